@@ -1,0 +1,308 @@
+#include "operations.h"
+
+#include <atomic>
+#include <condition_variable>
+#include <deque>
+#include <future>
+#include <thread>
+#include <unordered_map>
+
+#include "controller.h"
+#include "cpu_backend.h"
+#include "message.h"
+#include "timeline.h"
+
+namespace axon {
+
+namespace {
+
+constexpr const char* kShutdownBarrier = "__axon.shutdown.barrier__";
+
+struct Global {
+  std::atomic<bool> initialized{false};
+  std::atomic<bool> running{false};
+  std::atomic<bool> shutdown_requested{false};
+  int rank = 0, size = 1, local_rank = 0;
+  std::string control_path;
+  double cycle_ms = 1.0;
+
+  std::thread bg;
+  std::mutex mu;
+  std::condition_variable cv;
+  std::deque<TensorEntry> queue;
+
+  // Background-thread-only state.
+  std::unique_ptr<ControlPlane> cp;
+  std::unique_ptr<Controller> ctl;
+  std::unique_ptr<CpuBackend> cpu;
+  std::unique_ptr<GpuBackend> gpu;
+  std::unordered_map<std::string, TensorEntry> pending;
+  bool joined = false;
+  TensorEntry join_entry;
+  int32_t join_device = -1;
+  bool shutdown_barrier_sent = false;
+
+  Timeline tl;
+  std::mutex err_mu;
+  std::string fatal;
+};
+
+Global g;
+
+void set_fatal(const std::string& msg) {
+  std::lock_guard<std::mutex> lk(g.err_mu);
+  if (g.fatal.empty()) g.fatal = msg;
+}
+
+Request make_request(const TensorEntry& e) {
+  Request q;
+  q.rank = g.rank;
+  q.op = e.op;
+  q.red = e.red;
+  q.name = e.name;
+  if (e.in.defined()) {
+    q.dtype = dtype_id(e.in.scalar_type());
+    q.shape = e.in.sizes().vec();
+  }
+  q.wire_dtype =
+      e.wire_dtype == at::ScalarType::Undefined ? 0 : dtype_id(e.wire_dtype);
+  q.device = e.device;
+  q.root_rank = e.root_rank;
+  q.splits = e.splits;
+  return q;
+}
+
+TensorEntry make_zero_entry(const Response& r, size_t i) {
+  // A joined rank participates in peers' collectives with zero tensors
+  // (SURVEY.md §3.5 hvd.join semantics). Allreduce-shaped only.
+  TensorEntry e;
+  e.name = r.names[i];
+  e.op = r.op;
+  e.red = r.red;
+  auto opts = at::TensorOptions().dtype(id_dtype(r.dtype));
+  if (r.device >= 0)
+    opts = opts.device(at::Device(at::kCUDA, static_cast<int8_t>(g.join_device)));
+  e.in = at::zeros({r.numels.at(i)}, opts);
+  e.out = e.in;
+  e.device = r.device >= 0 ? g.join_device : -1;
+  return e;
+}
+
+void complete(TensorEntry& e, const Status& st) {
+  g.tl.end(e.name);
+  if (e.callback) e.callback(st);
+}
+
+void execute_response(const Response& r) {
+  if (r.op == OpType::JOIN) {
+    if (g.joined) {
+      if (g.join_entry.out.defined())
+        g.join_entry.out.fill_(static_cast<int64_t>(r.last_joined));
+      complete(g.join_entry, Status::OK());
+      g.join_entry = TensorEntry{};
+      g.joined = false;
+    }
+    return;
+  }
+
+  std::vector<TensorEntry> entries;
+  entries.reserve(r.names.size());
+  for (size_t i = 0; i < r.names.size(); ++i) {
+    auto it = g.pending.find(r.names[i]);
+    if (it != g.pending.end()) {
+      entries.push_back(std::move(it->second));
+      g.pending.erase(it);
+    } else if (g.joined && r.op == OpType::ALLREDUCE) {
+      entries.push_back(make_zero_entry(r, i));
+    } else {
+      AXON_LOG(LogLevel::ERROR,
+               "axon: response for unknown tensor '" + r.names[i] + "'");
+      return;
+    }
+  }
+
+  if (!r.error.empty()) {
+    for (auto& e : entries) complete(e, Status::Error(r.error));
+    return;
+  }
+
+  if (r.op == OpType::BARRIER) {
+    bool is_shutdown = !r.names.empty() && r.names[0] == kShutdownBarrier;
+    for (auto& e : entries) complete(e, Status::OK());
+    if (is_shutdown && g.rank == 0) g.ctl->coordinator_shutdown();
+    return;
+  }
+
+  for (auto& e : entries) g.tl.begin(e.name, r.device >= 0 ? "QUEUE_GPU" : "CPU_OP");
+
+  if (r.device < 0) {
+    Status st = g.cpu->execute(r, entries);
+    for (auto& e : entries) complete(e, st);
+  } else {
+    if (!g.gpu)
+      g.gpu.reset(create_rccl_backend(g.rank, g.size, g.cp.get(), &g.tl));
+    g.gpu->execute(r, std::move(entries));  // async; callbacks via poller
+  }
+}
+
+void abort_pending(const std::string& why) {
+  for (auto& kv : g.pending) complete(kv.second, Status::Aborted(why));
+  g.pending.clear();
+  std::deque<TensorEntry> leftover;
+  {
+    std::lock_guard<std::mutex> lk(g.mu);
+    leftover.swap(g.queue);
+  }
+  for (auto& e : leftover) complete(e, Status::Aborted(why));
+  if (g.joined) {
+    complete(g.join_entry, Status::Aborted(why));
+    g.joined = false;
+  }
+}
+
+void bg_loop(std::promise<void> ready) {
+  try {
+    double timeout = env_double("AXON_START_TIMEOUT", 120.0);
+    g.cp = std::make_unique<ControlPlane>();
+    if (g.rank == 0)
+      g.cp->init_coordinator(g.control_path, g.size, timeout);
+    else
+      g.cp->init_worker(g.control_path, g.rank, timeout);
+    g.ctl = std::make_unique<Controller>(g.rank, g.size, g.cp.get());
+    g.cpu = std::make_unique<CpuBackend>(g.rank, g.size, g.cp.get());
+    g.tl.init(g.rank);
+    ready.set_value();
+  } catch (...) {
+    ready.set_exception(std::current_exception());
+    return;
+  }
+
+  auto cycle = std::chrono::duration<double, std::milli>(g.cycle_ms);
+  bool shutdown = false;
+  try {
+    while (!shutdown) {
+      std::vector<TensorEntry> fresh;
+      {
+        std::unique_lock<std::mutex> lk(g.mu);
+        if (g.queue.empty() && !g.shutdown_requested.load())
+          g.cv.wait_for(lk, cycle);
+        while (!g.queue.empty()) {
+          fresh.push_back(std::move(g.queue.front()));
+          g.queue.pop_front();
+        }
+      }
+
+      std::vector<Request> reqs;
+      for (auto& e : fresh) {
+        if (e.op == OpType::JOIN) {
+          if (g.joined) {
+            complete(e, Status::Error("axon: join already in progress"));
+            continue;
+          }
+          g.joined = true;
+          g.join_entry = std::move(e);
+          g.join_device = g.join_entry.device;
+          reqs.push_back(make_request(g.join_entry));
+          continue;
+        }
+        if (g.pending.count(e.name)) {
+          complete(e, Status::Error("axon: tensor name '" + e.name +
+                                    "' already in flight — pass a unique name"));
+          continue;
+        }
+        g.tl.begin(e.name, "NEGOTIATE");
+        reqs.push_back(make_request(e));
+        g.pending.emplace(e.name, std::move(e));
+      }
+
+      if (g.shutdown_requested.load() && !g.shutdown_barrier_sent) {
+        TensorEntry e;
+        e.op = OpType::BARRIER;
+        e.name = kShutdownBarrier;
+        reqs.push_back(make_request(e));
+        g.pending.emplace(e.name, std::move(e));
+        g.shutdown_barrier_sent = true;
+      }
+
+      auto responses = g.ctl->run_cycle(reqs, &shutdown);
+      for (auto& r : responses) execute_response(r);
+      g.tl.mark_cycle();
+    }
+  } catch (const std::exception& ex) {
+    set_fatal(ex.what());
+    AXON_LOG(LogLevel::ERROR, std::string("axon background loop failed: ") + ex.what());
+  }
+
+  if (g.gpu) g.gpu->finalize();
+  abort_pending(g.fatal.empty() ? "axon shutting down" : g.fatal);
+  if (g.cp) g.cp->close_all();
+  g.running = false;
+}
+
+}  // namespace
+
+void initialize(int rank_, int size_, int local_rank_, const std::string& path) {
+  if (g.initialized.load())
+    throw std::runtime_error("axon already initialized");
+  g.rank = rank_;
+  g.size = size_;
+  g.local_rank = local_rank_;
+  g.control_path = path;
+  g.cycle_ms = env_double("AXON_CYCLE_TIME", 1.0);
+  g.shutdown_requested = false;
+  g.shutdown_barrier_sent = false;
+  {
+    std::lock_guard<std::mutex> lk(g.err_mu);
+    g.fatal.clear();
+  }
+  std::promise<void> ready;
+  auto fut = ready.get_future();
+  g.running = true;
+  g.bg = std::thread(bg_loop, std::move(ready));
+  try {
+    fut.get();
+  } catch (...) {
+    g.bg.join();
+    g.running = false;
+    throw;
+  }
+  g.initialized = true;
+}
+
+void shutdown() {
+  if (!g.initialized.load()) return;
+  g.shutdown_requested = true;
+  g.cv.notify_all();
+  if (g.bg.joinable()) g.bg.join();
+  g.initialized = false;
+  g.gpu.reset();
+  g.cpu.reset();
+  g.ctl.reset();
+  g.cp.reset();
+}
+
+bool is_initialized() { return g.initialized.load(); }
+int rank() { return g.rank; }
+int size() { return g.size; }
+int local_rank() { return g.local_rank; }
+
+std::string fatal_error() {
+  std::lock_guard<std::mutex> lk(g.err_mu);
+  return g.fatal;
+}
+
+void enqueue(TensorEntry entry) {
+  if (!g.initialized.load())
+    throw std::runtime_error("axon not initialized — call axon.init() first");
+  if (!g.running.load()) {
+    std::string f = fatal_error();
+    throw std::runtime_error(f.empty() ? "axon core stopped" : "axon core stopped: " + f);
+  }
+  {
+    std::lock_guard<std::mutex> lk(g.mu);
+    g.queue.push_back(std::move(entry));
+  }
+  g.cv.notify_all();
+}
+
+}  // namespace axon

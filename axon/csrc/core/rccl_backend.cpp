@@ -1,0 +1,651 @@
+// RCCL-over-xGMI GPU data plane (SURVEY.md §2.1 "NCCL ops -> RCCL ops",
+// §2.4 N1-N6, §5.8): a dedicated non-blocking HIP comm stream per rank,
+// lazily initialized RCCL communicator (unique id over the unix-socket
+// control plane), fusion-buffer pack/unpack via the CDNA4 kernels in
+// kernels/pack_unpack.hip, completion via a hipEvent poller thread, and a
+// distributed VHDD Adasum path (kernels/adasum.hip).
+//
+// MI355X notes: xGMI is 7 point-to-point links per GPU — RCCL stripes one
+// large buffer across links/channels, so the engine hands it FEW, LARGE
+// collectives (fusion threshold default 128 MiB, controller.cpp). Average
+// is folded into the collective itself (ncclAvg) instead of a separate
+// scale pass (in-collective average).
+#include <rccl/rccl.h>
+
+#include <atomic>
+#include <condition_variable>
+#include <deque>
+#include <limits>
+#include <map>
+#include <thread>
+
+#include "../kernels/kernels.h"
+#include "controller.h"
+#include "gpu_util.h"
+#include "message.h"
+#include "operations.h"
+#include "timeline.h"
+
+namespace axon {
+
+namespace {
+
+void nccl_check(ncclResult_t r, const char* what) {
+  if (r != ncclSuccess)
+    throw std::runtime_error(std::string("axon RCCL error in ") + what + ": " +
+                             ncclGetErrorString(r));
+}
+
+ncclDataType_t nccl_dtype(at::ScalarType t) {
+  switch (t) {
+    case at::ScalarType::Float: return ncclFloat32;
+    case at::ScalarType::Double: return ncclFloat64;
+    case at::ScalarType::Half: return ncclFloat16;
+    case at::ScalarType::BFloat16: return ncclBfloat16;
+    case at::ScalarType::Int: return ncclInt32;
+    case at::ScalarType::Long: return ncclInt64;
+    case at::ScalarType::Byte: return ncclUint8;
+    case at::ScalarType::Char: return ncclInt8;
+    case at::ScalarType::Bool: return ncclUint8;
+    default:
+      throw std::runtime_error("axon: unsupported GPU dtype " +
+                               std::string(at::toString(t)));
+  }
+}
+
+ncclRedOp_t nccl_red(ReduceOp r) {
+  switch (r) {
+    case ReduceOp::SUM: return ncclSum;
+    case ReduceOp::AVERAGE: return ncclAvg;
+    case ReduceOp::MIN: return ncclMin;
+    case ReduceOp::MAX: return ncclMax;
+    case ReduceOp::PRODUCT: return ncclProd;
+    default: throw std::runtime_error("axon: bad reduce op for RCCL");
+  }
+}
+
+constexpr size_t kAlign = 256;  // fusion-buffer sub-offsets: keep 16B vector
+                                // paths and xGMI-friendly alignment
+
+size_t align_up(size_t x) { return (x + kAlign - 1) & ~(kAlign - 1); }
+
+class RcclBackend : public GpuBackend {
+ public:
+  RcclBackend(int rank, int size, ControlPlane* cp, Timeline* tl)
+      : rank_(rank), size_(size), cp_(cp), tl_(tl) {}
+
+  ~RcclBackend() override { finalize(); }
+
+  void execute(const Response& r, std::vector<TensorEntry> entries) override {
+    try {
+      ensure_init(entries);
+      issue(r, entries);
+    } catch (const std::exception& e) {
+      AXON_LOG(LogLevel::ERROR, std::string("axon RCCL execute failed: ") + e.what());
+      for (auto& en : entries)
+        if (en.callback) en.callback(Status::Error(e.what()));
+      return;
+    }
+    // Completion event + poller hand-off.
+    hipEvent_t ev = EventPool::instance().get();
+    hip_check(hipEventRecord(ev, stream_), "hipEventRecord(done)");
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      inflight_.push_back(Inflight{ev, std::move(entries), std::move(used_slots_),
+                                   std::move(keep_alive_)});
+      used_slots_.clear();
+      keep_alive_.clear();
+    }
+    cv_.notify_all();
+  }
+
+  void finalize() override {
+    if (finalized_) return;
+    finalized_ = true;
+    {
+      std::unique_lock<std::mutex> lk(mu_);
+      cv_.wait(lk, [&] { return inflight_.empty(); });
+      stop_ = true;
+    }
+    cv_.notify_all();
+    if (poller_.joinable()) poller_.join();
+    if (comm_) {
+      hipStreamSynchronize(stream_);
+      ncclCommDestroy(comm_);
+      comm_ = nullptr;
+    }
+  }
+
+ private:
+  struct Slot {
+    ChunkDesc* host = nullptr;  // pinned
+    ChunkDesc* dev = nullptr;
+  };
+  struct Inflight {
+    hipEvent_t ev;
+    std::vector<TensorEntry> entries;
+    std::vector<int> slots;
+    std::vector<at::Tensor> keep;
+  };
+
+  void ensure_init(const std::vector<TensorEntry>& entries) {
+    if (comm_) return;
+    device_ = entries.at(0).device;
+    hip_check(hipSetDevice(device_), "hipSetDevice");
+    hip_check(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking),
+              "hipStreamCreate(comm)");
+    // RCCL unique id rendezvous over the control plane.
+    ncclUniqueId id;
+    if (rank_ == 0) {
+      nccl_check(ncclGetUniqueId(&id), "ncclGetUniqueId");
+      Writer w;
+      w.raw(&id, sizeof(id));
+      for (int p = 1; p < size_; ++p) cp_->send(p, MsgType::RENDEZVOUS, w.buf);
+    } else {
+      auto payload = cp_->recv_data(0);
+      if (payload.size() != sizeof(id))
+        throw std::runtime_error("axon: bad rccl unique id frame");
+      std::memcpy(&id, payload.data(), sizeof(id));
+    }
+    nccl_check(ncclCommInitRank(&comm_, size_, id, rank_), "ncclCommInitRank");
+
+    fuse_bytes_ = static_cast<size_t>(env_int("AXON_FUSION_THRESHOLD", 128 << 20));
+    hip_check(hipMalloc(&fusebuf_, fuse_bytes_ + kAlign), "hipMalloc(fusion buffer)");
+    hip_check(hipMalloc(&dots_mine_, 3 * sizeof(double)), "hipMalloc(dots)");
+    hip_check(hipMalloc(&dots_peer_, 3 * sizeof(double)), "hipMalloc(dots)");
+    poller_ = std::thread([this] { poll_loop(); });
+  }
+
+  int acquire_slot() {
+    std::lock_guard<std::mutex> lk(mu_);  // free_slots_ shared with poller
+    if (free_slots_.empty()) {
+      Slot s;
+      hip_check(hipHostMalloc(reinterpret_cast<void**>(&s.host),
+                              kMaxChunks * sizeof(ChunkDesc)),
+                "hipHostMalloc(descs)");
+      hip_check(hipMalloc(reinterpret_cast<void**>(&s.dev),
+                          kMaxChunks * sizeof(ChunkDesc)),
+                "hipMalloc(descs)");
+      slots_.push_back(s);
+      free_slots_.push_back(static_cast<int>(slots_.size()) - 1);
+    }
+    int id = free_slots_.back();
+    free_slots_.pop_back();
+    used_slots_.push_back(id);
+    return id;
+  }
+
+  void* ensure_scratch(size_t bytes) {
+    if (scratch_bytes_ >= bytes) return scratch_;
+    if (scratch_) retire_.push_back(scratch_);
+    scratch_bytes_ = std::max<size_t>(bytes, 64 << 20);
+    hip_check(hipMalloc(&scratch_, scratch_bytes_), "hipMalloc(scratch)");
+    return scratch_;
+  }
+
+  // Stage `chunks` to the device (one or more desc slots) and launch the
+  // given kernel over them.
+  template <typename LaunchFn>
+  void run_chunked(const std::vector<ChunkDesc>& chunks, LaunchFn&& fn) {
+    size_t off = 0;
+    while (off < chunks.size()) {
+      int n = static_cast<int>(std::min<size_t>(chunks.size() - off, kMaxChunks));
+      int sid = acquire_slot();
+      std::memcpy(slots_[sid].host, chunks.data() + off, n * sizeof(ChunkDesc));
+      hip_check(hipMemcpyAsync(slots_[sid].dev, slots_[sid].host,
+                               n * sizeof(ChunkDesc), hipMemcpyHostToDevice,
+                               stream_),
+                "hipMemcpyAsync(descs)");
+      fn(slots_[sid].dev, n);
+      off += static_cast<size_t>(n);
+    }
+  }
+
+  void wait_ready_events(std::vector<TensorEntry>& entries) {
+    for (auto& e : entries) {
+      if (e.ready_event) {
+        hipEvent_t ev = reinterpret_cast<hipEvent_t>(e.ready_event);
+        hip_check(hipStreamWaitEvent(stream_, ev, 0), "hipStreamWaitEvent(ready)");
+        EventPool::instance().put(ev);
+        e.ready_event = 0;
+      }
+    }
+  }
+
+  void issue(const Response& r, std::vector<TensorEntry>& entries) {
+    hip_check(hipSetDevice(device_), "hipSetDevice");
+    wait_ready_events(entries);
+    for (auto& e : entries) tl_->begin(e.name, "RCCL");
+    switch (r.op) {
+      case OpType::ALLREDUCE:
+        if (r.red == ReduceOp::ADASUM) adasum(entries);
+        else allreduce(r, entries);
+        break;
+      case OpType::BROADCAST: broadcast(r, entries); break;
+      case OpType::ALLGATHER: allgather(r, entries); break;
+      case OpType::ALLTOALL: alltoall(r, entries); break;
+      case OpType::REDUCESCATTER: reducescatter(r, entries); break;
+      default:
+        throw std::runtime_error("axon RCCL: unsupported op");
+    }
+  }
+
+  // Split [ptr, ptr+elems) into chunk descs of <= chunk_elems elements.
+  static void add_chunks(std::vector<ChunkDesc>& out, const void* src, void* dst,
+                         int64_t elems, size_t esize, int64_t chunk_elems) {
+    const char* s = static_cast<const char*>(src);
+    char* d = static_cast<char*>(dst);
+    for (int64_t off = 0; off < elems; off += chunk_elems) {
+      int64_t n = std::min<int64_t>(chunk_elems, elems - off);
+      out.push_back(ChunkDesc{s + off * esize, d + off * esize, n});
+    }
+  }
+
+  void allreduce(const Response& r, std::vector<TensorEntry>& entries) {
+    at::ScalarType dtype = id_dtype(r.dtype);
+    at::ScalarType wire = r.wire_dtype ? id_dtype(r.wire_dtype) : dtype;
+    bool any_scale = false;
+    for (auto& e : entries)
+      any_scale |= (e.prescale != 1.0 || e.postscale != 1.0);
+    ncclRedOp_t op = nccl_red(r.red);
+
+    if (entries.size() == 1 && wire == dtype && !any_scale) {
+      // Zero-copy path: single tensor, no transform — RCCL stripes the
+      // buffer across xGMI links itself.
+      auto& e = entries[0];
+      nccl_check(ncclAllReduce(e.in.data_ptr(), e.out.data_ptr(), e.in.numel(),
+                               nccl_dtype(dtype), op, comm_, stream_),
+                 "ncclAllReduce");
+      return;
+    }
+
+    // Fused path: pack (cast+prescale fused) -> allreduce(wire) -> unpack
+    // (cast back + postscale fused).
+    size_t wsize = dtype_size(wire);
+    size_t total = 0;
+    std::vector<size_t> offs(entries.size());
+    for (size_t i = 0; i < entries.size(); ++i) {
+      offs[i] = total;
+      total += align_up(static_cast<size_t>(entries[i].in.numel()) * wsize);
+    }
+    char* buf = static_cast<char*>(fusebuf_);
+    if (total > fuse_bytes_) buf = static_cast<char*>(ensure_scratch(total));
+
+    bool ident = wire == dtype;
+    size_t esize = dtype_size(dtype);
+    // Group pack chunks by prescale value (usually one group).
+    std::map<double, std::vector<ChunkDesc>> packs;
+    for (size_t i = 0; i < entries.size(); ++i) {
+      auto& e = entries[i];
+      if (ident && e.prescale == 1.0)
+        add_chunks(packs[std::numeric_limits<double>::infinity()],
+                   e.in.data_ptr(), buf + offs[i], e.in.numel() * esize, 1,
+                   kChunkBytes);
+      else
+        add_chunks(packs[e.prescale], e.in.data_ptr(), buf + offs[i],
+                   e.in.numel(), esize, kChunkElems);
+    }
+    for (auto& kv : packs) {
+      if (kv.first == std::numeric_limits<double>::infinity())
+        run_chunked(kv.second, [&](const ChunkDesc* d, int n) {
+          launch_copy_bytes(d, n, stream_);
+        });
+      else {
+        float sc = static_cast<float>(kv.first);
+        run_chunked(kv.second, [&](const ChunkDesc* d, int n) {
+          launch_xform(d, n, static_cast<int>(dtype), static_cast<int>(wire), sc,
+                       stream_);
+        });
+      }
+    }
+
+    int64_t wire_elems = static_cast<int64_t>(total / wsize);
+    nccl_check(ncclAllReduce(buf, buf, wire_elems, nccl_dtype(wire), op, comm_,
+                             stream_),
+               "ncclAllReduce(fused)");
+
+    std::map<double, std::vector<ChunkDesc>> unpacks;
+    for (size_t i = 0; i < entries.size(); ++i) {
+      auto& e = entries[i];
+      if (ident && e.postscale == 1.0)
+        add_chunks(unpacks[std::numeric_limits<double>::infinity()],
+                   buf + offs[i], e.out.data_ptr(), e.out.numel() * esize, 1,
+                   kChunkBytes);
+      else
+        add_chunks(unpacks[e.postscale], buf + offs[i], e.out.data_ptr(),
+                   e.out.numel(), wsize, kChunkElems);
+    }
+    for (auto& kv : unpacks) {
+      if (kv.first == std::numeric_limits<double>::infinity())
+        run_chunked(kv.second, [&](const ChunkDesc* d, int n) {
+          launch_copy_bytes(d, n, stream_);
+        });
+      else {
+        float sc = static_cast<float>(kv.first);
+        run_chunked(kv.second, [&](const ChunkDesc* d, int n) {
+          launch_xform(d, n, static_cast<int>(wire), static_cast<int>(dtype), sc,
+                       stream_);
+        });
+      }
+    }
+  }
+
+  void broadcast(const Response& r, std::vector<TensorEntry>& entries) {
+    for (auto& e : entries) {
+      nccl_check(ncclBroadcast(e.in.defined() ? e.in.data_ptr() : e.out.data_ptr(),
+                               e.out.data_ptr(), e.out.numel(),
+                               nccl_dtype(e.out.scalar_type()), r.root_rank,
+                               comm_, stream_),
+                 "ncclBroadcast");
+    }
+  }
+
+  void allgather(const Response& r, std::vector<TensorEntry>& entries) {
+    for (size_t t = 0; t < entries.size(); ++t) {
+      auto& e = entries[t];
+      const auto& fd = r.first_dims.at(t);
+      auto in_sizes = e.in.sizes().vec();
+      int64_t row = 1;
+      for (size_t d = 1; d < in_sizes.size(); ++d) row *= in_sizes[d];
+      int64_t total_rows = 0;
+      bool equal = true;
+      for (auto x : fd) {
+        total_rows += x;
+        equal &= (x == fd[0]);
+      }
+      auto out_sizes = in_sizes;
+      if (out_sizes.empty()) out_sizes.push_back(total_rows);
+      else out_sizes[0] = total_rows;
+      e.out.resize_(out_sizes);
+      size_t esize = e.in.element_size();
+      auto dt = nccl_dtype(e.in.scalar_type());
+      char* ob = static_cast<char*>(e.out.data_ptr());
+      if (equal && fd[0] > 0) {
+        nccl_check(ncclAllGather(e.in.data_ptr(), ob, fd[0] * row, dt, comm_,
+                                 stream_),
+                   "ncclAllGather");
+      } else {
+        // Variable first dim: direct p2p exchange over the 7 xGMI links
+        // (SURVEY.md §2.4 N3).
+        int64_t my_numel = fd[rank_] * row;
+        std::vector<int64_t> off(size_, 0);
+        for (int p = 1; p < size_; ++p) off[p] = off[p - 1] + fd[p - 1] * row;
+        nccl_check(ncclGroupStart(), "ncclGroupStart");
+        for (int p = 0; p < size_; ++p) {
+          if (p == rank_) continue;
+          if (my_numel > 0)
+            nccl_check(ncclSend(e.in.data_ptr(), my_numel, dt, p, comm_, stream_),
+                       "ncclSend");
+          int64_t pn = fd[p] * row;
+          if (pn > 0)
+            nccl_check(ncclRecv(ob + off[p] * esize, pn, dt, p, comm_, stream_),
+                       "ncclRecv");
+        }
+        nccl_check(ncclGroupEnd(), "ncclGroupEnd");
+        if (my_numel > 0)
+          hip_check(hipMemcpyAsync(ob + off[rank_] * esize, e.in.data_ptr(),
+                                   my_numel * esize, hipMemcpyDeviceToDevice,
+                                   stream_),
+                    "hipMemcpyAsync(self slab)");
+      }
+    }
+  }
+
+  void alltoall(const Response& r, std::vector<TensorEntry>& entries) {
+    auto m = [&](int src, int dst) {
+      return r.splits_matrix.at(static_cast<size_t>(src) * size_ + dst);
+    };
+    for (auto& e : entries) {
+      auto in_sizes = e.in.sizes().vec();
+      int64_t row = 1;
+      for (size_t d = 1; d < in_sizes.size(); ++d) row *= in_sizes[d];
+      int64_t rows_me = 0;
+      for (int src = 0; src < size_; ++src) rows_me += m(src, rank_);
+      auto out_sizes = in_sizes;
+      if (out_sizes.empty()) out_sizes.push_back(rows_me);
+      else out_sizes[0] = rows_me;
+      e.out.resize_(out_sizes);
+      size_t esize = e.in.element_size();
+      auto dt = nccl_dtype(e.in.scalar_type());
+      const char* ib = static_cast<const char*>(e.in.data_ptr());
+      char* ob = static_cast<char*>(e.out.data_ptr());
+      std::vector<int64_t> soff(size_, 0), roff(size_, 0);
+      for (int p = 1; p < size_; ++p) {
+        soff[p] = soff[p - 1] + m(rank_, p - 1) * row;
+        roff[p] = roff[p - 1] + m(p - 1, rank_) * row;
+      }
+      nccl_check(ncclGroupStart(), "ncclGroupStart");
+      for (int p = 0; p < size_; ++p) {
+        if (p == rank_) continue;
+        int64_t sn = m(rank_, p) * row, rn = m(p, rank_) * row;
+        if (sn > 0)
+          nccl_check(ncclSend(ib + soff[p] * esize, sn, dt, p, comm_, stream_),
+                     "ncclSend");
+        if (rn > 0)
+          nccl_check(ncclRecv(ob + roff[p] * esize, rn, dt, p, comm_, stream_),
+                     "ncclRecv");
+      }
+      nccl_check(ncclGroupEnd(), "ncclGroupEnd");
+      int64_t self_n = m(rank_, rank_) * row;
+      if (self_n > 0)
+        hip_check(hipMemcpyAsync(ob + roff[rank_] * esize, ib + soff[rank_] * esize,
+                                 self_n * esize, hipMemcpyDeviceToDevice, stream_),
+                  "hipMemcpyAsync(self)");
+      if (e.received_splits.defined()) {
+        auto rs = e.received_splits;
+        rs.resize_({size_});
+        for (int src = 0; src < size_; ++src) rs[src] = m(src, rank_);
+      }
+    }
+  }
+
+  void reducescatter(const Response& r, std::vector<TensorEntry>& entries) {
+    for (auto& e : entries) {
+      auto in_sizes = e.in.sizes().vec();
+      int64_t rows = in_sizes.empty() ? 1 : in_sizes[0];
+      int64_t row = 1;
+      for (size_t d = 1; d < in_sizes.size(); ++d) row *= in_sizes[d];
+      int64_t base = rows / size_, rem = rows % size_;
+      int64_t my_rows = base + (rank_ < rem ? 1 : 0);
+      auto out_sizes = in_sizes;
+      if (out_sizes.empty()) out_sizes.push_back(my_rows);
+      else out_sizes[0] = my_rows;
+      e.out.resize_(out_sizes);
+      auto dt = nccl_dtype(e.in.scalar_type());
+      ncclRedOp_t op = nccl_red(r.red);
+      if (rem == 0) {
+        nccl_check(ncclReduceScatter(e.in.data_ptr(), e.out.data_ptr(),
+                                     base * row, dt, op, comm_, stream_),
+                   "ncclReduceScatter");
+      } else {
+        // Uneven split: allreduce into scratch, slice locally. Correct but
+        // 2x traffic — documented limitation; even splits take the fast path.
+        size_t esize = e.in.element_size();
+        char* buf = static_cast<char*>(ensure_scratch(e.in.numel() * esize));
+        nccl_check(ncclAllReduce(e.in.data_ptr(), buf, e.in.numel(), dt, op,
+                                 comm_, stream_),
+                   "ncclAllReduce(rs-uneven)");
+        int64_t off = rank_ * base + std::min<int64_t>(rank_, rem);
+        hip_check(hipMemcpyAsync(e.out.data_ptr(), buf + off * row * esize,
+                                 my_rows * row * esize, hipMemcpyDeviceToDevice,
+                                 stream_),
+                  "hipMemcpyAsync(rs slice)");
+      }
+    }
+  }
+
+  // Distributed Adasum: vector-halving distance-doubling with per-level
+  // full-vector dot triples reduced over the 2^(k+1)-rank group by a
+  // butterfly of 24-byte p2p exchanges (SURVEY.md §2.1 "Adasum ops"; math
+  // parity with cpu_backend.cpp's fp64 oracle).
+  void adasum(std::vector<TensorEntry>& entries) {
+    if ((size_ & (size_ - 1)) != 0)
+      throw std::runtime_error("adasum requires a power-of-two world size");
+    for (auto& e : entries) {
+      int dtype = static_cast<int>(e.in.scalar_type());
+      size_t esize = e.in.element_size();
+      int64_t n = e.in.numel();
+      char* b = static_cast<char*>(e.out.data_ptr());
+      if (e.out.data_ptr() != e.in.data_ptr())
+        hip_check(hipMemcpyAsync(b, e.in.data_ptr(), n * esize,
+                                 hipMemcpyDeviceToDevice, stream_),
+                  "hipMemcpyAsync(adasum in->out)");
+      if (e.prescale != 1.0) {
+        std::vector<ChunkDesc> c;
+        add_chunks(c, b, b, n, esize, kChunkElems);
+        run_chunked(c, [&](const ChunkDesc* d, int nn) {
+          launch_xform(d, nn, dtype, dtype, static_cast<float>(e.prescale), stream_);
+        });
+      }
+      char* scratch =
+          static_cast<char*>(ensure_scratch(((n + 1) / 2 + kAlign) * esize));
+      auto dt = nccl_dtype(e.in.scalar_type());
+
+      int64_t off = 0, len = n;
+      for (int stride = 1; stride < size_; stride <<= 1) {
+        int peer = rank_ ^ stride;
+        bool a_side = (rank_ & stride) == 0;
+        int64_t len1 = (len + 1) / 2, len2 = len - len1;
+        int64_t keep_off = a_side ? off : off + len1;
+        int64_t keep_len = a_side ? len1 : len2;
+        int64_t send_off = a_side ? off + len1 : off;
+        int64_t send_len = a_side ? len2 : len1;
+        // 1) swap halves with the level peer.
+        nccl_check(ncclGroupStart(), "ncclGroupStart");
+        if (send_len > 0)
+          nccl_check(ncclSend(b + send_off * esize, send_len, dt, peer, comm_,
+                              stream_),
+                     "ncclSend(adasum half)");
+        if (keep_len > 0)
+          nccl_check(ncclRecv(scratch, keep_len, dt, peer, comm_, stream_),
+                     "ncclRecv(adasum half)");
+        nccl_check(ncclGroupEnd(), "ncclGroupEnd");
+        // 2) local partial dot triple (canonical a/b order).
+        hip_check(hipMemsetAsync(dots_mine_, 0, 3 * sizeof(double), stream_),
+                  "hipMemset(dots)");
+        if (keep_len > 0)
+          launch_adasum_dots(b + keep_off * esize, scratch, keep_len, dtype,
+                             dots_mine_, a_side, stream_);
+        // 3) butterfly-sum the triple across the 2*stride-rank group.
+        for (int j = 1; j <= stride; j <<= 1) {
+          int dpeer = rank_ ^ j;
+          nccl_check(ncclGroupStart(), "ncclGroupStart");
+          nccl_check(ncclSend(dots_mine_, 3, ncclFloat64, dpeer, comm_, stream_),
+                     "ncclSend(dots)");
+          nccl_check(ncclRecv(dots_peer_, 3, ncclFloat64, dpeer, comm_, stream_),
+                     "ncclRecv(dots)");
+          nccl_check(ncclGroupEnd(), "ncclGroupEnd");
+          launch_add3(dots_mine_, dots_peer_, stream_);
+        }
+        // 4) apply combine coefficients to my kept half.
+        if (keep_len > 0)
+          launch_adasum_apply(b + keep_off * esize, scratch, keep_len, dtype,
+                              dots_mine_, a_side, stream_);
+        off = keep_off;
+        len = keep_len;
+      }
+
+      // 5) allgather the final segments (deterministic replay of every
+      // rank's offsets) via direct p2p.
+      auto segment_of = [&](int rk) {
+        int64_t o = 0, l = n;
+        for (int stride = 1; stride < size_; stride <<= 1) {
+          int64_t l1 = (l + 1) / 2;
+          if ((rk & stride) == 0) l = l1;
+          else { o += l1; l -= l1; }
+        }
+        return std::make_pair(o, l);
+      };
+      nccl_check(ncclGroupStart(), "ncclGroupStart");
+      for (int p = 0; p < size_; ++p) {
+        if (p == rank_) continue;
+        if (len > 0)
+          nccl_check(ncclSend(b + off * esize, len, dt, p, comm_, stream_),
+                     "ncclSend(adasum gather)");
+        auto [po, pl] = segment_of(p);
+        if (pl > 0)
+          nccl_check(ncclRecv(b + po * esize, pl, dt, p, comm_, stream_),
+                     "ncclRecv(adasum gather)");
+      }
+      nccl_check(ncclGroupEnd(), "ncclGroupEnd");
+      if (e.postscale != 1.0) {
+        std::vector<ChunkDesc> c;
+        add_chunks(c, b, b, n, esize, kChunkElems);
+        run_chunked(c, [&](const ChunkDesc* d, int nn) {
+          launch_xform(d, nn, dtype, dtype, static_cast<float>(e.postscale), stream_);
+        });
+      }
+    }
+  }
+
+  void poll_loop() {
+    hip_check(hipSetDevice(device_), "hipSetDevice(poller)");
+    for (;;) {
+      Inflight* front = nullptr;
+      {
+        std::unique_lock<std::mutex> lk(mu_);
+        cv_.wait(lk, [&] { return stop_ || !inflight_.empty(); });
+        if (stop_ && inflight_.empty()) return;
+        front = &inflight_.front();
+      }
+      hipError_t q = hipEventQuery(front->ev);
+      if (q == hipErrorNotReady) {
+        std::this_thread::sleep_for(std::chrono::microseconds(100));
+        continue;
+      }
+      Status st = q == hipSuccess
+                      ? Status::OK()
+                      : Status::Error(std::string("axon: comm stream error: ") +
+                                      hipGetErrorString(q));
+      for (auto& e : front->entries) {
+        tl_->end(e.name);
+        if (e.callback) e.callback(st);
+      }
+      EventPool::instance().put(front->ev);
+      {
+        std::lock_guard<std::mutex> lk(mu_);
+        for (int sid : front->slots) free_slots_.push_back(sid);
+        inflight_.pop_front();
+        if (inflight_.empty()) {
+          for (void* p : retire_) hipFree(p);
+          retire_.clear();
+        }
+      }
+      cv_.notify_all();
+    }
+  }
+
+  int rank_, size_;
+  ControlPlane* cp_;
+  Timeline* tl_;
+  int device_ = 0;
+  ncclComm_t comm_ = nullptr;
+  hipStream_t stream_ = nullptr;
+  void* fusebuf_ = nullptr;
+  size_t fuse_bytes_ = 0;
+  void* scratch_ = nullptr;
+  size_t scratch_bytes_ = 0;
+  std::vector<void*> retire_;
+  double* dots_mine_ = nullptr;
+  double* dots_peer_ = nullptr;
+
+  std::vector<Slot> slots_;
+  std::vector<int> free_slots_;
+  std::vector<int> used_slots_;       // being assembled for current op
+  std::vector<at::Tensor> keep_alive_;
+
+  std::mutex mu_;
+  std::condition_variable cv_;
+  std::deque<Inflight> inflight_;
+  std::thread poller_;
+  std::atomic<bool> stop_{false};
+  bool finalized_ = false;
+};
+
+}  // namespace
+
+GpuBackend* create_rccl_backend(int rank, int size, ControlPlane* cp, Timeline* tl) {
+  return new RcclBackend(rank, size, cp, tl);
+}
+
+}  // namespace axon

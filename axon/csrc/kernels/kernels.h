@@ -1,0 +1,50 @@
+// Launch interface of the hand-written CDNA4 HIP kernels (gfx950).
+// K1/K2 batched fusion-buffer pack/unpack, K3 scale (fused
+// into pack/unpack), K4 fp16 wire compress/decompress (fused likewise),
+// K5 Adasum dots + scaled-add. K6 fused optimizers live in optimizers.hip.
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+namespace axon {
+
+// One contiguous transfer chunk: pointers are pre-offset to the chunk start.
+// n = element count (xform kernels) or byte count (byte-copy kernel).
+struct ChunkDesc {
+  const void* src;
+  void* dst;
+  int64_t n;
+};
+
+// Max chunks staged per launch (device descriptor buffer size).
+constexpr int kMaxChunks = 16384;
+// Elements processed per chunk/block (xform); bytes per chunk (copy).
+constexpr int64_t kChunkElems = 1 << 16;   // 64K elements
+constexpr int64_t kChunkBytes = 1 << 18;   // 256 KiB
+
+// Batched memcpy (dtype-agnostic, n = bytes). descs must be in device
+// memory; nchunks <= kMaxChunks.
+void launch_copy_bytes(const ChunkDesc* descs, int nchunks, hipStream_t stream);
+
+// Batched elementwise convert+scale: dst[i] = DstT(float(src[i]) * scale).
+// src_dtype/dst_dtype are at::ScalarType codes (float/half/bfloat16/double).
+// n = elements.
+void launch_xform(const ChunkDesc* descs, int nchunks, int src_dtype,
+                  int dst_dtype, float scale, hipStream_t stream);
+bool xform_supported(int src_dtype, int dst_dtype);
+
+// Adasum (K5): partial dot triple over this rank's segment.
+// out3 += {a.b, a.a, b.b} in canonical a/b order (a_side tells which of
+// kept/recv is 'a'). Accumulates in double; caller zeroes out3 first.
+void launch_adasum_dots(const void* kept, const void* recv, int64_t n,
+                        int dtype, double* out3, bool a_side, hipStream_t stream);
+// mine[0..2] += peer[0..2] (adasum partial-dot butterfly merge).
+void launch_add3(double* mine, const double* peer, hipStream_t stream);
+// kept = ca*kept + cb*recv (a_side: kept is 'a') with
+// ca = 1 - ab/(2*aa), cb = 1 - ab/(2*bb) read from dots3 on device.
+void launch_adasum_apply(void* kept, const void* recv, int64_t n, int dtype,
+                         const double* dots3, bool a_side, hipStream_t stream);
+
+}  // namespace axon

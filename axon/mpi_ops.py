@@ -1,0 +1,267 @@
+"""Tensor collective API: the 4-way (sync/async x in-place/out-of-place)
+op families plus synchronize/poll/join/barrier.
+
+Capability parity: horovod/torch/mpi_ops.py (SURVEY.md §2.2 "Tensor ops
+API"). Ops are enqueued to the C++ core (axon/_core.so) which coordinates
+order across ranks and runs RCCL on a dedicated HIP stream.
+"""
+import threading
+
+import torch
+
+from axon.basics import _ensure_init, rank, size
+
+
+class ReduceOp:
+    Average = 0
+    Sum = 1
+    Adasum = 2
+    Min = 3
+    Max = 4
+    Product = 5
+
+
+# Horovod-compatible module-level aliases.
+Average = ReduceOp.Average
+Sum = ReduceOp.Sum
+Adasum = ReduceOp.Adasum
+Min = ReduceOp.Min
+Max = ReduceOp.Max
+Product = ReduceOp.Product
+
+_lock = threading.Lock()
+_counters = {}
+_outputs = {}  # handle -> result payload returned by synchronize
+
+
+def _auto_name(kind, name):
+    if name is not None:
+        return f"{kind}.{name}"
+    # Deterministic per-op counter: every rank enqueues ops in the same
+    # program order, so the counter-derived names agree across ranks.
+    with _lock:
+        n = _counters.get(kind, 0)
+        _counters[kind] = n + 1
+    return f"{kind}.noname.{n}"
+
+
+def _register(handle, payload):
+    with _lock:
+        _outputs[handle] = payload
+    return handle
+
+
+def _wire(compression):
+    wd = getattr(compression, "wire_dtype", None) if compression is not None else None
+    if wd is None:
+        return -1
+    return {torch.float16: 5, torch.bfloat16: 15, torch.float32: 6}[wd]
+
+
+def _resolve_op(op, average):
+    if op is not None and average is not None:
+        raise ValueError("specify either op or average, not both")
+    if op is None:
+        if average is None or average:
+            return ReduceOp.Average
+        return ReduceOp.Sum
+    return op
+
+
+# --------------------------------------------------------------- allreduce
+
+def allreduce_async(tensor, average=None, name=None, op=None,
+                    prescale_factor=1.0, postscale_factor=1.0,
+                    compression=None):
+    """Out-of-place async allreduce; returns a handle."""
+    red = _resolve_op(op, average)
+    t = tensor.contiguous()
+    output = torch.empty_like(t)
+    core = _ensure_init()
+    h = core.allreduce_async(t, output, _auto_name("allreduce", name), red,
+                             prescale_factor, postscale_factor,
+                             _wire(compression))
+    return _register(h, output)
+
+
+def allreduce(tensor, average=None, name=None, op=None, prescale_factor=1.0,
+              postscale_factor=1.0, compression=None):
+    return synchronize(allreduce_async(tensor, average, name, op,
+                                       prescale_factor, postscale_factor,
+                                       compression))
+
+
+def allreduce_async_(tensor, average=None, name=None, op=None,
+                     prescale_factor=1.0, postscale_factor=1.0,
+                     compression=None):
+    """In-place async allreduce (the hot path for gradients)."""
+    if not tensor.is_contiguous():
+        raise ValueError("allreduce_ requires a contiguous tensor")
+    red = _resolve_op(op, average)
+    core = _ensure_init()
+    h = core.allreduce_async(tensor, tensor, _auto_name("allreduce", name), red,
+                             prescale_factor, postscale_factor,
+                             _wire(compression))
+    return _register(h, tensor)
+
+
+def allreduce_(tensor, average=None, name=None, op=None, prescale_factor=1.0,
+               postscale_factor=1.0, compression=None):
+    return synchronize(allreduce_async_(tensor, average, name, op,
+                                        prescale_factor, postscale_factor,
+                                        compression))
+
+
+def grouped_allreduce_async(tensors, average=None, name=None, op=None,
+                            prescale_factor=1.0, postscale_factor=1.0,
+                            compression=None):
+    """Allreduce a list of tensors as one logical group. The group is
+    negotiated in a single cycle, so the controller fuses it into as few
+    collectives as the fusion threshold allows."""
+    gname = _auto_name("grouped", name)
+    return [allreduce_async(t, average, f"{gname}.{i}", op, prescale_factor,
+                            postscale_factor, compression)
+            for i, t in enumerate(tensors)]
+
+
+def grouped_allreduce(tensors, **kw):
+    return [synchronize(h) for h in grouped_allreduce_async(tensors, **kw)]
+
+
+def grouped_allreduce_async_(tensors, average=None, name=None, op=None,
+                             prescale_factor=1.0, postscale_factor=1.0,
+                             compression=None):
+    gname = _auto_name("grouped", name)
+    return [allreduce_async_(t, average, f"{gname}.{i}", op, prescale_factor,
+                             postscale_factor, compression)
+            for i, t in enumerate(tensors)]
+
+
+def grouped_allreduce_(tensors, **kw):
+    return [synchronize(h) for h in grouped_allreduce_async_(tensors, **kw)]
+
+
+# --------------------------------------------------------------- broadcast
+
+def broadcast_async(tensor, root_rank, name=None):
+    t = tensor.contiguous()
+    output = torch.empty_like(t)
+    core = _ensure_init()
+    h = core.broadcast_async(t, output, root_rank, _auto_name("broadcast", name))
+    return _register(h, output)
+
+
+def broadcast(tensor, root_rank, name=None):
+    return synchronize(broadcast_async(tensor, root_rank, name))
+
+
+def broadcast_async_(tensor, root_rank, name=None):
+    if not tensor.is_contiguous():
+        raise ValueError("broadcast_ requires a contiguous tensor")
+    core = _ensure_init()
+    h = core.broadcast_async(tensor, tensor, root_rank,
+                             _auto_name("broadcast", name))
+    return _register(h, tensor)
+
+
+def broadcast_(tensor, root_rank, name=None):
+    return synchronize(broadcast_async_(tensor, root_rank, name))
+
+
+# --------------------------------------------------------------- allgather
+
+def allgather_async(tensor, name=None):
+    t = tensor.contiguous()
+    output = t.new_empty(0)
+    core = _ensure_init()
+    h = core.allgather_async(t, output, _auto_name("allgather", name))
+    return _register(h, output)
+
+
+def allgather(tensor, name=None):
+    return synchronize(allgather_async(tensor, name))
+
+
+# ---------------------------------------------------------------- alltoall
+
+def alltoall_async(tensor, splits=None, name=None):
+    t = tensor.contiguous()
+    n = size()
+    if splits is None:
+        rows = t.shape[0] if t.dim() > 0 else t.numel()
+        if rows % n != 0:
+            raise ValueError("alltoall without splits requires dim0 divisible "
+                             "by world size")
+        splits_list = [rows // n] * n
+    else:
+        splits_list = [int(x) for x in splits]
+    output = t.new_empty(0)
+    received_splits = torch.empty(n, dtype=torch.int64)
+    core = _ensure_init()
+    h = core.alltoall_async(t, output, splits_list, received_splits,
+                            _auto_name("alltoall", name))
+    return _register(h, (output, received_splits))
+
+
+def alltoall(tensor, splits=None, name=None):
+    out, received = synchronize(alltoall_async(tensor, splits, name))
+    if splits is None:
+        return out
+    return out, received
+
+
+# ----------------------------------------------------------- reducescatter
+
+def reducescatter_async(tensor, op=ReduceOp.Average, name=None,
+                        prescale_factor=1.0, postscale_factor=1.0):
+    if op not in (ReduceOp.Average, ReduceOp.Sum):
+        raise ValueError("reducescatter supports Average and Sum")
+    t = tensor.contiguous()
+    output = t.new_empty(0)
+    core = _ensure_init()
+    h = core.reducescatter_async(t, output, _auto_name("reducescatter", name),
+                                 op, prescale_factor, postscale_factor)
+    return _register(h, output)
+
+
+def reducescatter(tensor, op=ReduceOp.Average, name=None, prescale_factor=1.0,
+                  postscale_factor=1.0):
+    return synchronize(reducescatter_async(tensor, op, name, prescale_factor,
+                                           postscale_factor))
+
+
+# ------------------------------------------------------------ join/barrier
+
+def join(device=-1):
+    """Uneven-data barrier: a joined rank participates in
+    the remaining ranks' allreduces with zero tensors until every rank has
+    joined. Returns the rank of the last rank to join."""
+    core = _ensure_init()
+    out = torch.zeros((), dtype=torch.int64)
+    h = core.join_async(device, out)
+    _register(h, out)
+    synchronize(h)
+    return int(out.item())
+
+
+def barrier(name=None):
+    core = _ensure_init()
+    h = core.barrier_async(_auto_name("barrier", name))
+    _register(h, None)
+    synchronize(h)
+
+
+# ------------------------------------------------------------ handle mgmt
+
+def synchronize(handle):
+    """Block until the async op behind `handle` completes; returns its
+    output tensor (or tuple)."""
+    core = _ensure_init()
+    core.synchronize(handle)
+    with _lock:
+        return _outputs.pop(handle, None)
+
+
+def poll(handle):
+    core = _ensure_init()
+    return core.poll(handle)
