@@ -109,7 +109,10 @@ std::vector<Response> Controller::run_cycle(const std::vector<Request>& own,
     std::vector<Response> out;
     MsgType t;
     std::vector<uint8_t> payload;
-    while (cp_->try_recv_control(0, &t, &payload)) {
+    // Stop polling once the shutdown flag is seen: the coordinator closes
+    // its end right after the final broadcast, and polling past it would
+    // misread the EOF as a peer failure.
+    while (!*shutdown && cp_->try_recv_control(0, &t, &payload)) {
       if (t == MsgType::RESPONSE_LIST) {
         ResponseList rl = ResponseList::deserialize(payload);
         if (rl.shutdown) *shutdown = true;

@@ -178,6 +178,12 @@ void bg_loop(std::promise<void> ready) {
   }
 
   auto cycle = std::chrono::duration<double, std::milli>(g.cycle_ms);
+  // Bounded shutdown: a rank exiting early (sys.exit, exception) must not
+  // hang in atexit waiting for peers that never call shutdown(). After the
+  // timeout the loop force-exits; closing our sockets fail-fasts the peers
+  // (SURVEY.md §5.3b).
+  const double shutdown_timeout = env_double("AXON_SHUTDOWN_TIMEOUT", 30.0);
+  std::chrono::steady_clock::time_point shutdown_since{};
   bool shutdown = false;
   try {
     while (!shutdown) {
@@ -222,6 +228,16 @@ void bg_loop(std::promise<void> ready) {
         reqs.push_back(make_request(e));
         g.pending.emplace(e.name, std::move(e));
         g.shutdown_barrier_sent = true;
+        shutdown_since = std::chrono::steady_clock::now();
+      }
+      if (g.shutdown_barrier_sent && !shutdown &&
+          std::chrono::duration<double>(std::chrono::steady_clock::now() -
+                                        shutdown_since).count() > shutdown_timeout) {
+        AXON_LOG(LogLevel::WARN,
+                 "axon: shutdown barrier timed out after " +
+                     std::to_string(shutdown_timeout) +
+                     "s (peers still running?) — forcing local shutdown");
+        break;
       }
 
       auto responses = g.ctl->run_cycle(reqs, &shutdown);

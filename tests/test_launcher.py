@@ -1,0 +1,87 @@
+"""Launcher (axonrun) behavior: rank env pinning, GPU pinning env,
+fail-fast kill-tree on first failure, timeline env
+plumbing.
+"""
+import json
+import os
+import subprocess
+import sys
+import tempfile
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _run(np_, script, extra_args=(), timeout=120):
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["AXON_SHUTDOWN_TIMEOUT"] = "3"  # keep fail-fast tests quick
+    return subprocess.run(
+        [sys.executable, "-m", "axon.runner.run", "-np", str(np_),
+         *extra_args, sys.executable, "-c", script],
+        capture_output=True, text=True, env=env, timeout=timeout)
+
+
+def test_launcher_env_and_collective():
+    script = (
+        "import os, torch, axon as hvd\n"
+        "hvd.init()\n"
+        "assert hvd.size() == 2\n"
+        "assert os.environ['HIP_VISIBLE_DEVICES'] == str(hvd.rank())\n"
+        "out = hvd.allreduce(torch.ones(3) * (hvd.rank() + 1), average=True)\n"
+        "assert abs(out[0].item() - 1.5) < 1e-6\n"
+        "print('RANK_OK', hvd.rank())\n"
+        "hvd.shutdown()\n")
+    r = _run(2, script)
+    assert r.returncode == 0, r.stderr
+    assert "RANK_OK 0" in r.stdout and "RANK_OK 1" in r.stdout
+
+
+def test_launcher_failfast():
+    # rank 1 exits non-zero immediately; rank 0 would hang in a collective
+    # forever — the launcher must kill it and propagate the failure code.
+    script = (
+        "import os, sys, torch, axon as hvd\n"
+        "hvd.init()\n"
+        "if hvd.rank() == 1:\n"
+        "    sys.exit(7)\n"
+        "try:\n"
+        "    hvd.allreduce(torch.ones(3), average=True)\n"
+        "except RuntimeError:\n"
+        "    sys.exit(1)  # fail-fast surfaced as an exception: also fine\n"
+        "import time; time.sleep(600)\n")
+    r = _run(2, script, timeout=180)
+    assert r.returncode != 0
+    assert "terminating remaining ranks" in r.stderr
+
+
+def test_launcher_no_pin():
+    script = (
+        "import os, axon as hvd\n"
+        "hvd.init()\n"
+        "assert 'HIP_VISIBLE_DEVICES' not in os.environ or "
+        "os.environ.get('AXON_PINNED') is None\n"
+        "hvd.shutdown()\n")
+    r = _run(2, script, extra_args=("--no-pin",))
+    assert r.returncode == 0, r.stderr
+
+
+def test_timeline_written():
+    with tempfile.TemporaryDirectory() as d:
+        path = os.path.join(d, "tl.json")
+        script = (
+            "import torch, axon as hvd\n"
+            "hvd.init()\n"
+            "for i in range(3):\n"
+            "    hvd.allreduce(torch.ones(8), average=True, name=f't{i}')\n"
+            "hvd.shutdown()\n")
+        r = _run(2, script, extra_args=("--timeline", path))
+        assert r.returncode == 0, r.stderr
+        f0 = os.path.join(d, "tl.0.json")
+        assert os.path.exists(f0), os.listdir(d)
+        content = open(f0).read().rstrip().rstrip(",")
+        events = json.loads(content + "]")
+        names = {e.get("name") for e in events}
+        assert "NEGOTIATE" in names and "CPU_OP" in names
+        # per-tensor lanes got labeled
+        lanes = [e for e in events if e.get("name") == "thread_name"]
+        assert any("allreduce.t0" in json.dumps(e) for e in lanes)

@@ -1,0 +1,397 @@
+"""CPU multi-process semantics of every collective op (SURVEY.md §4:
+"every public op asserted against a locally computed expectation, run with
+small world sizes"). Closed-form expectations: rank r contributes
+deterministic tensors, so every rank can verify the reduction locally.
+"""
+import pytest
+import torch
+
+from conftest import run_distributed
+
+
+# ------------------------------------------------------------- allreduce
+
+def _allreduce_avg(rank, size):
+    import axon as hvd
+
+    t = torch.full((17,), float(rank + 1))
+    out = hvd.allreduce(t, average=True)
+    expect = sum(range(1, size + 1)) / size
+    assert torch.allclose(out, torch.full((17,), expect))
+    assert not torch.allclose(t, out) or size == 1  # out-of-place
+
+
+def test_allreduce_average():
+    run_distributed(_allreduce_avg, 2)
+
+
+def test_allreduce_average_ws4():
+    run_distributed(_allreduce_avg, 4)
+
+
+def _allreduce_ops(rank, size):
+    import axon as hvd
+
+    base = torch.arange(8, dtype=torch.float32) + rank
+    assert torch.allclose(
+        hvd.allreduce(base, op=hvd.Sum),
+        torch.arange(8, dtype=torch.float32) * size + sum(range(size)))
+    assert torch.allclose(hvd.allreduce(base, op=hvd.Min),
+                          torch.arange(8, dtype=torch.float32))
+    assert torch.allclose(hvd.allreduce(base, op=hvd.Max),
+                          torch.arange(8, dtype=torch.float32) + size - 1)
+    prod = torch.ones(8)
+    for r in range(size):
+        prod *= torch.arange(8, dtype=torch.float32) + r
+    assert torch.allclose(hvd.allreduce(base, op=hvd.Product), prod)
+
+
+def test_allreduce_ops():
+    run_distributed(_allreduce_ops, 2)
+
+
+def _allreduce_dtypes(rank, size):
+    import axon as hvd
+
+    for dt, tol in [(torch.float32, 1e-6), (torch.float64, 1e-12),
+                    (torch.bfloat16, 1e-2), (torch.float16, 1e-3)]:
+        t = (torch.arange(33, dtype=torch.float32) * (rank + 1)).to(dt)
+        out = hvd.allreduce(t, op=hvd.Sum, name=f"dt.{dt}")
+        expect = torch.zeros(33, dtype=torch.float32)
+        for r in range(size):
+            expect += (torch.arange(33, dtype=torch.float32) * (r + 1)).to(dt).float()
+        assert out.dtype == dt
+        assert torch.allclose(out.float(), expect, rtol=tol, atol=tol), dt
+    t = torch.arange(10, dtype=torch.int64) + rank
+    out = hvd.allreduce(t, op=hvd.Sum, name="dt.int64")
+    assert torch.equal(out, torch.arange(10, dtype=torch.int64) * size + sum(range(size)))
+
+
+def test_allreduce_dtypes():
+    run_distributed(_allreduce_dtypes, 2)
+
+
+def _allreduce_inplace(rank, size):
+    import axon as hvd
+
+    t = torch.full((5,), float(rank))
+    ret = hvd.allreduce_(t, average=True)
+    assert ret.data_ptr() == t.data_ptr()
+    assert torch.allclose(t, torch.full((5,), sum(range(size)) / size))
+
+
+def test_allreduce_inplace():
+    run_distributed(_allreduce_inplace, 2)
+
+
+def _allreduce_scales(rank, size):
+    import axon as hvd
+
+    t = torch.full((9,), 2.0)
+    out = hvd.allreduce(t, op=hvd.Sum, prescale_factor=0.5, postscale_factor=3.0)
+    assert torch.allclose(out, torch.full((9,), 2.0 * 0.5 * size * 3.0))
+
+
+def test_allreduce_scales():
+    run_distributed(_allreduce_scales, 2)
+
+
+def _allreduce_fp16_wire(rank, size):
+    import axon as hvd
+
+    t = torch.randn(257, generator=torch.Generator().manual_seed(rank))
+    out = hvd.allreduce(t, op=hvd.Sum, compression=hvd.Compression.fp16)
+    expect = torch.zeros(257)
+    for r in range(size):
+        expect += torch.randn(257, generator=torch.Generator().manual_seed(r)).half().float()
+    assert out.dtype == torch.float32
+    assert torch.allclose(out, expect, rtol=2e-3, atol=2e-3)
+
+
+def test_allreduce_fp16_wire():
+    run_distributed(_allreduce_fp16_wire, 2)
+
+
+def _allreduce_many_fused(rank, size):
+    import axon as hvd
+
+    handles = []
+    for i in range(60):
+        t = torch.full((128 + i,), float(rank + i))
+        handles.append(hvd.allreduce_async(t, op=hvd.Sum, name=f"many.{i}"))
+    for i, h in enumerate(handles):
+        out = hvd.synchronize(h)
+        expect = float(sum(r + i for r in range(size)))
+        assert torch.allclose(out, torch.full((128 + i,), expect)), i
+
+
+def test_allreduce_many_fused():
+    run_distributed(_allreduce_many_fused, 2)
+
+
+def _allreduce_shuffled_order(rank, size):
+    # Controller ordering invariant: ranks enqueue in different orders but
+    # every collective still completes correctly (SURVEY.md §7 hard part 1).
+    import random
+
+    import axon as hvd
+
+    idx = list(range(40))
+    random.Random(rank * 77).shuffle(idx)
+    handles = {}
+    for i in idx:
+        t = torch.full((64,), float(rank * 100 + i))
+        handles[i] = hvd.allreduce_async(t, op=hvd.Sum, name=f"shuf.{i}")
+    for i, h in handles.items():
+        out = hvd.synchronize(h)
+        expect = float(sum(r * 100 + i for r in range(size)))
+        assert torch.allclose(out, torch.full((64,), expect)), i
+
+
+def test_allreduce_shuffled_order():
+    run_distributed(_allreduce_shuffled_order, 4)
+
+
+def _grouped(rank, size):
+    import axon as hvd
+
+    ts = [torch.full((11 * (i + 1),), float(rank + i)) for i in range(5)]
+    outs = hvd.grouped_allreduce(ts, op=hvd.Sum)
+    for i, out in enumerate(outs):
+        assert torch.allclose(out, torch.full((11 * (i + 1),),
+                                              float(sum(r + i for r in range(size)))))
+
+
+def test_grouped_allreduce():
+    run_distributed(_grouped, 2)
+
+
+# ------------------------------------------------------------- broadcast
+
+def _broadcast(rank, size):
+    import axon as hvd
+
+    for root in range(size):
+        t = torch.full((7,), float(rank))
+        out = hvd.broadcast(t, root_rank=root, name=f"bc.{root}")
+        assert torch.allclose(out, torch.full((7,), float(root)))
+        t2 = torch.full((7,), float(rank))
+        hvd.broadcast_(t2, root_rank=root, name=f"bci.{root}")
+        assert torch.allclose(t2, torch.full((7,), float(root)))
+
+
+def test_broadcast():
+    run_distributed(_broadcast, 2)
+
+
+# ------------------------------------------------------------- allgather
+
+def _allgather_equal(rank, size):
+    import axon as hvd
+
+    t = torch.full((3, 4), float(rank))
+    out = hvd.allgather(t)
+    assert out.shape == (3 * size, 4)
+    for r in range(size):
+        assert torch.allclose(out[3 * r:3 * (r + 1)], torch.full((3, 4), float(r)))
+
+
+def _allgather_variable(rank, size):
+    import axon as hvd
+
+    t = torch.full((rank + 1, 2), float(rank))
+    out = hvd.allgather(t)
+    assert out.shape == (sum(r + 1 for r in range(size)), 2)
+    off = 0
+    for r in range(size):
+        assert torch.allclose(out[off:off + r + 1], torch.full((r + 1, 2), float(r)))
+        off += r + 1
+
+
+def test_allgather():
+    run_distributed(_allgather_equal, 2)
+    run_distributed(_allgather_variable, 3)
+
+
+# -------------------------------------------------------------- alltoall
+
+def _alltoall(rank, size):
+    import axon as hvd
+
+    # rank r sends (d+1) rows of value r*10+d to rank d
+    splits = [d + 1 for d in range(size)]
+    rows = sum(splits)
+    t = torch.cat([torch.full((d + 1, 3), float(rank * 10 + d)) for d in range(size)])
+    assert t.shape[0] == rows
+    out, received = hvd.alltoall(t, splits=torch.tensor(splits))
+    # I receive (rank+1) rows from each rank, value r*10+rank
+    assert out.shape == ((rank + 1) * size, 3)
+    off = 0
+    for r in range(size):
+        assert torch.allclose(out[off:off + rank + 1],
+                              torch.full((rank + 1, 3), float(r * 10 + rank)))
+        off += rank + 1
+    assert received.tolist() == [rank + 1] * size
+
+
+def _alltoall_even(rank, size):
+    import axon as hvd
+
+    t = torch.arange(size * 2, dtype=torch.float32) + rank * 100
+    out = hvd.alltoall(t)
+    expect = torch.cat([torch.arange(rank * 2, rank * 2 + 2, dtype=torch.float32) + r * 100
+                        for r in range(size)])
+    assert torch.allclose(out, expect)
+
+
+def test_alltoall():
+    run_distributed(_alltoall, 3)
+    run_distributed(_alltoall_even, 2)
+
+
+# --------------------------------------------------------- reducescatter
+
+def _reducescatter(rank, size):
+    import axon as hvd
+
+    for rows in (size * 3, size * 3 + 1):  # even + uneven
+        t = torch.ones(rows, 2) * (rank + 1)
+        out = hvd.reducescatter(t, op=hvd.Sum, name=f"rs.{rows}")
+        base, rem = rows // size, rows % size
+        my_rows = base + (1 if rank < rem else 0)
+        assert out.shape == (my_rows, 2)
+        assert torch.allclose(out, torch.full((my_rows, 2),
+                                              float(sum(range(1, size + 1)))))
+        out2 = hvd.reducescatter(t, op=hvd.Average, name=f"rsa.{rows}")
+        assert torch.allclose(out2, torch.full((my_rows, 2),
+                                               sum(range(1, size + 1)) / size))
+
+
+def test_reducescatter():
+    run_distributed(_reducescatter, 2)
+    run_distributed(_reducescatter, 3)
+
+
+# ------------------------------------------------------------------ adasum
+
+def adasum_oracle(vectors):
+    """Reference recursive-halving adasum tree in fp64 (same tree as the
+    engine: level k combines v[i] with v[i+2^k])."""
+    v = [x.double() for x in vectors]
+    n = len(v)
+    stride = 1
+    while stride < n:
+        for i in range(0, n, 2 * stride):
+            a, b = v[i], v[i + stride]
+            dot = torch.dot(a, b).item()
+            na = torch.dot(a, a).item()
+            nb = torch.dot(b, b).item()
+            ca = 1.0 - dot / (2 * na) if na > 0 else 1.0
+            cb = 1.0 - dot / (2 * nb) if nb > 0 else 1.0
+            v[i] = ca * a + cb * b
+        stride *= 2
+    return v[0]
+
+
+def _adasum(rank, size):
+    import axon as hvd
+
+    t = torch.randn(97, generator=torch.Generator().manual_seed(rank * 3 + 1))
+    out = hvd.allreduce(t, op=hvd.Adasum)
+    vectors = [torch.randn(97, generator=torch.Generator().manual_seed(r * 3 + 1))
+               for r in range(size)]
+    expect = adasum_oracle(vectors).float()
+    assert torch.allclose(out, expect, rtol=1e-5, atol=1e-5)
+
+
+def test_adasum():
+    run_distributed(_adasum, 2)
+    run_distributed(_adasum, 4)
+
+
+# -------------------------------------------------------------- join etc.
+
+def _join(rank, size):
+    import axon as hvd
+
+    # rank 0 runs 3 steps, rank 1 only 2 then joins.
+    steps = 3 if rank == 0 else 2
+    for i in range(steps):
+        t = torch.full((6,), float(rank + 1))
+        out = hvd.allreduce(t, op=hvd.Sum, name=f"j.{i}")
+        if i < 2:
+            assert torch.allclose(out, torch.full((6,), 3.0))
+        else:
+            assert torch.allclose(out, torch.full((6,), 1.0))  # peer joined -> zeros
+    last = hvd.join()
+    assert last in range(size)
+
+
+def test_join():
+    run_distributed(_join, 2)
+
+
+def _barrier(rank, size):
+    import axon as hvd
+
+    for _ in range(3):
+        hvd.barrier()
+
+
+def test_barrier():
+    run_distributed(_barrier, 4)
+
+
+def _duplicate_name(rank, size):
+    import time
+
+    import axon as hvd
+
+    # rank 0 enqueues "dup" twice before rank 1 enqueues it once: the first
+    # stays pending (awaiting rank 1), so the second must get the
+    # duplicate-name error deterministically.
+    if rank == 0:
+        h1 = hvd.allreduce_async(torch.ones(4), op=hvd.Sum, name="dup")
+        h2 = hvd.allreduce_async(torch.ones(4), op=hvd.Sum, name="dup")
+        try:
+            hvd.synchronize(h2)
+            raise AssertionError("expected duplicate-name error")
+        except RuntimeError as e:
+            assert "already in flight" in str(e)
+        hvd.barrier()  # let rank 1 enqueue its "dup" afterwards
+        hvd.synchronize(h1)
+    else:
+        hvd.barrier()
+        time.sleep(0.05)
+        hvd.allreduce(torch.ones(4), op=hvd.Sum, name="dup")
+
+
+def test_duplicate_name_error():
+    run_distributed(_duplicate_name, 2)
+
+
+def _shape_mismatch(rank, size):
+    import axon as hvd
+
+    t = torch.ones(3 + rank)
+    try:
+        hvd.allreduce(t, op=hvd.Sum, name="mismatch")
+        raise AssertionError("expected shape-mismatch error")
+    except RuntimeError as e:
+        assert "mismatch" in str(e)
+
+
+def test_shape_mismatch_error():
+    run_distributed(_shape_mismatch, 2)
+
+
+def _scalar_tensor(rank, size):
+    import axon as hvd
+
+    t = torch.tensor(float(rank + 1))
+    out = hvd.allreduce(t, average=True)
+    assert out.item() == pytest.approx(sum(range(1, size + 1)) / size)
+
+
+def test_scalar_tensor():
+    run_distributed(_scalar_tensor, 2)
