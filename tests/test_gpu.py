@@ -1,0 +1,209 @@
+"""GPU (MI355X) tests — run via `pytest -m gpu` on a GPU box (gpurun).
+Single-GPU world: RCCL 1-rank communicator + the CDNA4 kernels. Kernel
+numerics are asserted against the plain fp32 Python oracles
+(tests/test_fused_optim_cpu.py formulas / torch casts).
+"""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
+                                  reason="needs a GPU")
+
+
+@pytest.fixture(scope="module")
+def hvd():
+    import axon as hvd
+
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    hvd.init()
+    yield hvd
+    hvd.shutdown()
+
+
+DEV = "cuda:0"
+
+
+# ------------------------------------------------- engine collectives ws=1
+
+def test_allreduce_direct(hvd):
+    t = torch.randn(1 << 20, device=DEV)
+    out = hvd.allreduce(t, average=True)
+    assert torch.allclose(out, t)
+
+
+def test_allreduce_fused_pack_unpack(hvd):
+    ts = [torch.randn(1000 + 37 * i, device=DEV) for i in range(30)]
+    outs = hvd.grouped_allreduce(ts, op=hvd.Sum, name="gpu_fused")
+    for t, o in zip(ts, outs):
+        assert torch.allclose(o, t, rtol=1e-6, atol=1e-6)
+
+
+def test_allreduce_scales_gpu(hvd):
+    t = torch.full((12345,), 2.0, device=DEV)
+    out = hvd.allreduce(t, op=hvd.Sum, prescale_factor=0.5,
+                        postscale_factor=3.0, name="gpu_scaled")
+    assert torch.allclose(out, torch.full_like(t, 3.0))
+
+
+def test_fp16_wire_kernels(hvd):
+    ts = [torch.randn(777 + i, device=DEV) * 10 for i in range(5)]
+    outs = hvd.grouped_allreduce(ts, op=hvd.Sum, name="gpu_fp16",
+                                 compression=hvd.Compression.fp16)
+    for t, o in zip(ts, outs):
+        ref = t.half().float()  # fp32 torch oracle of the wire cast
+        assert o.dtype == torch.float32
+        assert torch.allclose(o, ref, rtol=1e-3, atol=1e-3)
+
+
+def test_bf16_fused(hvd):
+    ts = [torch.randn(513 + i, device=DEV).bfloat16() for i in range(8)]
+    outs = hvd.grouped_allreduce(ts, op=hvd.Sum, name="gpu_bf16")
+    for t, o in zip(ts, outs):
+        assert o.dtype == torch.bfloat16
+        assert torch.allclose(o.float(), t.float())
+
+
+def test_broadcast_gpu(hvd):
+    t = torch.randn(4096, device=DEV)
+    out = hvd.broadcast(t, root_rank=0)
+    assert torch.allclose(out, t)
+
+
+def test_allgather_gpu(hvd):
+    t = torch.randn(33, 5, device=DEV)
+    out = hvd.allgather(t)
+    assert out.shape == (33, 5)
+    assert torch.allclose(out, t)
+
+
+def test_alltoall_gpu(hvd):
+    t = torch.randn(8, 3, device=DEV)
+    out = hvd.alltoall(t)
+    assert torch.allclose(out, t)
+
+
+def test_reducescatter_gpu(hvd):
+    t = torch.randn(10, 4, device=DEV)
+    out = hvd.reducescatter(t, op=hvd.Sum)
+    assert torch.allclose(out, t)  # ws=1: identity
+
+
+def test_adasum_gpu_ws1(hvd):
+    t = torch.randn(999, device=DEV)
+    out = hvd.allreduce(t, op=hvd.Adasum, name="gpu_adasum")
+    assert torch.allclose(out, t)
+
+
+def test_large_single_tensor(hvd):
+    # > fusion threshold: exercises the scratch/direct path
+    t = torch.randn(40 << 20, device=DEV)  # 160 MB fp32
+    out = hvd.allreduce(t, op=hvd.Sum, name="gpu_big")
+    assert torch.allclose(out, t)
+
+
+# ------------------------------------------------------ fused optimizers
+
+def _models(dtype, seed=0):
+    import torch.nn as nn
+
+    torch.manual_seed(seed)
+    gpu = nn.Sequential(nn.Linear(64, 128), nn.GELU(),
+                        nn.Linear(128, 32)).to(DEV, dtype)
+    cpu = nn.Sequential(nn.Linear(64, 128), nn.GELU(), nn.Linear(128, 32))
+    cpu.load_state_dict({k: v.float().cpu() for k, v in gpu.state_dict().items()})
+    return gpu, cpu
+
+
+def _run_pair(gpu_model, cpu_model, gpu_opt, cpu_opt, dtype, steps=5):
+    for s in range(steps):
+        torch.manual_seed(100 + s)
+        x = torch.randn(16, 64)
+        xg = x.to(DEV, dtype)
+        (gpu_model(xg).float() ** 2).mean().backward()
+        (cpu_model(x) ** 2).mean().backward()
+        gpu_opt.step()
+        cpu_opt.step()
+        gpu_opt.zero_grad()
+        cpu_opt.zero_grad()
+
+
+@pytest.mark.parametrize("dtype,tol", [(torch.float32, 1e-4),
+                                       (torch.bfloat16, 5e-2)])
+def test_fused_sgd_gpu(hvd, dtype, tol):
+    from axon.optim import FusedSGD
+
+    gpu_m, cpu_m = _models(dtype, seed=1)
+    go = FusedSGD(gpu_m.parameters(), lr=0.05, momentum=0.9, weight_decay=1e-3)
+    co = FusedSGD(cpu_m.parameters(), lr=0.05, momentum=0.9, weight_decay=1e-3)
+    _run_pair(gpu_m, cpu_m, go, co, dtype)
+    for p, q in zip(gpu_m.parameters(), cpu_m.parameters()):
+        assert torch.allclose(p.float().cpu(), q.detach(), rtol=tol, atol=tol), \
+            (p.float().cpu() - q).abs().max()
+
+
+@pytest.mark.parametrize("cls", ["adam", "adamw", "lamb"])
+def test_fused_adam_family_gpu(hvd, cls):
+    from axon.optim import FusedAdam, FusedAdamW, FusedLAMB
+
+    ctor = {"adam": FusedAdam, "adamw": FusedAdamW, "lamb": FusedLAMB}[cls]
+    gpu_m, cpu_m = _models(torch.float32, seed=2)
+    go = ctor(gpu_m.parameters(), lr=2e-3, weight_decay=1e-2)
+    co = ctor(cpu_m.parameters(), lr=2e-3, weight_decay=1e-2)
+    _run_pair(gpu_m, cpu_m, go, co, torch.float32)
+    for p, q in zip(gpu_m.parameters(), cpu_m.parameters()):
+        assert torch.allclose(p.float().cpu(), q.detach(), rtol=1e-3, atol=1e-4), \
+            (p.float().cpu() - q).abs().max()
+
+
+def test_fused_sgd_matches_torch_gpu(hvd):
+    """Fused HIP SGD vs stock torch SGD on the SAME GPU data."""
+    import torch.nn as nn
+
+    from axon.optim import FusedSGD
+
+    torch.manual_seed(7)
+    m1 = nn.Linear(256, 256).to(DEV)
+    m2 = nn.Linear(256, 256).to(DEV)
+    m2.load_state_dict(m1.state_dict())
+    o1 = FusedSGD(m1.parameters(), lr=0.1, momentum=0.9)
+    o2 = torch.optim.SGD(m2.parameters(), lr=0.1, momentum=0.9)
+    for s in range(4):
+        torch.manual_seed(s)
+        x = torch.randn(32, 256, device=DEV)
+        (m1(x) ** 2).mean().backward()
+        (m2(x) ** 2).mean().backward()
+        o1.step()
+        o2.step()
+        o1.zero_grad()
+        o2.zero_grad()
+    for p, q in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p, q, rtol=1e-5, atol=1e-6)
+
+
+# ----------------------------------------------------------- end to end
+
+def test_resnet50_bf16_step(hvd):
+    import torch.nn.functional as F
+
+    from axon.models.resnet import resnet50
+    from axon.optim import FusedSGD
+
+    torch.manual_seed(0)
+    model = resnet50().to(DEV, torch.bfloat16)
+    opt = FusedSGD(model.parameters(), lr=0.1, momentum=0.9)
+    opt = hvd.DistributedOptimizer(opt, named_parameters=model.named_parameters())
+    hvd.broadcast_parameters(model.state_dict(), root_rank=0)
+    x = torch.randn(8, 3, 224, 224, device=DEV, dtype=torch.bfloat16)
+    y = torch.randint(0, 1000, (8,), device=DEV)
+    losses = []
+    for _ in range(3):
+        opt.zero_grad()
+        loss = F.cross_entropy(model(x).float(), y)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert losses[-1] < losses[0]  # overfits one batch fast
