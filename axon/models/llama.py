@@ -1,0 +1,150 @@
+"""Llama-3-style decoder for the pure-DP large-model config
+(BASELINE.json config 4: Llama-3 8B pure data-parallel bf16 — MI355X's
+288 GB HBM holds params + grads + Adam states per GPU with no sharding,
+SURVEY.md §5.7).
+
+Own implementation of the published architecture (Grattafiori et al.,
+The Llama 3 Herd of Models): RMSNorm, rotary embeddings (theta 500000),
+grouped-query attention (8 KV heads), SwiGLU FFN, tied-off LM head.
+"""
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+@dataclass
+class LlamaConfig:
+    vocab_size: int = 128256
+    dim: int = 4096
+    layers: int = 32
+    heads: int = 32
+    kv_heads: int = 8
+    ffn: int = 14336
+    max_seq: int = 8192
+    rope_theta: float = 500000.0
+    norm_eps: float = 1e-5
+
+
+def llama3_8b_config():
+    return LlamaConfig()
+
+
+def llama3_1b_config():
+    """Roughly Llama-3.2-1B shaped (smaller bench/dev variant)."""
+    return LlamaConfig(dim=2048, layers=16, heads=32, kv_heads=8, ffn=8192)
+
+
+def llama_tiny_config():
+    """For tests: same code path, tiny shapes."""
+    return LlamaConfig(vocab_size=256, dim=64, layers=2, heads=4, kv_heads=2,
+                       ffn=128, max_seq=128)
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, dim, eps):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.eps = eps
+
+    def forward(self, x):
+        dt = x.dtype
+        x = x.float()
+        x = x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + self.eps)
+        return (x * self.weight.float()).to(dt)
+
+
+def _rope_cache(dim, seq, theta, device, dtype):
+    inv = 1.0 / (theta ** (torch.arange(0, dim, 2, device=device).float() / dim))
+    t = torch.arange(seq, device=device).float()
+    freqs = torch.outer(t, inv)
+    return torch.cos(freqs).to(dtype), torch.sin(freqs).to(dtype)
+
+
+def _apply_rope(x, cos, sin):
+    # x: (b, h, s, d); cos/sin: (s, d/2)
+    x1, x2 = x[..., 0::2], x[..., 1::2]
+    c = cos.view(1, 1, cos.shape[0], -1)
+    s = sin.view(1, 1, sin.shape[0], -1)
+    out = torch.empty_like(x)
+    out[..., 0::2] = x1 * c - x2 * s
+    out[..., 1::2] = x2 * c + x1 * s
+    return out
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, cfg):
+        super().__init__()
+        self.heads = cfg.heads
+        self.kv_heads = cfg.kv_heads
+        self.head_dim = cfg.dim // cfg.heads
+        self.wq = nn.Linear(cfg.dim, cfg.dim, bias=False)
+        self.wk = nn.Linear(cfg.dim, cfg.kv_heads * self.head_dim, bias=False)
+        self.wv = nn.Linear(cfg.dim, cfg.kv_heads * self.head_dim, bias=False)
+        self.wo = nn.Linear(cfg.dim, cfg.dim, bias=False)
+
+    def forward(self, x, cos, sin):
+        b, s, _ = x.shape
+        q = self.wq(x).view(b, s, self.heads, self.head_dim).transpose(1, 2)
+        k = self.wk(x).view(b, s, self.kv_heads, self.head_dim).transpose(1, 2)
+        v = self.wv(x).view(b, s, self.kv_heads, self.head_dim).transpose(1, 2)
+        q = _apply_rope(q, cos, sin)
+        k = _apply_rope(k, cos, sin)
+        o = F.scaled_dot_product_attention(q, k, v, is_causal=True,
+                                           enable_gqa=True)
+        return self.wo(o.transpose(1, 2).reshape(b, s, -1))
+
+
+class LlamaBlock(nn.Module):
+    def __init__(self, cfg):
+        super().__init__()
+        self.attn_norm = RMSNorm(cfg.dim, cfg.norm_eps)
+        self.attn = LlamaAttention(cfg)
+        self.ffn_norm = RMSNorm(cfg.dim, cfg.norm_eps)
+        self.w1 = nn.Linear(cfg.dim, cfg.ffn, bias=False)  # gate
+        self.w3 = nn.Linear(cfg.dim, cfg.ffn, bias=False)  # up
+        self.w2 = nn.Linear(cfg.ffn, cfg.dim, bias=False)  # down
+
+    def forward(self, x, cos, sin):
+        x = x + self.attn(self.attn_norm(x), cos, sin)
+        h = self.ffn_norm(x)
+        return x + self.w2(F.silu(self.w1(h)) * self.w3(h))
+
+
+class LlamaForCausalLM(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.embed = nn.Embedding(cfg.vocab_size, cfg.dim)
+        self.blocks = nn.ModuleList(LlamaBlock(cfg) for _ in range(cfg.layers))
+        self.norm = RMSNorm(cfg.dim, cfg.norm_eps)
+        self.lm_head = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
+        self.apply(self._init)
+        self._rope = {}
+
+    def _init(self, m):
+        if isinstance(m, (nn.Linear, nn.Embedding)):
+            nn.init.normal_(m.weight, std=0.02)
+
+    def _rope_for(self, s, device, dtype):
+        key = (s, device, dtype)
+        if key not in self._rope:
+            self._rope[key] = _rope_cache(self.cfg.dim // self.cfg.heads, s,
+                                          self.cfg.rope_theta, device, dtype)
+        return self._rope[key]
+
+    def forward(self, input_ids):
+        b, s = input_ids.shape
+        x = self.embed(input_ids)
+        cos, sin = self._rope_for(s, x.device, x.dtype)
+        for blk in self.blocks:
+            x = blk(x, cos, sin)
+        return self.lm_head(self.norm(x))
+
+    def lm_loss(self, input_ids):
+        """Next-token loss on the sequence itself (synthetic pretrain)."""
+        logits = self.forward(input_ids)
+        return F.cross_entropy(
+            logits[:, :-1].float().reshape(-1, self.cfg.vocab_size),
+            input_ids[:, 1:].reshape(-1))
