@@ -96,8 +96,10 @@ Controller::Controller(int rank, int size, ControlPlane* cp)
 }
 
 std::vector<Response> Controller::run_cycle(const std::vector<Request>& own,
-                                            bool* shutdown) {
+                                            bool* shutdown,
+                                            double* new_cycle_ms) {
   *shutdown = false;
+  *new_cycle_ms = 0.0;
   if (rank_ != 0) {
     // Worker: ship requests, collect any pending response lists.
     if (!own.empty()) {
@@ -116,6 +118,7 @@ std::vector<Response> Controller::run_cycle(const std::vector<Request>& own,
       if (t == MsgType::RESPONSE_LIST) {
         ResponseList rl = ResponseList::deserialize(payload);
         if (rl.shutdown) *shutdown = true;
+        if (rl.new_cycle_ms > 0) *new_cycle_ms = rl.new_cycle_ms;
         for (auto& r : rl.responses) out.push_back(std::move(r));
       }
     }
@@ -138,14 +141,17 @@ std::vector<Response> Controller::run_cycle(const std::vector<Request>& own,
   std::vector<Response> rs = assemble();
   check_stalls();
 
-  if (!rs.empty() || pending_shutdown_) {
+  if (!rs.empty() || pending_shutdown_ || pending_cycle_ms_ > 0) {
     ResponseList rl;
     rl.responses = rs;
     rl.shutdown = pending_shutdown_;
+    rl.new_cycle_ms = pending_cycle_ms_;
     auto buf = rl.serialize();
     for (int peer = 1; peer < size_; ++peer)
       cp_->send(peer, MsgType::RESPONSE_LIST, buf);
     if (pending_shutdown_) *shutdown = true;
+    if (pending_cycle_ms_ > 0) *new_cycle_ms = pending_cycle_ms_;
+    pending_cycle_ms_ = 0.0;
   }
   return rs;
 }

@@ -54,8 +54,16 @@ class Controller {
 
   // One coordination step. `own` = requests newly generated on this rank
   // this cycle. Returns responses to execute, in the globally agreed order.
-  // Sets *shutdown when the coordinator announced shutdown.
-  std::vector<Response> run_cycle(const std::vector<Request>& own, bool* shutdown);
+  // Sets *shutdown when the coordinator announced shutdown; sets
+  // *new_cycle_ms > 0 when the autotuner shipped a cycle-time update.
+  std::vector<Response> run_cycle(const std::vector<Request>& own, bool* shutdown,
+                                  double* new_cycle_ms);
+
+  // Autotuner hooks (rank 0): change the fusion threshold locally and
+  // attach a cycle-time update to the next broadcast.
+  void set_fusion_threshold(int64_t t) { fusion_threshold_ = t; }
+  int64_t fusion_threshold() const { return fusion_threshold_; }
+  void announce_cycle_time(double ms) { pending_cycle_ms_ = ms; }
 
   // Coordinator only: announce shutdown to all workers (drained on their
   // next cycle).
@@ -92,6 +100,7 @@ class Controller {
   int64_t fusion_threshold_;
   double stall_warn_sec_;
   bool pending_shutdown_ = false;
+  double pending_cycle_ms_ = 0.0;
 };
 
 }  // namespace axon

@@ -5,7 +5,16 @@
 namespace axon {
 
 namespace {
+// Storage-flat alias: collectives are elementwise over dense storage, so a
+// channels-last (or otherwise permuted-but-dense) tensor participates
+// without copies — every rank has identical strides by construction.
 at::Tensor flat(const at::Tensor& t) {
+  if (t.is_contiguous()) return t.view({t.numel()});
+  if (t.is_non_overlapping_and_dense()) {
+    at::Tensor alias = at::empty({0}, t.options());
+    alias.set_(t.storage(), t.storage_offset(), {t.numel()}, {1});
+    return alias;
+  }
   return t.contiguous().view({t.numel()});
 }
 }  // namespace
@@ -89,7 +98,7 @@ Status CpuBackend::allreduce(const Response& r, std::vector<TensorEntry>& entrie
       send_tensor(0, t);
       result = recv_tensor(0, wire, t.numel());
     }
-    at::Tensor out = e.out.view({e.out.numel()});
+    at::Tensor out = flat(e.out);
     out.copy_(result);  // casts wire -> dtype if compressed
     if (e.postscale != 1.0) out.mul_(e.postscale);
   }
@@ -132,7 +141,7 @@ Status CpuBackend::adasum(const Response& r, std::vector<TensorEntry>& entries) 
       send_tensor(0, t.contiguous());
       result = recv_tensor(0, dtype, t.numel());
     }
-    at::Tensor out = e.out.view({e.out.numel()});
+    at::Tensor out = flat(e.out);
     out.copy_(result);
     if (e.postscale != 1.0) out.mul_(e.postscale);
   }
@@ -155,7 +164,7 @@ Status CpuBackend::broadcast(const Response& r, std::vector<TensorEntry>& entrie
       buf = recv_tensor(0, id_dtype(r.dtype), e.out.numel());
     }
     if (rank_ != root || !e.out.is_same(e.in))
-      e.out.view({e.out.numel()}).copy_(buf);
+      flat(e.out).copy_(buf);
   }
   return Status::OK();
 }
@@ -190,7 +199,7 @@ Status CpuBackend::allgather(const Response& r, std::vector<TensorEntry>& entrie
     auto out_sizes = in_sizes;
     if (out_sizes.empty()) out_sizes.push_back(total_rows); else out_sizes[0] = total_rows;
     e.out.resize_(out_sizes);
-    e.out.view({e.out.numel()}).copy_(full);
+    flat(e.out).copy_(full);
   }
   return Status::OK();
 }
@@ -240,7 +249,7 @@ Status CpuBackend::alltoall(const Response& r, std::vector<TensorEntry>& entries
     int64_t rows_me = mine.numel() / std::max<int64_t>(row, 1);
     if (out_sizes.empty()) out_sizes.push_back(rows_me); else out_sizes[0] = rows_me;
     e.out.resize_(out_sizes);
-    e.out.view({e.out.numel()}).copy_(mine);
+    flat(e.out).copy_(mine);
     if (e.received_splits.defined()) {
       auto rs = e.received_splits;
       rs.resize_({size_});
@@ -283,7 +292,7 @@ Status CpuBackend::reducescatter(const Response& r, std::vector<TensorEntry>& en
     if (out_sizes.empty()) out_sizes.push_back(rank_rows(rank_));
     else out_sizes[0] = rank_rows(rank_);
     e.out.resize_(out_sizes);
-    e.out.view({e.out.numel()}).copy_(mine);
+    flat(e.out).copy_(mine);
     if (e.postscale != 1.0) e.out.mul_(e.postscale);
   }
   return Status::OK();

@@ -84,6 +84,7 @@ Response Response::deserialize(Reader& r) {
 std::vector<uint8_t> ResponseList::serialize() const {
   Writer w;
   w.u8(shutdown ? 1 : 0);
+  w.f64(new_cycle_ms);
   w.u32(static_cast<uint32_t>(responses.size()));
   for (auto& r : responses) r.serialize(w);
   return std::move(w.buf);
@@ -93,6 +94,7 @@ ResponseList ResponseList::deserialize(const std::vector<uint8_t>& buf) {
   Reader r(buf);
   ResponseList l;
   l.shutdown = r.u8() != 0;
+  l.new_cycle_ms = r.f64();
   uint32_t n = r.u32();
   l.responses.resize(n);
   for (uint32_t i = 0; i < n; ++i) l.responses[i] = Response::deserialize(r);

@@ -50,6 +50,8 @@ struct Response {
 struct ResponseList {
   std::vector<Response> responses;
   bool shutdown = false;
+  // Autotuner cycle-time update (ms); 0 = unchanged (parameter_manager.h).
+  double new_cycle_ms = 0.0;
 
   std::vector<uint8_t> serialize() const;
   static ResponseList deserialize(const std::vector<uint8_t>& buf);

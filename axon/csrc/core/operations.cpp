@@ -10,6 +10,7 @@
 #include "controller.h"
 #include "cpu_backend.h"
 #include "message.h"
+#include "parameter_manager.h"
 #include "timeline.h"
 
 namespace axon {
@@ -36,6 +37,7 @@ struct Global {
   std::unique_ptr<Controller> ctl;
   std::unique_ptr<CpuBackend> cpu;
   std::unique_ptr<GpuBackend> gpu;
+  std::unique_ptr<ParameterManager> pm;
   std::unordered_map<std::string, TensorEntry> pending;
   bool joined = false;
   TensorEntry join_entry;
@@ -170,6 +172,9 @@ void bg_loop(std::promise<void> ready) {
       g.cp->init_worker(g.control_path, g.rank, timeout);
     g.ctl = std::make_unique<Controller>(g.rank, g.size, g.cp.get());
     g.cpu = std::make_unique<CpuBackend>(g.rank, g.size, g.cp.get());
+    if (g.rank == 0)
+      g.pm = std::make_unique<ParameterManager>(g.ctl->fusion_threshold(),
+                                                g.cycle_ms);
     g.tl.init(g.rank);
     ready.set_value();
   } catch (...) {
@@ -240,8 +245,33 @@ void bg_loop(std::promise<void> ready) {
         break;
       }
 
-      auto responses = g.ctl->run_cycle(reqs, &shutdown);
-      for (auto& r : responses) execute_response(r);
+      // Autotuner (rank 0): score by GPU-allreduced bytes/sec, ship knob
+      // updates with the next broadcast (SURVEY.md §2.1 parameter manager).
+      if (g.pm && g.pm->enabled()) {
+        int64_t thr = g.ctl->fusion_threshold();
+        double cyc = g.cycle_ms;
+        if (g.pm->maybe_update(&thr, &cyc)) {
+          g.ctl->set_fusion_threshold(thr);
+          g.ctl->announce_cycle_time(cyc);
+        }
+      }
+
+      double new_cycle_ms = 0.0;
+      auto responses = g.ctl->run_cycle(reqs, &shutdown, &new_cycle_ms);
+      if (new_cycle_ms > 0) {
+        g.cycle_ms = new_cycle_ms;
+        cycle = std::chrono::duration<double, std::milli>(g.cycle_ms);
+      }
+      for (auto& r : responses) {
+        if (g.pm && r.op == OpType::ALLREDUCE && r.device >= 0 &&
+            r.error.empty()) {
+          int64_t bytes = 0;
+          size_t esz = dtype_size(id_dtype(r.wire_dtype ? r.wire_dtype : r.dtype));
+          for (auto nn : r.numels) bytes += nn * static_cast<int64_t>(esz);
+          g.pm->record_bytes(bytes);
+        }
+        execute_response(r);
+      }
       g.tl.mark_cycle();
     }
   } catch (const std::exception& ex) {
@@ -293,6 +323,7 @@ void shutdown() {
   g.initialized = false;
   g.gpu.reset();
   g.cpu.reset();
+  g.pm.reset();
   g.ctl.reset();
   g.cp.reset();
 }

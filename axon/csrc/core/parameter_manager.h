@@ -1,0 +1,52 @@
+// Online autotuner for the two engine knobs: fusion threshold (bytes) and
+// cycle time (ms) — capability parity with the reference's ParameterManager
+// (SURVEY.md §2.1 "Parameter manager (autotuner)"), re-designed as a
+// coordinator-driven grid hill-climb scored by allreduced bytes/sec:
+// xGMI-era defaults are already large (128 MiB), so the tuner explores a
+// coarse 2-D neighborhood instead of running full Bayesian optimization.
+//
+// Enable with AXON_AUTOTUNE=1; AXON_AUTOTUNE_LOG=1 logs every sample.
+// Runs on rank 0 only: the fusion threshold is a coordinator decision, and
+// cycle-time updates ship to workers inside the ResponseList.
+#pragma once
+
+#include <chrono>
+#include <cstdint>
+#include <vector>
+
+namespace axon {
+
+class ParameterManager {
+ public:
+  ParameterManager(int64_t threshold, double cycle_ms);
+
+  bool enabled() const { return enabled_; }
+  // Account bytes moved by GPU allreduce responses this cycle.
+  void record_bytes(int64_t bytes);
+  // Called once per cycle on the coordinator. Returns true when the knobs
+  // changed; outputs the new values (to apply locally + broadcast).
+  bool maybe_update(int64_t* threshold, double* cycle_ms);
+
+ private:
+  struct Sample {
+    int64_t threshold;
+    double cycle_ms;
+    double score = -1.0;
+  };
+  void advance();
+
+  bool enabled_;
+  bool log_;
+  bool done_ = false;
+  int64_t bytes_ = 0;
+  std::chrono::steady_clock::time_point window_start_;
+  double warmup_until_ = 3.0;  // seconds before sampling starts
+  double sample_sec_;
+  std::chrono::steady_clock::time_point t0_;
+  std::vector<Sample> plan_;
+  size_t idx_ = 0;
+  Sample best_;
+  bool started_ = false;
+};
+
+}  // namespace axon

@@ -58,6 +58,19 @@ def _wire(compression):
     return {torch.float16: 5, torch.bfloat16: 15, torch.float32: 6}[wd]
 
 
+def _dense(t):
+    """Dense (non-overlapping) tensors of ANY memory format (e.g.
+    channels_last) participate in elementwise collectives directly: every
+    rank constructs identical strides, so storage order agrees."""
+    if t.is_contiguous():
+        return True
+    try:
+        return torch.ops.aten.is_non_overlapping_and_dense(t)
+    except Exception:
+        return t.is_contiguous(memory_format=torch.channels_last) or \
+            t.is_contiguous(memory_format=torch.channels_last_3d)
+
+
 def _resolve_op(op, average):
     if op is not None and average is not None:
         raise ValueError("specify either op or average, not both")
@@ -75,7 +88,7 @@ def allreduce_async(tensor, average=None, name=None, op=None,
                     compression=None):
     """Out-of-place async allreduce; returns a handle."""
     red = _resolve_op(op, average)
-    t = tensor.contiguous()
+    t = tensor if _dense(tensor) else tensor.contiguous()
     output = torch.empty_like(t)
     core = _ensure_init()
     h = core.allreduce_async(t, output, _auto_name("allreduce", name), red,
@@ -95,8 +108,8 @@ def allreduce_async_(tensor, average=None, name=None, op=None,
                      prescale_factor=1.0, postscale_factor=1.0,
                      compression=None):
     """In-place async allreduce (the hot path for gradients)."""
-    if not tensor.is_contiguous():
-        raise ValueError("allreduce_ requires a contiguous tensor")
+    if not _dense(tensor):
+        raise ValueError("allreduce_ requires a dense tensor")
     red = _resolve_op(op, average)
     core = _ensure_init()
     h = core.allreduce_async(tensor, tensor, _auto_name("allreduce", name), red,
@@ -144,7 +157,7 @@ def grouped_allreduce_(tensors, **kw):
 # --------------------------------------------------------------- broadcast
 
 def broadcast_async(tensor, root_rank, name=None):
-    t = tensor.contiguous()
+    t = tensor if _dense(tensor) else tensor.contiguous()
     output = torch.empty_like(t)
     core = _ensure_init()
     h = core.broadcast_async(t, output, root_rank, _auto_name("broadcast", name))
@@ -156,8 +169,8 @@ def broadcast(tensor, root_rank, name=None):
 
 
 def broadcast_async_(tensor, root_rank, name=None):
-    if not tensor.is_contiguous():
-        raise ValueError("broadcast_ requires a contiguous tensor")
+    if not _dense(tensor):
+        raise ValueError("broadcast_ requires a dense tensor")
     core = _ensure_init()
     h = core.broadcast_async(tensor, tensor, root_rank,
                              _auto_name("broadcast", name))

@@ -1,12 +1,18 @@
-"""Flagship benchmark: ResNet-50, synthetic ImageNet-shaped data, bf16,
-data-parallel over N MI355X GPUs (BASELINE.json metric: images/sec
-ResNet-50 synthetic at 1/2/4/8 GPUs).
+"""Benchmark entry point (driver contract).
 
-Contract (driver): `python bench.py --gpus N --steps K --warmup W`;
-for N>1 launched via torch.distributed.run with one rank per GPU. W
-untimed warmup steps, then exactly K timed steps bracketed by barrier +
+Flagship (default, BASELINE.json metric): ResNet-50, synthetic
+ImageNet-shaped data, bf16, data-parallel over N MI355X GPUs —
+whole-job images/sec.
+
+`python bench.py --gpus N --steps K --warmup W`; for N>1 the driver
+launches one rank per GPU via torch.distributed.run. W untimed warmup
+steps, then exactly K timed steps bracketed by barrier +
 torch.cuda.synchronize on both sides; elapsed = MAX over ranks; rank 0
-prints ONE JSON line with the whole-job aggregate images/sec.
+prints ONE JSON line with the whole-job aggregate.
+
+Secondary configs (BASELINE.json configs 3-5) are reachable with
+--model bert-large (LAMB + fp16 compression), --model llama3-8b
+(pure-DP bf16 + fused AdamW), --reduction adasum (ResNet-50 Adasum).
 """
 import argparse
 import json
@@ -24,14 +30,109 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--batch", type=int, default=256, help="per-GPU batch size")
+    p.add_argument("--batch", type=int, default=None, help="per-GPU batch size")
     p.add_argument("--image-size", type=int, default=224)
-    p.add_argument("--model", default="resnet50")
-    p.add_argument("--fused-optimizer", default=None,
-                   help="'sgd' to force the fused HIP SGD; default: auto "
-                        "(fused on GPU, torch SGD on CPU)")
-    p.add_argument("--compression", default="none", choices=["none", "fp16"])
+    p.add_argument("--seq-len", type=int, default=None)
+    p.add_argument("--model", default="resnet50",
+                   choices=["resnet50", "bert-large", "llama3-8b", "llama3-1b"])
+    p.add_argument("--reduction", default="average", choices=["average", "adasum"])
+    p.add_argument("--compression", default=None, choices=[None, "none", "fp16"])
+    p.add_argument("--no-fused-optimizer", action="store_true")
     return p.parse_args()
+
+
+def build_resnet(args, device, dtype):
+    from axon.models.resnet import resnet50
+    from axon.optim import FusedSGD
+
+    torch.manual_seed(1234)
+    model = resnet50().to(device=device, dtype=dtype)
+    if device.type == "cuda":
+        # NHWC: MIOpen's MFMA conv kernels want channels-last on CDNA4.
+        model = model.to(memory_format=torch.channels_last)
+    batch = args.batch or 256
+    if args.no_fused_optimizer or device.type == "cpu":
+        opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
+                              weight_decay=1e-4)
+    else:
+        opt = FusedSGD(model.parameters(), lr=0.1, momentum=0.9,
+                       weight_decay=1e-4)
+    g = torch.Generator().manual_seed(hvd.rank())
+    x = torch.randn(batch, 3, args.image_size, args.image_size,
+                    generator=g).to(device=device, dtype=dtype)
+    if device.type == "cuda":
+        x = x.to(memory_format=torch.channels_last)
+    y = torch.randint(0, 1000, (batch,), generator=g).to(device)
+
+    def step():
+        opt.zero_grad(set_to_none=False)
+        loss = F.cross_entropy(model(x).float(), y)
+        loss.backward()
+        opt.step()
+        return loss
+
+    cfg = {"model": "resnet50", "global_batch": batch * hvd.size(),
+           "image_size": args.image_size}
+    return model, opt, step, batch, "images/sec", cfg
+
+
+def build_bert(args, device, dtype):
+    from axon.models.bert import BertForPreTraining, bert_large_config
+    from axon.optim import FusedLAMB
+
+    torch.manual_seed(1234)
+    model = BertForPreTraining(bert_large_config()).to(device=device, dtype=dtype)
+    batch = args.batch or 64
+    seq = args.seq_len or 128
+    if args.no_fused_optimizer or device.type == "cpu":
+        opt = torch.optim.AdamW(model.parameters(), lr=1e-4, weight_decay=0.01)
+    else:
+        opt = FusedLAMB(model.parameters(), lr=1e-3, weight_decay=0.01)
+    g = torch.Generator().manual_seed(hvd.rank())
+    ids = torch.randint(0, 30522, (batch, seq), generator=g).to(device)
+    labels = ids.clone()
+    mask = torch.rand(batch, seq, generator=g) > 0.15
+    labels[mask.to(device)] = -100
+
+    def step():
+        opt.zero_grad(set_to_none=False)
+        loss = model.mlm_loss(ids, labels)
+        loss.backward()
+        opt.step()
+        return loss
+
+    cfg = {"model": "bert-large", "global_batch": batch * hvd.size(),
+           "seq_len": seq, "optimizer": "fused_lamb"}
+    return model, opt, step, batch * seq, "tokens/sec", cfg
+
+
+def build_llama(args, device, dtype, which="llama3-8b"):
+    from axon.models.llama import (LlamaForCausalLM, llama3_1b_config,
+                                   llama3_8b_config)
+    from axon.optim import FusedAdamW
+
+    torch.manual_seed(1234)
+    cfgm = llama3_8b_config() if which == "llama3-8b" else llama3_1b_config()
+    model = LlamaForCausalLM(cfgm).to(device=device, dtype=dtype)
+    batch = args.batch or (2 if which == "llama3-8b" else 8)
+    seq = args.seq_len or 4096
+    if args.no_fused_optimizer or device.type == "cpu":
+        opt = torch.optim.AdamW(model.parameters(), lr=3e-4)
+    else:
+        opt = FusedAdamW(model.parameters(), lr=3e-4)
+    g = torch.Generator().manual_seed(hvd.rank())
+    ids = torch.randint(0, cfgm.vocab_size, (batch, seq), generator=g).to(device)
+
+    def step():
+        opt.zero_grad(set_to_none=False)
+        loss = model.lm_loss(ids)
+        loss.backward()
+        opt.step()
+        return loss
+
+    cfg = {"model": which, "global_batch": batch * hvd.size(), "seq_len": seq,
+           "optimizer": "fused_adamw"}
+    return model, opt, step, batch * seq, "tokens/sec", cfg
 
 
 def main():
@@ -43,43 +144,24 @@ def main():
     device = torch.device("cuda" if use_gpu else "cpu")
     dtype = torch.bfloat16 if use_gpu else torch.float32
 
-    from axon.models.resnet import resnet50
+    if args.model == "resnet50":
+        model, opt, step, units_per_step, metric, cfg = build_resnet(args, device, dtype)
+        default_comp = "none"
+    elif args.model == "bert-large":
+        model, opt, step, units_per_step, metric, cfg = build_bert(args, device, dtype)
+        default_comp = "fp16"
+    else:
+        model, opt, step, units_per_step, metric, cfg = build_llama(
+            args, device, dtype, args.model)
+        default_comp = "none"
 
-    torch.manual_seed(1234)
-    model = resnet50().to(device=device, dtype=dtype)
-
-    lr = 0.1
-    fused = args.fused_optimizer == "sgd" or (args.fused_optimizer is None and use_gpu)
-    if fused:
-        try:
-            from axon.optim import FusedSGD
-
-            opt = FusedSGD(model.parameters(), lr=lr, momentum=0.9,
-                           weight_decay=1e-4)
-        except (ImportError, RuntimeError):
-            fused = False
-    if not fused:
-        opt = torch.optim.SGD(model.parameters(), lr=lr, momentum=0.9,
-                              weight_decay=1e-4)
-    compression = hvd.Compression.fp16 if args.compression == "fp16" \
+    comp_name = args.compression or default_comp
+    compression = hvd.Compression.fp16 if comp_name == "fp16" \
         else hvd.Compression.none
+    red = hvd.Adasum if args.reduction == "adasum" else hvd.Average
     opt = hvd.DistributedOptimizer(opt, named_parameters=model.named_parameters(),
-                                   compression=compression)
+                                   compression=compression, op=red)
     hvd.broadcast_parameters(model.state_dict(), root_rank=0)
-
-    # Synthetic data, fixed per-GPU batch (weak scaling).
-    g = torch.Generator().manual_seed(rank)
-    x = torch.randn(args.batch, 3, args.image_size, args.image_size,
-                    generator=g).to(device=device, dtype=dtype)
-    y = torch.randint(0, 1000, (args.batch,), generator=g).to(device)
-
-    def step():
-        opt.zero_grad(set_to_none=False)
-        out = model(x)
-        loss = F.cross_entropy(out.float(), y)
-        loss.backward()
-        opt.step()
-        return loss
 
     for _ in range(args.warmup):
         step()
@@ -96,17 +178,18 @@ def main():
         torch.cuda.synchronize()
     hvd.barrier(name="bench.stop")
     elapsed = time.perf_counter() - t0
-    # MAX over ranks is the honest job time.
     elapsed = hvd.allreduce(torch.tensor([elapsed], dtype=torch.float64),
                             op=hvd.Max, name="bench.elapsed").item()
 
     if rank == 0:
-        total_images = args.batch * n * args.steps
-        value = total_images / elapsed
+        value = units_per_step * n * args.steps / elapsed
+        cfg["parallelism"] = f"dp{n}"
+        cfg["reduction"] = args.reduction
+        cfg["compression"] = comp_name
         print(json.dumps({
-            "metric": "images/sec",
+            "metric": metric,
             "value": round(value, 2),
-            "unit": "images/sec",
+            "unit": metric,
             "n_gpus": n,
             "steps": args.steps,
             "warmup": args.warmup,
@@ -116,14 +199,7 @@ def main():
             "vs_baseline": None,
             "dtype": "bf16" if use_gpu else "fp32",
             "data": "synthetic",
-            "config": {
-                "model": args.model,
-                "global_batch": args.batch * n,
-                "image_size": args.image_size,
-                "parallelism": f"dp{n}",
-                "optimizer": "fused_sgd" if fused else "torch_sgd",
-                "compression": args.compression,
-            },
+            "config": cfg,
         }), flush=True)
     hvd.shutdown()
 

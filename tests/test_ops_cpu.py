@@ -395,3 +395,21 @@ def _scalar_tensor(rank, size):
 
 def test_scalar_tensor():
     run_distributed(_scalar_tensor, 2)
+
+
+def _channels_last(rank, size):
+    import axon as hvd
+
+    t = (torch.arange(2 * 4 * 3 * 3, dtype=torch.float32).reshape(2, 4, 3, 3)
+         * (rank + 1)).to(memory_format=torch.channels_last)
+    out = hvd.allreduce(t, op=hvd.Sum, name="cl")
+    expect = torch.arange(2 * 4 * 3 * 3, dtype=torch.float32).reshape(
+        2, 4, 3, 3) * sum(r + 1 for r in range(size))
+    assert torch.allclose(out, expect)
+    b = (torch.ones(2, 4, 3, 3) * rank).to(memory_format=torch.channels_last)
+    hvd.broadcast_(b, root_rank=1, name="clb")
+    assert torch.allclose(b, torch.ones(2, 4, 3, 3))
+
+
+def test_channels_last_dense_tensors():
+    run_distributed(_channels_last, 2)
