@@ -93,6 +93,23 @@ Controller::Controller(int rank, int size, ControlPlane* cp)
   //  — 128 MiB vs the NVLink-era 64 MiB default.
   fusion_threshold_ = env_int("AXON_FUSION_THRESHOLD", 128 << 20);
   stall_warn_sec_ = env_double("AXON_STALL_CHECK_TIME", 60.0);
+  cache_capacity_ = env_int("AXON_CACHE_CAPACITY", 2048);
+}
+
+std::string Controller::fingerprint(const Request& q) {
+  std::string fp;
+  fp.reserve(32 + q.shape.size() * 8);
+  fp += static_cast<char>(q.op);
+  fp += static_cast<char>(q.red);
+  fp += static_cast<char>(q.dtype);
+  fp += static_cast<char>(q.wire_dtype);
+  fp += q.device < 0 ? 'c' : 'g';
+  fp += std::to_string(q.root_rank);
+  for (auto d : q.shape) {
+    fp += ':';
+    fp += std::to_string(d);
+  }
+  return fp;
 }
 
 std::vector<Response> Controller::run_cycle(const std::vector<Request>& own,
@@ -101,11 +118,27 @@ std::vector<Response> Controller::run_cycle(const std::vector<Request>& own,
   *shutdown = false;
   *new_cycle_ms = 0.0;
   if (rank_ != 0) {
-    // Worker: ship requests, collect any pending response lists.
+    // Worker: ship requests (cached steady-state allreduces as bare ids),
+    // collect any pending response lists.
     if (!own.empty()) {
+      std::vector<const Request*> full;
+      std::vector<uint32_t> ids;
+      for (auto& q : own) {
+        if (q.op == OpType::ALLREDUCE) {
+          auto it = cache_.find(q.name);
+          if (it != cache_.end() && cache_fp_[q.name] == fingerprint(q)) {
+            ids.push_back(it->second);
+            continue;
+          }
+          sent_fp_[q.name] = fingerprint(q);
+        }
+        full.push_back(&q);
+      }
       Writer w;
-      w.u32(static_cast<uint32_t>(own.size()));
-      for (auto& q : own) q.serialize(w);
+      w.u32(static_cast<uint32_t>(full.size()));
+      for (auto* q : full) q->serialize(w);
+      w.u32(static_cast<uint32_t>(ids.size()));
+      for (uint32_t id : ids) w.u32(id);
       cp_->send(0, MsgType::REQUEST, w.buf);
     }
     std::vector<Response> out;
@@ -119,7 +152,10 @@ std::vector<Response> Controller::run_cycle(const std::vector<Request>& own,
         ResponseList rl = ResponseList::deserialize(payload);
         if (rl.shutdown) *shutdown = true;
         if (rl.new_cycle_ms > 0) *new_cycle_ms = rl.new_cycle_ms;
-        for (auto& r : rl.responses) out.push_back(std::move(r));
+        for (auto& r : rl.responses) {
+          learn(r);
+          out.push_back(std::move(r));
+        }
       }
     }
     return out;
@@ -135,6 +171,22 @@ std::vector<Response> Controller::run_cycle(const std::vector<Request>& own,
       Reader r(payload);
       uint32_t n = r.u32();
       for (uint32_t i = 0; i < n; ++i) ingest(Request::deserialize(r));
+      if (!r.done()) {
+        uint32_t nc = r.u32();
+        for (uint32_t i = 0; i < nc; ++i) {
+          uint32_t id = r.u32();
+          auto it = templates_.find(id);
+          if (it == templates_.end()) {
+            AXON_LOG(LogLevel::ERROR, "axon: unknown cache id " +
+                                          std::to_string(id) + " from rank " +
+                                          std::to_string(peer));
+            continue;
+          }
+          Request q = it->second;
+          q.rank = peer;
+          ingest(q);
+        }
+      }
     }
   }
 
@@ -189,8 +241,27 @@ void Controller::ingest(const Request& q) {
   }
   pt.reqs[q.rank] = q;
   pt.have |= (1u << q.rank);
+  // Metadata change on a cached tensor: refresh the coordinator template
+  // so later bare-id submissions reconstruct the new request.
+  if (q.op == OpType::ALLREDUCE) {
+    auto it = cache_.find(q.name);
+    if (it != cache_.end()) templates_[it->second] = q;
+  }
   uint32_t need = (1u << size_) - 1;
   if ((pt.have | joined_mask_) == need) ready_.push_back(q.name);
+}
+
+void Controller::learn(const Response& r) {
+  if (r.cache_ids.size() != r.names.size()) return;
+  for (size_t i = 0; i < r.names.size(); ++i) {
+    if (r.cache_ids[i] < 0) continue;
+    cache_[r.names[i]] = static_cast<uint32_t>(r.cache_ids[i]);
+    auto it = sent_fp_.find(r.names[i]);
+    if (it != sent_fp_.end()) {
+      cache_fp_[r.names[i]] = std::move(it->second);
+      sent_fp_.erase(it);
+    }
+  }
 }
 
 std::vector<Response> Controller::assemble() {
@@ -265,6 +336,21 @@ Response Controller::make_response(PendingTensor& pt) {
   }
 
   r.numels.push_back(numel0);
+  // Assign (or reuse) a response-cache id for allreduce tensors.
+  if (t0->op == OpType::ALLREDUCE) {
+    int32_t id = -1;
+    auto it = cache_.find(t0->name);
+    if (it != cache_.end()) {
+      id = static_cast<int32_t>(it->second);
+    } else if (static_cast<int64_t>(cache_.size()) < cache_capacity_) {
+      id = static_cast<int32_t>(next_cache_id_++);
+      cache_[t0->name] = static_cast<uint32_t>(id);
+      templates_[static_cast<uint32_t>(id)] = *t0;
+    }
+    r.cache_ids.push_back(id);
+  } else {
+    r.cache_ids.push_back(-1);
+  }
   if (t0->op == OpType::ALLGATHER) {
     std::vector<int64_t> fd(size_, 0);
     for (int rk = 0; rk < size_; ++rk)
@@ -301,6 +387,7 @@ void Controller::fuse(std::vector<Response>& rs) {
         if (pbytes <= fusion_threshold_) {
           p.names.push_back(std::move(r.names[0]));
           p.numels.push_back(r.numels[0]);
+          p.cache_ids.push_back(r.cache_ids.empty() ? -1 : r.cache_ids[0]);
           continue;
         }
       }

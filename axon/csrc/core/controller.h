@@ -84,6 +84,7 @@ class Controller {
   };
 
   void ingest(const Request& q);
+  void learn(const Response& r);  // adopt coordinator-assigned cache ids
   std::vector<Response> assemble();            // coordinator
   Response make_response(PendingTensor& pt);   // validate + build
   void fuse(std::vector<Response>& rs);
@@ -91,6 +92,18 @@ class Controller {
 
   int rank_, size_;
   ControlPlane* cp_;
+  // --- response cache (both roles) ---
+  // Steady-state allreduces re-submit as 4-byte ids: the coordinator
+  // assigns an id the first time a tensor's response is emitted; every
+  // rank learns the id from the response; subsequent submissions with an
+  // unchanged fingerprint go on the wire as the id alone.
+  static std::string fingerprint(const Request& q);
+  std::unordered_map<std::string, uint32_t> cache_;       // name -> id
+  std::unordered_map<std::string, std::string> cache_fp_; // name -> fp
+  std::unordered_map<std::string, std::string> sent_fp_;  // awaiting id
+  std::unordered_map<uint32_t, Request> templates_;       // coordinator
+  uint32_t next_cache_id_ = 0;
+  int64_t cache_capacity_;
   // --- coordinator state ---
   std::unordered_map<std::string, PendingTensor> table_;
   std::deque<std::string> ready_;   // names ready this cycle, arrival order

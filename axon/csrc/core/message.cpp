@@ -58,6 +58,8 @@ void Response::serialize(Writer& w) const {
   for (auto& v : first_dims) put_i64_vec(w, v);
   put_i64_vec(w, splits_matrix);
   w.i32(last_joined);
+  w.u32(static_cast<uint32_t>(cache_ids.size()));
+  for (int32_t id : cache_ids) w.i32(id);
 }
 
 Response Response::deserialize(Reader& r) {
@@ -78,6 +80,9 @@ Response Response::deserialize(Reader& r) {
   for (uint32_t i = 0; i < fd; ++i) s.first_dims[i] = get_i64_vec(r);
   s.splits_matrix = get_i64_vec(r);
   s.last_joined = r.i32();
+  uint32_t nc = r.u32();
+  s.cache_ids.resize(nc);
+  for (uint32_t i = 0; i < nc; ++i) s.cache_ids[i] = r.i32();
   return s;
 }
 

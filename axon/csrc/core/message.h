@@ -42,6 +42,10 @@ struct Response {
   // ALLTOALL: row-major [src_rank][dst_rank] row counts.
   std::vector<int64_t> splits_matrix;
   int32_t last_joined = -1;  // JOIN result
+  // Response-cache ids assigned by the coordinator, parallel to `names`
+  // (-1 = uncached). Workers re-submit steady-state allreduces as 4-byte
+  // ids instead of full Requests (SURVEY.md §2.1 "Response cache").
+  std::vector<int32_t> cache_ids;
 
   void serialize(Writer& w) const;
   static Response deserialize(Reader& r);

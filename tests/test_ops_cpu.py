@@ -413,3 +413,25 @@ def _channels_last(rank, size):
 
 def test_channels_last_dense_tensors():
     run_distributed(_channels_last, 2)
+
+
+def _steady_state_cache(rank, size):
+    # Same tensor names re-submitted across many steps: after step 1 the
+    # workers ship bare cache ids (SURVEY.md §2.1 response cache) — results
+    # must stay correct in every iteration.
+    import axon as hvd
+
+    for it in range(6):
+        handles = [
+            hvd.allreduce_async(torch.full((50 + i,), float(rank + i + it)),
+                                op=hvd.Sum, name=f"steady.{i}")
+            for i in range(8)
+        ]
+        for i, h in enumerate(handles):
+            out = hvd.synchronize(h)
+            expect = float(sum(r + i + it for r in range(size)))
+            assert torch.allclose(out, torch.full((50 + i,), expect)), (it, i)
+
+
+def test_steady_state_response_cache():
+    run_distributed(_steady_state_cache, 4)
