@@ -37,8 +37,18 @@ ChunkPlan build_chunks(const std::vector<at::Tensor>& ps,
   int64_t uoff = 0;
   for (size_t t = 0; t < ps.size(); ++t) {
     auto& p = ps[t];
-    TORCH_CHECK(p.is_contiguous() && gs[t].is_contiguous(),
-                "fused optimizer requires contiguous params/grads");
+    // Elementwise over dense storage: any non-overlapping-and-dense layout
+    // (incl. channels_last) works as long as param/grad/state agree on
+    // strides — same storage index = same logical element on all of them.
+    TORCH_CHECK(p.is_non_overlapping_and_dense() &&
+                    gs[t].is_non_overlapping_and_dense(),
+                "fused optimizer requires dense params/grads");
+    TORCH_CHECK(p.numel() <= 1 || gs[t].strides() == p.strides(),
+                "fused optimizer: grad layout must match param layout");
+    TORCH_CHECK(ms.empty() || p.numel() <= 1 || ms[t].strides() == p.strides(),
+                "fused optimizer: state layout must match param layout");
+    TORCH_CHECK(vs.empty() || p.numel() <= 1 || vs[t].strides() == p.strides(),
+                "fused optimizer: state layout must match param layout");
     size_t esize = p.element_size();
     char* pp = static_cast<char*>(p.data_ptr());
     char* gg = static_cast<char*>(gs[t].data_ptr());

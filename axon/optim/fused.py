@@ -5,6 +5,19 @@ import torch
 from torch.optim import Optimizer
 
 
+def _like_param(p, t):
+    """The fused kernels walk flat storage, so grad/state must share the
+    param's strides (channels_last params included). Autograd or a loaded
+    checkpoint can hand us a differently-laid-out tensor — restride it once.
+    Grads are read-only in the kernels, so a copy is safe."""
+    if t is None or p.numel() <= 1 or t.stride() == p.stride():
+        return t
+    out = torch.empty_strided(p.shape, p.stride(), dtype=t.dtype,
+                              device=t.device)
+    out.copy_(t)
+    return out
+
+
 def _grouped(params_with_grad):
     """Group (param, grad, state...) lists by (device, dtype)."""
     groups = {}
@@ -43,7 +56,10 @@ class FusedSGD(Optimizer):
                 st = self.state[p]
                 if mu != 0 and "momentum_buffer" not in st:
                     st["momentum_buffer"] = torch.zeros_like(p)
-                work.append((p, p.grad, st["momentum_buffer"] if mu != 0 else None))
+                elif mu != 0:
+                    st["momentum_buffer"] = _like_param(p, st["momentum_buffer"])
+                work.append((p, _like_param(p, p.grad),
+                             st["momentum_buffer"] if mu != 0 else None))
             for tuples in _grouped(work):
                 ps = [t[0] for t in tuples]
                 gs = [t[1] for t in tuples]
@@ -99,8 +115,10 @@ class _FusedAdamBase(Optimizer):
                     st["exp_avg"] = torch.zeros_like(p)
                     st["exp_avg_sq"] = torch.zeros_like(p)
                 st["step"] += 1
-                work.append((p, p.grad, st["exp_avg"], st["exp_avg_sq"],
-                             int(st["step"].item())))
+                st["exp_avg"] = _like_param(p, st["exp_avg"])
+                st["exp_avg_sq"] = _like_param(p, st["exp_avg_sq"])
+                work.append((p, _like_param(p, p.grad), st["exp_avg"],
+                             st["exp_avg_sq"], int(st["step"].item())))
             b1, b2 = group["betas"]
             for tuples in _grouped(work):
                 # all params in a (device,dtype) bucket share the step count
@@ -187,8 +205,10 @@ class FusedLAMB(Optimizer):
                     st["exp_avg"] = torch.zeros_like(p)
                     st["exp_avg_sq"] = torch.zeros_like(p)
                 st["step"] += 1
-                work.append((p, p.grad, st["exp_avg"], st["exp_avg_sq"],
-                             int(st["step"].item())))
+                st["exp_avg"] = _like_param(p, st["exp_avg"])
+                st["exp_avg_sq"] = _like_param(p, st["exp_avg_sq"])
+                work.append((p, _like_param(p, p.grad), st["exp_avg"],
+                             st["exp_avg_sq"], int(st["step"].item())))
             b1, b2 = group["betas"]
             for tuples in _grouped(work):
                 by_step = {}

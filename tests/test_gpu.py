@@ -183,6 +183,38 @@ def test_fused_sgd_matches_torch_gpu(hvd):
         assert torch.allclose(p, q, rtol=1e-5, atol=1e-6)
 
 
+def test_fused_sgd_channels_last_gpu(hvd):
+    """channels_last params/grads (ResNet NHWC path): the flat-storage
+    kernels must accept matching non-contiguous dense layouts."""
+    import torch.nn as nn
+
+    from axon.optim import FusedSGD
+
+    torch.manual_seed(11)
+    m1 = nn.Sequential(nn.Conv2d(8, 16, 3, padding=1), nn.ReLU(),
+                       nn.Conv2d(16, 8, 3, padding=1)).to(DEV)
+    m2 = nn.Sequential(nn.Conv2d(8, 16, 3, padding=1), nn.ReLU(),
+                       nn.Conv2d(16, 8, 3, padding=1)).to(DEV)
+    m2.load_state_dict(m1.state_dict())
+    m1 = m1.to(memory_format=torch.channels_last)
+    o1 = FusedSGD(m1.parameters(), lr=0.05, momentum=0.9, weight_decay=1e-3)
+    o2 = torch.optim.SGD(m2.parameters(), lr=0.05, momentum=0.9,
+                         weight_decay=1e-3)
+    for s in range(4):
+        torch.manual_seed(20 + s)
+        x = torch.randn(4, 8, 16, 16, device=DEV)
+        (m1(x.to(memory_format=torch.channels_last)) ** 2).mean().backward()
+        (m2(x) ** 2).mean().backward()
+        o1.step()
+        o2.step()
+        o1.zero_grad()
+        o2.zero_grad()
+    for p, q in zip(m1.parameters(), m2.parameters()):
+        assert p.is_non_overlapping_and_dense()
+        assert torch.allclose(p.contiguous(), q.contiguous(),
+                              rtol=1e-4, atol=1e-5)
+
+
 # ----------------------------------------------------------- end to end
 
 def test_resnet50_bf16_step(hvd):
