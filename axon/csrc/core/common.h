@@ -56,6 +56,25 @@ struct Status {
 
 using StatusCallback = std::function<void(const Status&)>;
 
+// Process set: a named subgroup of ranks with its own collectives and (on
+// GPU) its own RCCL communicator (SURVEY.md §2.1 "Process sets"). Id 0 is
+// the global set. Registration must happen collectively and in identical
+// order on every rank (enforced by the Python API via allgather).
+struct ProcessSetInfo {
+  int id = 0;
+  std::vector<int> ranks;  // sorted global ranks
+  uint32_t mask = 0;       // bit r set iff rank r is a member
+  int my_index = -1;       // this rank's position in `ranks`; -1 = non-member
+  bool valid = true;       // false after remove_process_set
+  int size() const { return static_cast<int>(ranks.size()); }
+  bool member(int r) const { return (mask >> r) & 1u; }
+  int index_of(int r) const {
+    for (size_t i = 0; i < ranks.size(); ++i)
+      if (ranks[i] == r) return static_cast<int>(i);
+    return -1;
+  }
+};
+
 // One pending collective operation on one rank.
 struct TensorEntry {
   std::string name;
@@ -70,6 +89,7 @@ struct TensorEntry {
   std::vector<int64_t> splits;          // alltoall send splits (rows per rank)
   at::Tensor received_splits;           // alltoall: out tensor for recv splits
   int32_t device = -1;        // hip device index, -1 = CPU
+  int32_t process_set = 0;    // process-set id (0 = global)
   uint64_t ready_event = 0;   // hipEvent_t recorded on producer stream (opaque)
   uint64_t handle = 0;        // binding-side handle id
   StatusCallback callback;

@@ -3,6 +3,8 @@
 #include <algorithm>
 #include <sstream>
 
+#include "operations.h"  // ps_get/ps_count (process-set registry)
+
 namespace axon {
 
 // ---------------------------------------------------------------- ControlPlane
@@ -105,6 +107,8 @@ std::string Controller::fingerprint(const Request& q) {
   fp += static_cast<char>(q.wire_dtype);
   fp += q.device < 0 ? 'c' : 'g';
   fp += std::to_string(q.root_rank);
+  fp += '/';
+  fp += std::to_string(q.process_set);
   for (auto d : q.shape) {
     fp += ':';
     fp += std::to_string(d);
@@ -190,6 +194,12 @@ std::vector<Response> Controller::run_cycle(const std::vector<Request>& own,
     }
   }
 
+  // A process set registered after its members' requests arrived: re-check
+  // readiness of everything deferred on an unknown set.
+  if (ps_count() != last_ps_count_) {
+    last_ps_count_ = ps_count();
+    rescan_ready();
+  }
   std::vector<Response> rs = assemble();
   check_stalls();
 
@@ -214,18 +224,9 @@ void Controller::ingest(const Request& q) {
   if (q.op == OpType::JOIN) {
     joined_mask_ |= (1u << q.rank);
     last_joined_ = q.rank;
-    // A new join may complete tensors that were waiting on this rank, and
-    // may complete the join itself.
-    std::vector<std::string> now_ready;
-    for (auto& kv : table_) {
-      uint32_t need = (1u << size_) - 1;
-      if ((kv.second.have | joined_mask_) == need) now_ready.push_back(kv.first);
-    }
-    std::sort(now_ready.begin(), now_ready.end(),
-              [&](const std::string& a, const std::string& b) {
-                return table_[a].order < table_[b].order;
-              });
-    for (auto& n : now_ready) ready_.push_back(n);
+    // A new join may complete tensors that were waiting on this rank (in
+    // any process set the rank belongs to), and may complete the join.
+    rescan_ready();
     return;
   }
   auto& pt = table_[q.name];
@@ -233,6 +234,7 @@ void Controller::ingest(const Request& q) {
     pt.reqs.resize(size_);
     pt.first_seen = std::chrono::steady_clock::now();
     pt.order = order_counter_++;
+    pt.pset = q.process_set;
   }
   if (pt.have & (1u << q.rank)) {
     AXON_LOG(LogLevel::WARN, "axon: duplicate request for tensor '" + q.name +
@@ -247,8 +249,29 @@ void Controller::ingest(const Request& q) {
     auto it = cache_.find(q.name);
     if (it != cache_.end()) templates_[it->second] = q;
   }
-  uint32_t need = (1u << size_) - 1;
-  if ((pt.have | joined_mask_) == need) ready_.push_back(q.name);
+  maybe_ready(q.name, pt);
+}
+
+void Controller::maybe_ready(const std::string& name, PendingTensor& pt) {
+  if (pt.queued) return;
+  ProcessSetInfo psi = ps_get(pt.pset);
+  if (!psi.valid) return;  // set not registered here yet; re-checked per cycle
+  uint32_t need = psi.mask;
+  if ((pt.have | (joined_mask_ & need)) == need) {
+    pt.queued = true;
+    ready_.push_back(name);
+  }
+}
+
+void Controller::rescan_ready() {
+  std::vector<std::string> names;
+  for (auto& kv : table_)
+    if (!kv.second.queued) names.push_back(kv.first);
+  std::sort(names.begin(), names.end(),
+            [&](const std::string& a, const std::string& b) {
+              return table_[a].order < table_[b].order;
+            });
+  for (auto& n : names) maybe_ready(n, table_[n]);
 }
 
 void Controller::learn(const Response& r) {
@@ -300,11 +323,19 @@ Response Controller::make_response(PendingTensor& pt) {
   r.wire_dtype = t0->wire_dtype;
   r.device = t0->device;
   r.root_rank = t0->root_rank;
+  r.process_set = t0->process_set;
+  ProcessSetInfo psi = ps_get(t0->process_set);
+  const int set_size = psi.size();
 
   std::ostringstream err;
   int64_t numel0 = 1;
   for (auto d : t0->shape) numel0 *= d;
-  for (int rk = 0; rk < size_; ++rk) {
+  if (!psi.valid) err << "process set " << t0->process_set << " was removed";
+  else if (t0->op == OpType::BROADCAST && !psi.member(t0->root_rank))
+    err << "broadcast root " << t0->root_rank << " not in process set on '"
+        << t0->name << "'";
+  for (int rk : psi.ranks) {
+    if (err.tellp() > 0) break;
     const Request& q = pt.reqs[rk];
     if (q.name.empty()) continue;  // joined rank
     if (q.op != t0->op)
@@ -325,10 +356,11 @@ Response Controller::make_response(PendingTensor& pt) {
                       t0->shape.begin() + (t0->shape.empty() ? 0 : 1)))
         err << "non-first-dim shape mismatch on allgather '" << t0->name << "'";
     } else if (q.op == OpType::ALLTOALL) {
-      if (static_cast<int>(q.splits.size()) != size_)
-        err << "alltoall splits size != world_size on '" << t0->name << "'";
+      if (static_cast<int>(q.splits.size()) != set_size)
+        err << "alltoall splits size != process-set size on '" << t0->name << "'";
     }
-    if (err.tellp() > 0) break;
+    if (q.process_set != t0->process_set)
+      err << "process-set mismatch on '" << t0->name << "'";
   }
   if (err.tellp() > 0) {
     r.error = err.str();
@@ -352,17 +384,26 @@ Response Controller::make_response(PendingTensor& pt) {
     r.cache_ids.push_back(-1);
   }
   if (t0->op == OpType::ALLGATHER) {
-    std::vector<int64_t> fd(size_, 0);
-    for (int rk = 0; rk < size_; ++rk)
-      if (!pt.reqs[rk].name.empty() && !pt.reqs[rk].shape.empty())
-        fd[rk] = pt.reqs[rk].shape[0];
+    // first_dims[i] = dim-0 extent from the i-th SET MEMBER (set order).
+    std::vector<int64_t> fd(set_size, 0);
+    for (int i = 0; i < set_size; ++i) {
+      const Request& q = pt.reqs[psi.ranks[i]];
+      if (!q.name.empty() && !q.shape.empty()) fd[i] = q.shape[0];
+    }
     r.first_dims.push_back(std::move(fd));
   } else if (t0->op == OpType::ALLTOALL) {
-    r.splits_matrix.assign(static_cast<size_t>(size_) * size_, 0);
-    for (int rk = 0; rk < size_; ++rk)
-      if (!pt.reqs[rk].name.empty())
-        for (int j = 0; j < size_; ++j)
-          r.splits_matrix[static_cast<size_t>(rk) * size_ + j] = pt.reqs[rk].splits[j];
+    // [src_member][dst_member] row counts, set-relative indexing.
+    r.splits_matrix.assign(static_cast<size_t>(set_size) * set_size, 0);
+    for (int i = 0; i < set_size; ++i) {
+      const Request& q = pt.reqs[psi.ranks[i]];
+      if (!q.name.empty())
+        for (int j = 0; j < set_size; ++j)
+          r.splits_matrix[static_cast<size_t>(i) * set_size + j] = q.splits[j];
+    }
+  } else if (t0->op == OpType::REDUCESCATTER) {
+    // Dim-0 extent (rows to split across members) — lets a non-member
+    // rank-0 relay compute the per-member slices without the shape.
+    r.first_dims.push_back({t0->shape.empty() ? 1 : t0->shape[0]});
   }
   return r;
 }
@@ -379,7 +420,8 @@ void Controller::fuse(std::vector<Response>& rs) {
       Response& p = out.back();
       if (p.op == OpType::ALLREDUCE && p.error.empty() &&
           p.red == r.red && p.dtype == r.dtype &&
-          p.wire_dtype == r.wire_dtype && p.device == r.device) {
+          p.wire_dtype == r.wire_dtype && p.device == r.device &&
+          p.process_set == r.process_set) {
         size_t esz = dtype_size(id_dtype(r.wire_dtype ? r.wire_dtype : r.dtype));
         int64_t pbytes = 0;
         for (auto n : p.numels) pbytes += n;
@@ -405,13 +447,17 @@ void Controller::check_stalls() {
     if (pt.warned) continue;
     double age = std::chrono::duration<double>(now - pt.first_seen).count();
     if (age > stall_warn_sec_) {
+      uint32_t need = ps_get(pt.pset).mask;
+      if (!need) need = (1u << size_) - 1;
       std::ostringstream os;
       os << "axon stall inspector: tensor '" << kv.first << "' submitted by ranks [";
       for (int rk = 0; rk < size_; ++rk)
         if (pt.have & (1u << rk)) os << rk << " ";
       os << "] but missing from ranks [";
       for (int rk = 0; rk < size_; ++rk)
-        if (!(pt.have & (1u << rk)) && !(joined_mask_ & (1u << rk))) os << rk << " ";
+        if ((need & (1u << rk)) && !(pt.have & (1u << rk)) &&
+            !(joined_mask_ & (1u << rk)))
+          os << rk << " ";
       os << "] for " << static_cast<int>(age) << "s — one or more ranks may be stuck";
       AXON_LOG(LogLevel::WARN, os.str());
       pt.warned = true;

@@ -76,14 +76,18 @@ class Controller {
 
  private:
   struct PendingTensor {
-    std::vector<Request> reqs;      // indexed by rank; empty name = missing
+    std::vector<Request> reqs;      // indexed by GLOBAL rank; empty name = missing
     uint32_t have = 0;              // bitmask of ranks that requested
     std::chrono::steady_clock::time_point first_seen;
     bool warned = false;
+    bool queued = false;            // already pushed to ready_
     int order = 0;                  // arrival order of first request
+    int pset = 0;                   // process-set id
   };
 
   void ingest(const Request& q);
+  void maybe_ready(const std::string& name, PendingTensor& pt);
+  void rescan_ready();  // after joins / late process-set registration
   void learn(const Response& r);  // adopt coordinator-assigned cache ids
   std::vector<Response> assemble();            // coordinator
   Response make_response(PendingTensor& pt);   // validate + build
@@ -110,6 +114,7 @@ class Controller {
   uint32_t joined_mask_ = 0;
   int last_joined_ = -1;
   int order_counter_ = 0;
+  int last_ps_count_ = 1;  // re-scan readiness when new sets register
   int64_t fusion_threshold_;
   double stall_warn_sec_;
   bool pending_shutdown_ = false;

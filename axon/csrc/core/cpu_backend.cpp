@@ -2,6 +2,8 @@
 
 #include <ATen/Functions.h>
 
+#include "operations.h"  // ps_get
+
 namespace axon {
 
 namespace {
@@ -19,6 +21,15 @@ at::Tensor flat(const at::Tensor& t) {
 }
 }  // namespace
 
+// All CPU collectives run hub-and-spoke over the unix-socket star: global
+// rank 0 is ALWAYS the hub (it is the only rank every worker has a channel
+// to). For a process set that does not contain rank 0, the hub acts as a
+// pure relay: it is invoked with empty `entries` at the same globally
+// agreed point in the response order as the members, receives the members'
+// payloads, combines/routes them, and sends results back (SURVEY.md §2.1
+// "Process sets"; the GPU data plane instead gets a per-set RCCL
+// communicator and needs rank 0 only for the unique-id rendezvous).
+
 void CpuBackend::send_tensor(int peer, const at::Tensor& t) {
   at::Tensor c = t.contiguous();
   Writer w;
@@ -35,6 +46,19 @@ at::Tensor CpuBackend::recv_tensor(int peer, at::ScalarType dtype, int64_t numel
                              std::to_string(payload.size()) + " vs " +
                              std::to_string(want) + ")");
   std::memcpy(t.data_ptr(), payload.data(), want);
+  return t;
+}
+
+at::Tensor CpuBackend::recv_tensor_any(int peer, at::ScalarType dtype) {
+  // Size inferred from the payload — used by the relay, which does not
+  // always know the logical shape.
+  auto payload = cp_->recv_data(peer);
+  size_t esize = dtype_size(dtype);
+  if (payload.size() % esize != 0)
+    throw std::runtime_error("axon cpu backend: payload not a dtype multiple");
+  int64_t numel = static_cast<int64_t>(payload.size() / esize);
+  at::Tensor t = at::empty({numel}, at::TensorOptions().dtype(dtype));
+  std::memcpy(t.data_ptr(), payload.data(), payload.size());
   return t;
 }
 
@@ -60,19 +84,33 @@ Status CpuBackend::execute(const Response& r, std::vector<TensorEntry>& entries)
 }
 
 Status CpuBackend::allreduce(const Response& r, std::vector<TensorEntry>& entries) {
+  ProcessSetInfo psi = ps_get(r.process_set);
+  const bool im_member = psi.member(rank_);
   at::ScalarType dtype = id_dtype(r.dtype);
   at::ScalarType wire = r.wire_dtype ? id_dtype(r.wire_dtype) : dtype;
-  for (auto& e : entries) {
-    at::Tensor t = flat(e.in);
-    if (e.prescale != 1.0) t = t.mul(e.prescale);
-    if (wire != dtype) t = t.to(wire);
-    t = t.contiguous();
+  const size_t ntensors = im_member ? entries.size() : r.names.size();
+  const double nmem = static_cast<double>(psi.size());
+
+  for (size_t ti = 0; ti < ntensors; ++ti) {
+    at::Tensor t;
+    if (im_member) {
+      auto& e = entries[ti];
+      t = flat(e.in);
+      if (e.prescale != 1.0) t = t.mul(e.prescale);
+      if (wire != dtype) t = t.to(wire);
+      t = t.contiguous();
+    }
 
     at::Tensor result;
     if (rank_ == 0) {
-      at::Tensor acc = t.clone();
-      for (int peer = 1; peer < size_; ++peer) {
-        at::Tensor s = recv_tensor(peer, wire, t.numel());
+      at::Tensor acc;
+      bool first = true;
+      if (im_member) { acc = t.clone(); first = false; }
+      for (int peer : psi.ranks) {
+        if (peer == 0) continue;
+        at::Tensor s = first ? recv_tensor_any(peer, wire)
+                             : recv_tensor(peer, wire, acc.numel());
+        if (first) { acc = s; first = false; continue; }
         switch (r.red) {
           case ReduceOp::SUM:
           case ReduceOp::AVERAGE:
@@ -91,16 +129,20 @@ Status CpuBackend::allreduce(const Response& r, std::vector<TensorEntry>& entrie
             return Status::Error("bad reduce op");
         }
       }
-      if (r.red == ReduceOp::AVERAGE) acc.div_(static_cast<double>(size_));
-      for (int peer = 1; peer < size_; ++peer) send_tensor(peer, acc);
+      if (r.red == ReduceOp::AVERAGE) acc.div_(nmem);
+      for (int peer : psi.ranks)
+        if (peer != 0) send_tensor(peer, acc);
       result = acc;
     } else {
       send_tensor(0, t);
       result = recv_tensor(0, wire, t.numel());
     }
-    at::Tensor out = flat(e.out);
-    out.copy_(result);  // casts wire -> dtype if compressed
-    if (e.postscale != 1.0) out.mul_(e.postscale);
+    if (im_member) {
+      auto& e = entries[ti];
+      at::Tensor out = flat(e.out);
+      out.copy_(result);  // casts wire -> dtype if compressed
+      if (e.postscale != 1.0) out.mul_(e.postscale);
+    }
   }
   return Status::OK();
 }
@@ -117,118 +159,174 @@ at::Tensor adasum_combine_cpu(const at::Tensor& a, const at::Tensor& b) {
 
 Status CpuBackend::adasum(const Response& r, std::vector<TensorEntry>& entries) {
   // Central simulation of the recursive-halving pairwise combine tree
-  // (level k pairs rank r with r^2^k): gather on rank 0, fold the tree in
-  // fp64, fan the result out. CPU path is a correctness oracle, not a perf
-  // path — the GPU path (rccl_backend) does the real distributed recursion.
-  if ((size_ & (size_ - 1)) != 0)
-    return Status::Error("adasum requires a power-of-two world size");
+  // (level k pairs member i with i^2^k): gather on the hub, fold the tree
+  // in fp64, fan the result out. CPU path is a correctness oracle, not a
+  // perf path — the GPU path (rccl_backend) does the real distributed
+  // recursion.
+  ProcessSetInfo psi = ps_get(r.process_set);
+  const bool im_member = psi.member(rank_);
+  const int nmem = psi.size();
+  if ((nmem & (nmem - 1)) != 0)
+    return Status::Error("adasum requires a power-of-two process-set size");
   at::ScalarType dtype = id_dtype(r.dtype);
-  for (auto& e : entries) {
-    at::Tensor t = flat(e.in);
-    if (e.prescale != 1.0) t = t.mul(e.prescale);
+  const size_t ntensors = im_member ? entries.size() : r.names.size();
+  for (size_t ti = 0; ti < ntensors; ++ti) {
+    at::Tensor t;
+    if (im_member) {
+      auto& e = entries[ti];
+      t = flat(e.in);
+      if (e.prescale != 1.0) t = t.mul(e.prescale);
+    }
     at::Tensor result;
     if (rank_ == 0) {
-      std::vector<at::Tensor> v(size_);
-      v[0] = t.to(at::kDouble);
-      for (int peer = 1; peer < size_; ++peer)
-        v[peer] = recv_tensor(peer, dtype, t.numel()).to(at::kDouble);
-      for (int stride = 1; stride < size_; stride *= 2)
-        for (int i = 0; i + stride < size_; i += 2 * stride)
+      std::vector<at::Tensor> v(nmem);
+      for (int i = 0; i < nmem; ++i) {
+        int peer = psi.ranks[i];
+        v[i] = peer == 0 ? t.to(at::kDouble)
+                         : recv_tensor_any(peer, dtype).to(at::kDouble);
+      }
+      for (int stride = 1; stride < nmem; stride *= 2)
+        for (int i = 0; i + stride < nmem; i += 2 * stride)
           v[i] = adasum_combine_cpu(v[i], v[i + stride]);
       result = v[0].to(dtype);
-      for (int peer = 1; peer < size_; ++peer) send_tensor(peer, result);
+      for (int peer : psi.ranks)
+        if (peer != 0) send_tensor(peer, result);
     } else {
       send_tensor(0, t.contiguous());
       result = recv_tensor(0, dtype, t.numel());
     }
-    at::Tensor out = flat(e.out);
-    out.copy_(result);
-    if (e.postscale != 1.0) out.mul_(e.postscale);
+    if (im_member) {
+      auto& e = entries[ti];
+      at::Tensor out = flat(e.out);
+      out.copy_(result);
+      if (e.postscale != 1.0) out.mul_(e.postscale);
+    }
   }
   return Status::OK();
 }
 
 Status CpuBackend::broadcast(const Response& r, std::vector<TensorEntry>& entries) {
-  for (auto& e : entries) {
-    int root = r.root_rank;
+  ProcessSetInfo psi = ps_get(r.process_set);
+  const bool im_member = psi.member(rank_);
+  const size_t ntensors = im_member ? entries.size() : r.names.size();
+  const int root = r.root_rank;
+  for (size_t ti = 0; ti < ntensors; ++ti) {
     at::Tensor buf;
-    if (rank_ == root) {
-      buf = flat(e.in);
+    if (im_member && rank_ == root) {
+      buf = flat(entries[ti].in);
       if (root != 0) send_tensor(0, buf);
     }
     if (rank_ == 0) {
-      if (root != 0) buf = recv_tensor(root, id_dtype(r.dtype), e.out.numel());
-      for (int peer = 1; peer < size_; ++peer)
-        if (peer != root) send_tensor(peer, buf);
+      if (root != 0) buf = recv_tensor_any(root, id_dtype(r.dtype));
+      for (int peer : psi.ranks)
+        if (peer != 0 && peer != root) send_tensor(peer, buf);
     } else if (rank_ != root) {
-      buf = recv_tensor(0, id_dtype(r.dtype), e.out.numel());
+      buf = recv_tensor(0, id_dtype(r.dtype), entries[ti].out.numel());
     }
-    if (rank_ != root || !e.out.is_same(e.in))
-      flat(e.out).copy_(buf);
+    if (im_member) {
+      auto& e = entries[ti];
+      if (rank_ != root || !e.out.is_same(e.in)) flat(e.out).copy_(buf);
+    }
   }
   return Status::OK();
 }
 
 Status CpuBackend::allgather(const Response& r, std::vector<TensorEntry>& entries) {
+  ProcessSetInfo psi = ps_get(r.process_set);
+  const bool im_member = psi.member(rank_);
+  const int nmem = psi.size();
   at::ScalarType dtype = id_dtype(r.dtype);
-  for (size_t ti = 0; ti < entries.size(); ++ti) {
-    auto& e = entries[ti];
-    const auto& fd = r.first_dims.at(ti);
-    auto in_sizes = e.in.sizes().vec();
-    int64_t row = 1;
-    for (size_t d = 1; d < in_sizes.size(); ++d) row *= in_sizes[d];
-    int64_t total_rows = 0;
-    for (auto x : fd) total_rows += x;
+  const size_t ntensors = im_member ? entries.size() : r.names.size();
+  for (size_t ti = 0; ti < ntensors; ++ti) {
+    const auto& fd = r.first_dims.at(ti);  // indexed by set member
 
     at::Tensor full;
     if (rank_ == 0) {
-      full = at::empty({total_rows * row}, at::TensorOptions().dtype(dtype));
+      std::vector<at::Tensor> parts(nmem);
+      int64_t total = 0;
+      for (int i = 0; i < nmem; ++i) {
+        int peer = psi.ranks[i];
+        if (fd[i] == 0) { parts[i] = at::empty({0}, at::TensorOptions().dtype(dtype)); continue; }
+        parts[i] = peer == 0 ? flat(entries[ti].in).contiguous()
+                             : recv_tensor_any(peer, dtype);
+        total += parts[i].numel();
+      }
+      full = at::empty({total}, at::TensorOptions().dtype(dtype));
       int64_t off = 0;
-      for (int rk = 0; rk < size_; ++rk) {
-        int64_t n = fd[rk] * row;
-        if (n == 0) continue;
-        at::Tensor part = rk == 0 ? flat(e.in) : recv_tensor(rk, dtype, n);
-        full.narrow(0, off, n).copy_(part);
+      for (int i = 0; i < nmem; ++i) {
+        int64_t n = parts[i].numel();
+        if (n) full.narrow(0, off, n).copy_(parts[i]);
         off += n;
       }
-      for (int peer = 1; peer < size_; ++peer) send_tensor(peer, full);
+      for (int peer : psi.ranks)
+        if (peer != 0) send_tensor(peer, full);
     } else {
+      auto& e = entries[ti];
       send_tensor(0, flat(e.in));
+      auto in_sizes = e.in.sizes().vec();
+      int64_t row = 1;
+      for (size_t d = 1; d < in_sizes.size(); ++d) row *= in_sizes[d];
+      int64_t total_rows = 0;
+      for (auto x : fd) total_rows += x;
       full = recv_tensor(0, dtype, total_rows * row);
     }
-    auto out_sizes = in_sizes;
-    if (out_sizes.empty()) out_sizes.push_back(total_rows); else out_sizes[0] = total_rows;
-    e.out.resize_(out_sizes);
-    flat(e.out).copy_(full);
+    if (im_member) {
+      auto& e = entries[ti];
+      auto in_sizes = e.in.sizes().vec();
+      int64_t row = 1;
+      for (size_t d = 1; d < in_sizes.size(); ++d) row *= in_sizes[d];
+      int64_t total_rows = 0;
+      for (auto x : fd) total_rows += x;
+      auto out_sizes = in_sizes;
+      if (out_sizes.empty()) out_sizes.push_back(total_rows);
+      else out_sizes[0] = total_rows;
+      e.out.resize_(out_sizes);
+      flat(e.out).copy_(full);
+    }
   }
   return Status::OK();
 }
 
 Status CpuBackend::alltoall(const Response& r, std::vector<TensorEntry>& entries) {
+  ProcessSetInfo psi = ps_get(r.process_set);
+  const bool im_member = psi.member(rank_);
+  const int nmem = psi.size();
+  const int my_idx = psi.my_index;
   at::ScalarType dtype = id_dtype(r.dtype);
+  // splits_matrix is [src_member][dst_member] row counts.
   auto m = [&](int src, int dst) {
-    return r.splits_matrix.at(static_cast<size_t>(src) * size_ + dst);
+    return r.splits_matrix.at(static_cast<size_t>(src) * nmem + dst);
   };
-  for (auto& e : entries) {
-    auto in_sizes = e.in.sizes().vec();
+  const size_t ntensors = im_member ? entries.size() : r.names.size();
+  for (size_t ti = 0; ti < ntensors; ++ti) {
     int64_t row = 1;
-    for (size_t d = 1; d < in_sizes.size(); ++d) row *= in_sizes[d];
+    if (im_member) {
+      auto in_sizes = entries[ti].in.sizes().vec();
+      for (size_t d = 1; d < in_sizes.size(); ++d) row *= in_sizes[d];
+    }
 
     at::Tensor mine;
     if (rank_ == 0) {
-      std::vector<at::Tensor> inputs(size_);
-      inputs[0] = flat(e.in);
-      for (int rk = 1; rk < size_; ++rk) {
-        int64_t rows = 0;
-        for (int j = 0; j < size_; ++j) rows += m(rk, j);
-        inputs[rk] = recv_tensor(rk, dtype, rows * row);
+      std::vector<at::Tensor> inputs(nmem);
+      for (int i = 0; i < nmem; ++i) {
+        int peer = psi.ranks[i];
+        inputs[i] = peer == 0 ? flat(entries[ti].in).contiguous()
+                              : recv_tensor_any(peer, dtype);
+        if (!im_member && row == 1) {
+          // relay: derive the row element count from any member's payload
+          int64_t rows_i = 0;
+          for (int j = 0; j < nmem; ++j) rows_i += m(i, j);
+          if (rows_i > 0) row = inputs[i].numel() / rows_i;
+        }
       }
-      for (int dst = 0; dst < size_; ++dst) {
+      if (row == 0) row = 1;
+      for (int dst = 0; dst < nmem; ++dst) {
         int64_t rows_dst = 0;
-        for (int src = 0; src < size_; ++src) rows_dst += m(src, dst);
-        at::Tensor outbuf = at::empty({rows_dst * row}, at::TensorOptions().dtype(dtype));
+        for (int src = 0; src < nmem; ++src) rows_dst += m(src, dst);
+        at::Tensor outbuf =
+            at::empty({rows_dst * row}, at::TensorOptions().dtype(dtype));
         int64_t off = 0;
-        for (int src = 0; src < size_; ++src) {
+        for (int src = 0; src < nmem; ++src) {
           int64_t pre = 0;  // rows src sends to dst' < dst
           for (int j = 0; j < dst; ++j) pre += m(src, j);
           int64_t n = m(src, dst) * row;
@@ -236,64 +334,90 @@ Status CpuBackend::alltoall(const Response& r, std::vector<TensorEntry>& entries
             outbuf.narrow(0, off, n).copy_(inputs[src].narrow(0, pre * row, n));
           off += n;
         }
-        if (dst == 0) mine = outbuf;
-        else send_tensor(dst, outbuf);
+        int peer = psi.ranks[dst];
+        if (peer == 0) mine = outbuf;
+        else send_tensor(peer, outbuf);
       }
     } else {
-      send_tensor(0, flat(e.in));
+      send_tensor(0, flat(entries[ti].in));
       int64_t rows_me = 0;
-      for (int src = 0; src < size_; ++src) rows_me += m(src, rank_);
+      for (int src = 0; src < nmem; ++src) rows_me += m(src, my_idx);
       mine = recv_tensor(0, dtype, rows_me * row);
     }
-    auto out_sizes = in_sizes;
-    int64_t rows_me = mine.numel() / std::max<int64_t>(row, 1);
-    if (out_sizes.empty()) out_sizes.push_back(rows_me); else out_sizes[0] = rows_me;
-    e.out.resize_(out_sizes);
-    flat(e.out).copy_(mine);
-    if (e.received_splits.defined()) {
-      auto rs = e.received_splits;
-      rs.resize_({size_});
-      for (int src = 0; src < size_; ++src)
-        rs[src] = m(src, rank_);
+    if (im_member) {
+      auto& e = entries[ti];
+      auto out_sizes = e.in.sizes().vec();
+      int64_t rows_me = mine.numel() / std::max<int64_t>(row, 1);
+      if (out_sizes.empty()) out_sizes.push_back(rows_me);
+      else out_sizes[0] = rows_me;
+      e.out.resize_(out_sizes);
+      flat(e.out).copy_(mine);
+      if (e.received_splits.defined()) {
+        auto rs = e.received_splits;
+        rs.resize_({nmem});
+        for (int src = 0; src < nmem; ++src) rs[src] = m(src, my_idx);
+      }
     }
   }
   return Status::OK();
 }
 
 Status CpuBackend::reducescatter(const Response& r, std::vector<TensorEntry>& entries) {
+  ProcessSetInfo psi = ps_get(r.process_set);
+  const bool im_member = psi.member(rank_);
+  const int nmem = psi.size();
+  const int my_idx = psi.my_index;
   at::ScalarType dtype = id_dtype(r.dtype);
-  for (auto& e : entries) {
-    auto in_sizes = e.in.sizes().vec();
-    int64_t rows = in_sizes.empty() ? 1 : in_sizes[0];
-    int64_t row = 1;
-    for (size_t d = 1; d < in_sizes.size(); ++d) row *= in_sizes[d];
-    int64_t base = rows / size_, rem = rows % size_;
-    auto rank_rows = [&](int rk) { return base + (rk < rem ? 1 : 0); };
-    auto rank_off = [&](int rk) {
-      return static_cast<int64_t>(rk) * base + std::min<int64_t>(rk, rem);
+  const size_t ntensors = im_member ? entries.size() : r.names.size();
+  for (size_t ti = 0; ti < ntensors; ++ti) {
+    // Dim-0 rows from the response (also works for the non-member relay).
+    int64_t rows = r.first_dims.at(ti).at(0);
+    int64_t row = r.numels.at(ti) / std::max<int64_t>(rows, 1);
+    int64_t base = rows / nmem, rem = rows % nmem;
+    auto member_rows = [&](int i) { return base + (i < rem ? 1 : 0); };
+    auto member_off = [&](int i) {
+      return static_cast<int64_t>(i) * base + std::min<int64_t>(i, rem);
     };
 
-    at::Tensor t = flat(e.in);
-    if (e.prescale != 1.0) t = t.mul(e.prescale);
+    at::Tensor t;
+    if (im_member) {
+      auto& e = entries[ti];
+      t = flat(e.in);
+      if (e.prescale != 1.0) t = t.mul(e.prescale);
+    }
     at::Tensor mine;
     if (rank_ == 0) {
-      at::Tensor acc = t.clone();
-      for (int peer = 1; peer < size_; ++peer)
-        acc.add_(recv_tensor(peer, dtype, t.numel()));
-      if (r.red == ReduceOp::AVERAGE) acc.div_(static_cast<double>(size_));
-      for (int peer = 1; peer < size_; ++peer)
-        send_tensor(peer, acc.narrow(0, rank_off(peer) * row, rank_rows(peer) * row));
-      mine = acc.narrow(0, 0, rank_rows(0) * row);
+      at::Tensor acc;
+      bool first = true;
+      if (im_member) { acc = t.clone(); first = false; }
+      for (int peer : psi.ranks) {
+        if (peer == 0) continue;
+        at::Tensor s = first ? recv_tensor_any(peer, dtype)
+                             : recv_tensor(peer, dtype, acc.numel());
+        if (first) { acc = s; first = false; }
+        else acc.add_(s);
+      }
+      if (r.red == ReduceOp::AVERAGE) acc.div_(static_cast<double>(nmem));
+      for (int i = 0; i < nmem; ++i) {
+        int peer = psi.ranks[i];
+        if (peer == 0) continue;
+        send_tensor(peer, acc.narrow(0, member_off(i) * row, member_rows(i) * row));
+      }
+      if (im_member)
+        mine = acc.narrow(0, member_off(my_idx) * row, member_rows(my_idx) * row);
     } else {
       send_tensor(0, t.contiguous());
-      mine = recv_tensor(0, dtype, rank_rows(rank_) * row);
+      mine = recv_tensor(0, dtype, member_rows(my_idx) * row);
     }
-    auto out_sizes = in_sizes;
-    if (out_sizes.empty()) out_sizes.push_back(rank_rows(rank_));
-    else out_sizes[0] = rank_rows(rank_);
-    e.out.resize_(out_sizes);
-    flat(e.out).copy_(mine);
-    if (e.postscale != 1.0) e.out.mul_(e.postscale);
+    if (im_member) {
+      auto& e = entries[ti];
+      auto out_sizes = e.in.sizes().vec();
+      if (out_sizes.empty()) out_sizes.push_back(member_rows(my_idx));
+      else out_sizes[0] = member_rows(my_idx);
+      e.out.resize_(out_sizes);
+      flat(e.out).copy_(mine);
+      if (e.postscale != 1.0) e.out.mul_(e.postscale);
+    }
   }
   return Status::OK();
 }

@@ -24,6 +24,7 @@ void Request::serialize(Writer& w) const {
   w.u8(wire_dtype);
   w.i32(device);
   w.i32(root_rank);
+  w.i32(process_set);
   put_i64_vec(w, shape);
   put_i64_vec(w, splits);
 }
@@ -38,6 +39,7 @@ Request Request::deserialize(Reader& r) {
   q.wire_dtype = r.u8();
   q.device = r.i32();
   q.root_rank = r.i32();
+  q.process_set = r.i32();
   q.shape = get_i64_vec(r);
   q.splits = get_i64_vec(r);
   return q;
@@ -53,6 +55,7 @@ void Response::serialize(Writer& w) const {
   w.u8(wire_dtype);
   w.i32(device);
   w.i32(root_rank);
+  w.i32(process_set);
   put_i64_vec(w, numels);
   w.u32(static_cast<uint32_t>(first_dims.size()));
   for (auto& v : first_dims) put_i64_vec(w, v);
@@ -74,6 +77,7 @@ Response Response::deserialize(Reader& r) {
   s.wire_dtype = r.u8();
   s.device = r.i32();
   s.root_rank = r.i32();
+  s.process_set = r.i32();
   s.numels = get_i64_vec(r);
   uint32_t fd = r.u32();
   s.first_dims.resize(fd);

@@ -18,8 +18,9 @@ struct Request {
   uint8_t wire_dtype = 0;
   int32_t device = -1;  // -1 = CPU
   int32_t root_rank = 0;
+  int32_t process_set = 0;  // process-set id (0 = global)
   std::vector<int64_t> shape;
-  std::vector<int64_t> splits;  // alltoall: rows sent to each rank
+  std::vector<int64_t> splits;  // alltoall: rows sent to each set member
 
   void serialize(Writer& w) const;
   static Request deserialize(Reader& r);
@@ -34,6 +35,7 @@ struct Response {
   uint8_t wire_dtype = 0;
   int32_t device = -1;
   int32_t root_rank = 0;
+  int32_t process_set = 0;  // process-set id (0 = global)
   // Per-tensor total element count (fusion sizing; zero-fill for joined ranks).
   std::vector<int64_t> numels;
   // ALLGATHER: first_dims[t][r] = dim-0 extent contributed by rank r for

@@ -1,5 +1,6 @@
 #include "operations.h"
 
+#include <algorithm>
 #include <atomic>
 #include <condition_variable>
 #include <deque>
@@ -47,6 +48,11 @@ struct Global {
   Timeline tl;
   std::mutex err_mu;
   std::string fatal;
+
+  // Process-set registry: written by the user thread (collective
+  // registration), read by the background thread each cycle.
+  std::mutex ps_mu;
+  std::vector<ProcessSetInfo> psets;
 };
 
 Global g;
@@ -70,6 +76,7 @@ Request make_request(const TensorEntry& e) {
       e.wire_dtype == at::ScalarType::Undefined ? 0 : dtype_id(e.wire_dtype);
   q.device = e.device;
   q.root_rank = e.root_rank;
+  q.process_set = e.process_set;
   q.splits = e.splits;
   return q;
 }
@@ -87,6 +94,7 @@ TensorEntry make_zero_entry(const Response& r, size_t i) {
   e.in = at::zeros({r.numels.at(i)}, opts);
   e.out = e.in;
   e.device = r.device >= 0 ? g.join_device : -1;
+  e.process_set = r.process_set;
   return e;
 }
 
@@ -107,25 +115,41 @@ void execute_response(const Response& r) {
     return;
   }
 
+  ProcessSetInfo psi = ps_get(r.process_set);
+  if (!psi.valid) {
+    AXON_LOG(LogLevel::ERROR, "axon: response for unknown process set " +
+                                  std::to_string(r.process_set));
+    return;
+  }
+  const bool member = psi.member(g.rank);
+  // Non-members are uninvolved — except rank 0, which is the control-plane
+  // hub: for CPU-data ops of sets it doesn't belong to it acts as a pure
+  // relay, and for GPU ops it relays the RCCL unique-id rendezvous. The
+  // global response order is identical on every rank, so the relay executes
+  // at the same logical point as the members.
+  if (!member && g.rank != 0) return;
+
   std::vector<TensorEntry> entries;
-  entries.reserve(r.names.size());
-  for (size_t i = 0; i < r.names.size(); ++i) {
-    auto it = g.pending.find(r.names[i]);
-    if (it != g.pending.end()) {
-      entries.push_back(std::move(it->second));
-      g.pending.erase(it);
-    } else if (g.joined && r.op == OpType::ALLREDUCE) {
-      entries.push_back(make_zero_entry(r, i));
-    } else {
-      AXON_LOG(LogLevel::ERROR,
-               "axon: response for unknown tensor '" + r.names[i] + "'");
-      return;
+  if (member) {
+    entries.reserve(r.names.size());
+    for (size_t i = 0; i < r.names.size(); ++i) {
+      auto it = g.pending.find(r.names[i]);
+      if (it != g.pending.end()) {
+        entries.push_back(std::move(it->second));
+        g.pending.erase(it);
+      } else if (g.joined && r.op == OpType::ALLREDUCE) {
+        entries.push_back(make_zero_entry(r, i));
+      } else {
+        AXON_LOG(LogLevel::ERROR,
+                 "axon: response for unknown tensor '" + r.names[i] + "'");
+        return;
+      }
     }
   }
 
   if (!r.error.empty()) {
     for (auto& e : entries) complete(e, Status::Error(r.error));
-    return;
+    return;  // errors are delivered locally on every rank — nothing to relay
   }
 
   if (r.op == OpType::BARRIER) {
@@ -138,12 +162,14 @@ void execute_response(const Response& r) {
   for (auto& e : entries) g.tl.begin(e.name, r.device >= 0 ? "QUEUE_GPU" : "CPU_OP");
 
   if (r.device < 0) {
-    Status st = g.cpu->execute(r, entries);
+    Status st = g.cpu->execute(r, entries);  // member or rank-0 relay
     for (auto& e : entries) complete(e, st);
   } else {
     if (!g.gpu)
       g.gpu.reset(create_rccl_backend(g.rank, g.size, g.cp.get(), &g.tl));
     g.gpu->execute(r, std::move(entries));  // async; callbacks via poller
+    // (rank-0 non-member: execute() only relays the set's unique-id
+    // rendezvous on its first op, then returns with nothing enqueued)
   }
 }
 
@@ -301,6 +327,18 @@ void initialize(int rank_, int size_, int local_rank_, const std::string& path) 
     std::lock_guard<std::mutex> lk(g.err_mu);
     g.fatal.clear();
   }
+  {
+    std::lock_guard<std::mutex> lk(g.ps_mu);
+    g.psets.clear();
+    ProcessSetInfo global;
+    global.id = 0;
+    for (int r = 0; r < size_; ++r) {
+      global.ranks.push_back(r);
+      global.mask |= (1u << r);
+    }
+    global.my_index = rank_;
+    g.psets.push_back(std::move(global));
+  }
   std::promise<void> ready;
   auto fut = ready.get_future();
   g.running = true;
@@ -336,6 +374,50 @@ int local_rank() { return g.local_rank; }
 std::string fatal_error() {
   std::lock_guard<std::mutex> lk(g.err_mu);
   return g.fatal;
+}
+
+int ps_register(const std::vector<int>& ranks) {
+  std::vector<int> sorted = ranks;
+  std::sort(sorted.begin(), sorted.end());
+  sorted.erase(std::unique(sorted.begin(), sorted.end()), sorted.end());
+  if (sorted.empty() || sorted.front() < 0 || sorted.back() >= g.size)
+    throw std::runtime_error("axon: process-set ranks out of range");
+  ProcessSetInfo psi;
+  psi.ranks = std::move(sorted);
+  for (int r : psi.ranks) psi.mask |= (1u << r);
+  psi.my_index = psi.index_of(g.rank);
+  std::lock_guard<std::mutex> lk(g.ps_mu);
+  psi.id = static_cast<int>(g.psets.size());
+  g.psets.push_back(psi);
+  return psi.id;
+}
+
+ProcessSetInfo ps_get(int id) {
+  std::lock_guard<std::mutex> lk(g.ps_mu);
+  if (id < 0 || id >= static_cast<int>(g.psets.size())) {
+    ProcessSetInfo bad;
+    bad.id = id;
+    bad.valid = false;
+    return bad;
+  }
+  return g.psets[id];
+}
+
+bool ps_known(int id) {
+  std::lock_guard<std::mutex> lk(g.ps_mu);
+  return id >= 0 && id < static_cast<int>(g.psets.size()) && g.psets[id].valid;
+}
+
+void ps_remove(int id) {
+  std::lock_guard<std::mutex> lk(g.ps_mu);
+  if (id <= 0 || id >= static_cast<int>(g.psets.size()))
+    throw std::runtime_error("axon: cannot remove process set " + std::to_string(id));
+  g.psets[id].valid = false;
+}
+
+int ps_count() {
+  std::lock_guard<std::mutex> lk(g.ps_mu);
+  return static_cast<int>(g.psets.size());
 }
 
 void enqueue(TensorEntry entry) {

@@ -31,6 +31,16 @@ void enqueue(TensorEntry entry);
 // Non-empty when the background loop died unexpectedly (peer failure).
 std::string fatal_error();
 
+// Process-set registry (SURVEY.md §2.1 "Process sets"). Id 0 (the global
+// set) is created by initialize(). ps_register must be called collectively
+// and in identical order on every rank (the Python API enforces this with
+// an allgather of the member list); it returns the same id everywhere.
+int ps_register(const std::vector<int>& ranks);
+ProcessSetInfo ps_get(int id);  // copy; unknown id -> invalid info (valid=false, empty)
+bool ps_known(int id);
+void ps_remove(int id);  // marks invalid; ops on it then error
+int ps_count();
+
 // GPU backend factory seam (implemented in rccl_backend.cpp).
 class ControlPlane;
 class Timeline;

@@ -210,7 +210,8 @@ def test_fused_sgd_channels_last_gpu(hvd):
         o1.zero_grad()
         o2.zero_grad()
     for p, q in zip(m1.parameters(), m2.parameters()):
-        assert p.is_non_overlapping_and_dense()
+        if p.dim() == 4:  # conv weights stayed channels_last through step()
+            assert p.is_contiguous(memory_format=torch.channels_last)
         assert torch.allclose(p.contiguous(), q.contiguous(),
                               rtol=1e-4, atol=1e-5)
 
