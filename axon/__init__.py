@@ -61,6 +61,13 @@ from axon.mpi_ops import (  # noqa: F401,E402
     synchronize,
 )
 from axon.optimizer import DistributedOptimizer  # noqa: F401,E402
+from axon.process_sets import (  # noqa: F401,E402
+    ProcessSet,
+    add_process_set,
+    global_process_set,
+    process_set_ranks,
+    remove_process_set,
+)
 
 
 def nccl_built():

@@ -47,6 +47,8 @@ def init():
     local_rank = _env_int("AXON_LOCAL_RANK", "LOCAL_RANK",
                           "OMPI_COMM_WORLD_LOCAL_RANK", default=rank)
     _core.init(rank, size, local_rank, _control_path())
+    from axon.process_sets import _refresh_global
+    _refresh_global()
     if torch.cuda.is_available():
         ndev = torch.cuda.device_count()
         torch.cuda.set_device(local_rank % ndev if ndev > 1 else 0)

@@ -96,7 +96,7 @@ uint64_t submit(TensorEntry e) {
 
 uint64_t do_allreduce(at::Tensor in, at::Tensor out, const std::string& name,
                       int red, double prescale, double postscale,
-                      int wire_dtype) {
+                      int wire_dtype, int process_set) {
   TensorEntry e;
   e.name = name;
   e.op = OpType::ALLREDUCE;
@@ -107,37 +107,42 @@ uint64_t do_allreduce(at::Tensor in, at::Tensor out, const std::string& name,
   e.postscale = postscale;
   e.wire_dtype = wire_dtype < 0 ? at::ScalarType::Undefined
                                 : static_cast<at::ScalarType>(wire_dtype);
+  e.process_set = process_set;
   e.device = dev_index_of(in);
   e.ready_event = maybe_record_ready(in);
   return submit(std::move(e));
 }
 
 uint64_t do_broadcast(at::Tensor in, at::Tensor out, int root,
-                      const std::string& name) {
+                      const std::string& name, int process_set) {
   TensorEntry e;
   e.name = name;
   e.op = OpType::BROADCAST;
   e.in = in;
   e.out = out;
   e.root_rank = root;
+  e.process_set = process_set;
   e.device = dev_index_of(in);
   e.ready_event = maybe_record_ready(in);
   return submit(std::move(e));
 }
 
-uint64_t do_allgather(at::Tensor in, at::Tensor out, const std::string& name) {
+uint64_t do_allgather(at::Tensor in, at::Tensor out, const std::string& name,
+                      int process_set) {
   TensorEntry e;
   e.name = name;
   e.op = OpType::ALLGATHER;
   e.in = in;
   e.out = out;
+  e.process_set = process_set;
   e.device = dev_index_of(in);
   e.ready_event = maybe_record_ready(in);
   return submit(std::move(e));
 }
 
 uint64_t do_alltoall(at::Tensor in, at::Tensor out, std::vector<int64_t> splits,
-                     at::Tensor received_splits, const std::string& name) {
+                     at::Tensor received_splits, const std::string& name,
+                     int process_set) {
   TensorEntry e;
   e.name = name;
   e.op = OpType::ALLTOALL;
@@ -145,13 +150,15 @@ uint64_t do_alltoall(at::Tensor in, at::Tensor out, std::vector<int64_t> splits,
   e.out = out;
   e.splits = std::move(splits);
   e.received_splits = received_splits;
+  e.process_set = process_set;
   e.device = dev_index_of(in);
   e.ready_event = maybe_record_ready(in);
   return submit(std::move(e));
 }
 
 uint64_t do_reducescatter(at::Tensor in, at::Tensor out, const std::string& name,
-                          int red, double prescale, double postscale) {
+                          int red, double prescale, double postscale,
+                          int process_set) {
   TensorEntry e;
   e.name = name;
   e.op = OpType::REDUCESCATTER;
@@ -160,6 +167,7 @@ uint64_t do_reducescatter(at::Tensor in, at::Tensor out, const std::string& name
   e.out = out;
   e.prescale = prescale;
   e.postscale = postscale;
+  e.process_set = process_set;
   e.device = dev_index_of(in);
   e.ready_event = maybe_record_ready(in);
   return submit(std::move(e));
@@ -221,16 +229,27 @@ PYBIND11_MODULE(_core, m) {
 
   m.def("allreduce_async", &axon::do_allreduce, py::arg("in"), py::arg("out"),
         py::arg("name"), py::arg("red"), py::arg("prescale") = 1.0,
-        py::arg("postscale") = 1.0, py::arg("wire_dtype") = -1);
+        py::arg("postscale") = 1.0, py::arg("wire_dtype") = -1,
+        py::arg("process_set") = 0);
   m.def("broadcast_async", &axon::do_broadcast, py::arg("in"), py::arg("out"),
-        py::arg("root"), py::arg("name"));
+        py::arg("root"), py::arg("name"), py::arg("process_set") = 0);
   m.def("allgather_async", &axon::do_allgather, py::arg("in"), py::arg("out"),
-        py::arg("name"));
+        py::arg("name"), py::arg("process_set") = 0);
   m.def("alltoall_async", &axon::do_alltoall, py::arg("in"), py::arg("out"),
-        py::arg("splits"), py::arg("received_splits"), py::arg("name"));
+        py::arg("splits"), py::arg("received_splits"), py::arg("name"),
+        py::arg("process_set") = 0);
   m.def("reducescatter_async", &axon::do_reducescatter, py::arg("in"),
         py::arg("out"), py::arg("name"), py::arg("red"),
-        py::arg("prescale") = 1.0, py::arg("postscale") = 1.0);
+        py::arg("prescale") = 1.0, py::arg("postscale") = 1.0,
+        py::arg("process_set") = 0);
+  m.def("ps_register", &axon::ps_register, py::arg("ranks"));
+  m.def("ps_remove", &axon::ps_remove, py::arg("id"));
+  m.def("ps_count", &axon::ps_count);
+  m.def("ps_ranks", [](int id) {
+    auto psi = axon::ps_get(id);
+    if (!psi.valid) throw std::runtime_error("axon: unknown/removed process set");
+    return psi.ranks;
+  }, py::arg("id"));
   m.def("join_async", &axon::do_join, py::arg("device"), py::arg("out"));
   m.def("barrier_async", &axon::do_barrier, py::arg("name"));
   m.def("synchronize", &axon::do_synchronize, py::arg("handle"));

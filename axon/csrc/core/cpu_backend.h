@@ -30,6 +30,7 @@ class CpuBackend {
 
   void send_tensor(int peer, const at::Tensor& t);
   at::Tensor recv_tensor(int peer, at::ScalarType dtype, int64_t numel);
+  at::Tensor recv_tensor_any(int peer, at::ScalarType dtype);
 
   int rank_, size_;
   ControlPlane* cp_;

@@ -18,6 +18,8 @@
 #include <limits>
 #include <map>
 #include <thread>
+#include <unordered_map>
+#include <unordered_set>
 
 #include "../kernels/kernels.h"
 #include "controller.h"
@@ -78,8 +80,18 @@ class RcclBackend : public GpuBackend {
 
   void execute(const Response& r, std::vector<TensorEntry> entries) override {
     try {
-      ensure_init(entries);
-      issue(r, entries);
+      ProcessSetInfo psi = ps_get(r.process_set);
+      if (!psi.member(rank_)) {
+        // Rank-0 relay for a set we don't belong to: forward the set's
+        // unique-id rendezvous on its first GPU op, nothing else (the data
+        // plane is the set's own RCCL communicator). Deterministic: the
+        // global response order is identical on every rank.
+        relay_rendezvous(psi);
+        return;
+      }
+      ensure_device(entries);
+      SetComm& sc = ensure_comm(psi);
+      issue(r, entries, sc);
     } catch (const std::exception& e) {
       AXON_LOG(LogLevel::ERROR, std::string("axon RCCL execute failed: ") + e.what());
       for (auto& en : entries)
@@ -109,14 +121,19 @@ class RcclBackend : public GpuBackend {
     }
     cv_.notify_all();
     if (poller_.joinable()) poller_.join();
-    if (comm_) {
-      hipStreamSynchronize(stream_);
-      ncclCommDestroy(comm_);
-      comm_ = nullptr;
-    }
+    if (stream_) hipStreamSynchronize(stream_);
+    for (auto& kv : comms_)
+      if (kv.second.comm) ncclCommDestroy(kv.second.comm);
+    comms_.clear();
   }
 
  private:
+  struct SetComm {
+    ncclComm_t comm = nullptr;
+    int rank = 0;   // set-relative rank (== comm rank)
+    int size = 1;   // number of set members
+    std::vector<int> ranks;  // global ranks, set order (for root lookup)
+  };
   struct Slot {
     ChunkDesc* host = nullptr;  // pinned
     ChunkDesc* dev = nullptr;
@@ -128,32 +145,61 @@ class RcclBackend : public GpuBackend {
     std::vector<at::Tensor> keep;
   };
 
-  void ensure_init(const std::vector<TensorEntry>& entries) {
-    if (comm_) return;
+  void ensure_device(const std::vector<TensorEntry>& entries) {
+    if (device_init_) return;
+    device_init_ = true;
     device_ = entries.at(0).device;
     hip_check(hipSetDevice(device_), "hipSetDevice");
     hip_check(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking),
               "hipStreamCreate(comm)");
-    // RCCL unique id rendezvous over the control plane.
+    fuse_bytes_ = static_cast<size_t>(env_int("AXON_FUSION_THRESHOLD", 128 << 20));
+    hip_check(hipMalloc(&fusebuf_, fuse_bytes_ + kAlign), "hipMalloc(fusion buffer)");
+    hip_check(hipMalloc(&dots_mine_, 3 * sizeof(double)), "hipMalloc(dots)");
+    hip_check(hipMalloc(&dots_peer_, 3 * sizeof(double)), "hipMalloc(dots)");
+    poller_ = std::thread([this] { poll_loop(); });
+  }
+
+  // Lazy per-process-set communicator. RCCL unique id rendezvous over the
+  // control plane: the lowest set member generates the
+  // id. All worker channels go through rank 0, so when rank 0 is not in
+  // the set it relays the id (relay_rendezvous, called at the same point
+  // in the global response order).
+  SetComm& ensure_comm(const ProcessSetInfo& psi) {
+    auto it = comms_.find(psi.id);
+    if (it != comms_.end()) return it->second;
+    SetComm sc;
+    sc.rank = psi.my_index;
+    sc.size = psi.size();
+    sc.ranks = psi.ranks;
     ncclUniqueId id;
-    if (rank_ == 0) {
+    const int root = psi.ranks.front();  // == 0 iff rank 0 is a member
+    if (rank_ == root) {
       nccl_check(ncclGetUniqueId(&id), "ncclGetUniqueId");
       Writer w;
       w.raw(&id, sizeof(id));
-      for (int p = 1; p < size_; ++p) cp_->send(p, MsgType::RENDEZVOUS, w.buf);
+      if (root == 0) {
+        for (int p : psi.ranks)
+          if (p != 0) cp_->send(p, MsgType::RENDEZVOUS, w.buf);
+      } else {
+        cp_->send(0, MsgType::RENDEZVOUS, w.buf);  // hub relays to members
+      }
     } else {
       auto payload = cp_->recv_data(0);
       if (payload.size() != sizeof(id))
         throw std::runtime_error("axon: bad rccl unique id frame");
       std::memcpy(&id, payload.data(), sizeof(id));
     }
-    nccl_check(ncclCommInitRank(&comm_, size_, id, rank_), "ncclCommInitRank");
+    nccl_check(ncclCommInitRank(&sc.comm, sc.size, id, sc.rank),
+               "ncclCommInitRank");
+    return comms_.emplace(psi.id, sc).first->second;
+  }
 
-    fuse_bytes_ = static_cast<size_t>(env_int("AXON_FUSION_THRESHOLD", 128 << 20));
-    hip_check(hipMalloc(&fusebuf_, fuse_bytes_ + kAlign), "hipMalloc(fusion buffer)");
-    hip_check(hipMalloc(&dots_mine_, 3 * sizeof(double)), "hipMalloc(dots)");
-    hip_check(hipMalloc(&dots_peer_, 3 * sizeof(double)), "hipMalloc(dots)");
-    poller_ = std::thread([this] { poll_loop(); });
+  void relay_rendezvous(const ProcessSetInfo& psi) {
+    if (relayed_sets_.count(psi.id)) return;
+    relayed_sets_.insert(psi.id);
+    auto payload = cp_->recv_data(psi.ranks.front());
+    for (int p : psi.ranks)
+      if (p != psi.ranks.front()) cp_->send(p, MsgType::RENDEZVOUS, payload);
   }
 
   int acquire_slot() {
@@ -212,19 +258,19 @@ class RcclBackend : public GpuBackend {
     }
   }
 
-  void issue(const Response& r, std::vector<TensorEntry>& entries) {
+  void issue(const Response& r, std::vector<TensorEntry>& entries, SetComm& sc) {
     hip_check(hipSetDevice(device_), "hipSetDevice");
     wait_ready_events(entries);
     for (auto& e : entries) tl_->begin(e.name, "RCCL");
     switch (r.op) {
       case OpType::ALLREDUCE:
-        if (r.red == ReduceOp::ADASUM) adasum(entries);
-        else allreduce(r, entries);
+        if (r.red == ReduceOp::ADASUM) adasum(entries, sc);
+        else allreduce(r, entries, sc);
         break;
-      case OpType::BROADCAST: broadcast(r, entries); break;
-      case OpType::ALLGATHER: allgather(r, entries); break;
-      case OpType::ALLTOALL: alltoall(r, entries); break;
-      case OpType::REDUCESCATTER: reducescatter(r, entries); break;
+      case OpType::BROADCAST: broadcast(r, entries, sc); break;
+      case OpType::ALLGATHER: allgather(r, entries, sc); break;
+      case OpType::ALLTOALL: alltoall(r, entries, sc); break;
+      case OpType::REDUCESCATTER: reducescatter(r, entries, sc); break;
       default:
         throw std::runtime_error("axon RCCL: unsupported op");
     }
@@ -241,7 +287,7 @@ class RcclBackend : public GpuBackend {
     }
   }
 
-  void allreduce(const Response& r, std::vector<TensorEntry>& entries) {
+  void allreduce(const Response& r, std::vector<TensorEntry>& entries, SetComm& sc) {
     at::ScalarType dtype = id_dtype(r.dtype);
     at::ScalarType wire = r.wire_dtype ? id_dtype(r.wire_dtype) : dtype;
     bool any_scale = false;
@@ -254,7 +300,7 @@ class RcclBackend : public GpuBackend {
       // buffer across xGMI links itself.
       auto& e = entries[0];
       nccl_check(ncclAllReduce(e.in.data_ptr(), e.out.data_ptr(), e.in.numel(),
-                               nccl_dtype(dtype), op, comm_, stream_),
+                               nccl_dtype(dtype), op, sc.comm, stream_),
                  "ncclAllReduce");
       return;
     }
@@ -300,7 +346,7 @@ class RcclBackend : public GpuBackend {
     }
 
     int64_t wire_elems = static_cast<int64_t>(total / wsize);
-    nccl_check(ncclAllReduce(buf, buf, wire_elems, nccl_dtype(wire), op, comm_,
+    nccl_check(ncclAllReduce(buf, buf, wire_elems, nccl_dtype(wire), op, sc.comm,
                              stream_),
                "ncclAllReduce(fused)");
 
@@ -330,17 +376,21 @@ class RcclBackend : public GpuBackend {
     }
   }
 
-  void broadcast(const Response& r, std::vector<TensorEntry>& entries) {
+  void broadcast(const Response& r, std::vector<TensorEntry>& entries, SetComm& sc) {
+    // r.root_rank is a GLOBAL rank; RCCL wants the comm-relative root.
+    int root = 0;
+    for (int i = 0; i < sc.size; ++i)
+      if (sc.ranks[i] == r.root_rank) { root = i; break; }
     for (auto& e : entries) {
       nccl_check(ncclBroadcast(e.in.defined() ? e.in.data_ptr() : e.out.data_ptr(),
                                e.out.data_ptr(), e.out.numel(),
-                               nccl_dtype(e.out.scalar_type()), r.root_rank,
-                               comm_, stream_),
+                               nccl_dtype(e.out.scalar_type()), root,
+                               sc.comm, stream_),
                  "ncclBroadcast");
     }
   }
 
-  void allgather(const Response& r, std::vector<TensorEntry>& entries) {
+  void allgather(const Response& r, std::vector<TensorEntry>& entries, SetComm& sc) {
     for (size_t t = 0; t < entries.size(); ++t) {
       auto& e = entries[t];
       const auto& fd = r.first_dims.at(t);
@@ -361,29 +411,29 @@ class RcclBackend : public GpuBackend {
       auto dt = nccl_dtype(e.in.scalar_type());
       char* ob = static_cast<char*>(e.out.data_ptr());
       if (equal && fd[0] > 0) {
-        nccl_check(ncclAllGather(e.in.data_ptr(), ob, fd[0] * row, dt, comm_,
+        nccl_check(ncclAllGather(e.in.data_ptr(), ob, fd[0] * row, dt, sc.comm,
                                  stream_),
                    "ncclAllGather");
       } else {
         // Variable first dim: direct p2p exchange over the 7 xGMI links
         // (SURVEY.md §2.4 N3).
-        int64_t my_numel = fd[rank_] * row;
-        std::vector<int64_t> off(size_, 0);
-        for (int p = 1; p < size_; ++p) off[p] = off[p - 1] + fd[p - 1] * row;
+        int64_t my_numel = fd[sc.rank] * row;
+        std::vector<int64_t> off(sc.size, 0);
+        for (int p = 1; p < sc.size; ++p) off[p] = off[p - 1] + fd[p - 1] * row;
         nccl_check(ncclGroupStart(), "ncclGroupStart");
-        for (int p = 0; p < size_; ++p) {
-          if (p == rank_) continue;
+        for (int p = 0; p < sc.size; ++p) {
+          if (p == sc.rank) continue;
           if (my_numel > 0)
-            nccl_check(ncclSend(e.in.data_ptr(), my_numel, dt, p, comm_, stream_),
+            nccl_check(ncclSend(e.in.data_ptr(), my_numel, dt, p, sc.comm, stream_),
                        "ncclSend");
           int64_t pn = fd[p] * row;
           if (pn > 0)
-            nccl_check(ncclRecv(ob + off[p] * esize, pn, dt, p, comm_, stream_),
+            nccl_check(ncclRecv(ob + off[p] * esize, pn, dt, p, sc.comm, stream_),
                        "ncclRecv");
         }
         nccl_check(ncclGroupEnd(), "ncclGroupEnd");
         if (my_numel > 0)
-          hip_check(hipMemcpyAsync(ob + off[rank_] * esize, e.in.data_ptr(),
+          hip_check(hipMemcpyAsync(ob + off[sc.rank] * esize, e.in.data_ptr(),
                                    my_numel * esize, hipMemcpyDeviceToDevice,
                                    stream_),
                     "hipMemcpyAsync(self slab)");
@@ -391,16 +441,16 @@ class RcclBackend : public GpuBackend {
     }
   }
 
-  void alltoall(const Response& r, std::vector<TensorEntry>& entries) {
+  void alltoall(const Response& r, std::vector<TensorEntry>& entries, SetComm& sc) {
     auto m = [&](int src, int dst) {
-      return r.splits_matrix.at(static_cast<size_t>(src) * size_ + dst);
+      return r.splits_matrix.at(static_cast<size_t>(src) * sc.size + dst);
     };
     for (auto& e : entries) {
       auto in_sizes = e.in.sizes().vec();
       int64_t row = 1;
       for (size_t d = 1; d < in_sizes.size(); ++d) row *= in_sizes[d];
       int64_t rows_me = 0;
-      for (int src = 0; src < size_; ++src) rows_me += m(src, rank_);
+      for (int src = 0; src < sc.size; ++src) rows_me += m(src, sc.rank);
       auto out_sizes = in_sizes;
       if (out_sizes.empty()) out_sizes.push_back(rows_me);
       else out_sizes[0] = rows_me;
@@ -409,44 +459,44 @@ class RcclBackend : public GpuBackend {
       auto dt = nccl_dtype(e.in.scalar_type());
       const char* ib = static_cast<const char*>(e.in.data_ptr());
       char* ob = static_cast<char*>(e.out.data_ptr());
-      std::vector<int64_t> soff(size_, 0), roff(size_, 0);
-      for (int p = 1; p < size_; ++p) {
-        soff[p] = soff[p - 1] + m(rank_, p - 1) * row;
-        roff[p] = roff[p - 1] + m(p - 1, rank_) * row;
+      std::vector<int64_t> soff(sc.size, 0), roff(sc.size, 0);
+      for (int p = 1; p < sc.size; ++p) {
+        soff[p] = soff[p - 1] + m(sc.rank, p - 1) * row;
+        roff[p] = roff[p - 1] + m(p - 1, sc.rank) * row;
       }
       nccl_check(ncclGroupStart(), "ncclGroupStart");
-      for (int p = 0; p < size_; ++p) {
-        if (p == rank_) continue;
-        int64_t sn = m(rank_, p) * row, rn = m(p, rank_) * row;
+      for (int p = 0; p < sc.size; ++p) {
+        if (p == sc.rank) continue;
+        int64_t sn = m(sc.rank, p) * row, rn = m(p, sc.rank) * row;
         if (sn > 0)
-          nccl_check(ncclSend(ib + soff[p] * esize, sn, dt, p, comm_, stream_),
+          nccl_check(ncclSend(ib + soff[p] * esize, sn, dt, p, sc.comm, stream_),
                      "ncclSend");
         if (rn > 0)
-          nccl_check(ncclRecv(ob + roff[p] * esize, rn, dt, p, comm_, stream_),
+          nccl_check(ncclRecv(ob + roff[p] * esize, rn, dt, p, sc.comm, stream_),
                      "ncclRecv");
       }
       nccl_check(ncclGroupEnd(), "ncclGroupEnd");
-      int64_t self_n = m(rank_, rank_) * row;
+      int64_t self_n = m(sc.rank, sc.rank) * row;
       if (self_n > 0)
-        hip_check(hipMemcpyAsync(ob + roff[rank_] * esize, ib + soff[rank_] * esize,
+        hip_check(hipMemcpyAsync(ob + roff[sc.rank] * esize, ib + soff[sc.rank] * esize,
                                  self_n * esize, hipMemcpyDeviceToDevice, stream_),
                   "hipMemcpyAsync(self)");
       if (e.received_splits.defined()) {
         auto rs = e.received_splits;
-        rs.resize_({size_});
-        for (int src = 0; src < size_; ++src) rs[src] = m(src, rank_);
+        rs.resize_({sc.size});
+        for (int src = 0; src < sc.size; ++src) rs[src] = m(src, sc.rank);
       }
     }
   }
 
-  void reducescatter(const Response& r, std::vector<TensorEntry>& entries) {
+  void reducescatter(const Response& r, std::vector<TensorEntry>& entries, SetComm& sc) {
     for (auto& e : entries) {
       auto in_sizes = e.in.sizes().vec();
       int64_t rows = in_sizes.empty() ? 1 : in_sizes[0];
       int64_t row = 1;
       for (size_t d = 1; d < in_sizes.size(); ++d) row *= in_sizes[d];
-      int64_t base = rows / size_, rem = rows % size_;
-      int64_t my_rows = base + (rank_ < rem ? 1 : 0);
+      int64_t base = rows / sc.size, rem = rows % sc.size;
+      int64_t my_rows = base + (sc.rank < rem ? 1 : 0);
       auto out_sizes = in_sizes;
       if (out_sizes.empty()) out_sizes.push_back(my_rows);
       else out_sizes[0] = my_rows;
@@ -455,7 +505,7 @@ class RcclBackend : public GpuBackend {
       ncclRedOp_t op = nccl_red(r.red);
       if (rem == 0) {
         nccl_check(ncclReduceScatter(e.in.data_ptr(), e.out.data_ptr(),
-                                     base * row, dt, op, comm_, stream_),
+                                     base * row, dt, op, sc.comm, stream_),
                    "ncclReduceScatter");
       } else {
         // Uneven split: allreduce into scratch, slice locally. Correct but
@@ -463,9 +513,9 @@ class RcclBackend : public GpuBackend {
         size_t esize = e.in.element_size();
         char* buf = static_cast<char*>(ensure_scratch(e.in.numel() * esize));
         nccl_check(ncclAllReduce(e.in.data_ptr(), buf, e.in.numel(), dt, op,
-                                 comm_, stream_),
+                                 sc.comm, stream_),
                    "ncclAllReduce(rs-uneven)");
-        int64_t off = rank_ * base + std::min<int64_t>(rank_, rem);
+        int64_t off = sc.rank * base + std::min<int64_t>(sc.rank, rem);
         hip_check(hipMemcpyAsync(e.out.data_ptr(), buf + off * row * esize,
                                  my_rows * row * esize, hipMemcpyDeviceToDevice,
                                  stream_),
@@ -478,9 +528,9 @@ class RcclBackend : public GpuBackend {
   // full-vector dot triples reduced over the 2^(k+1)-rank group by a
   // butterfly of 24-byte p2p exchanges (SURVEY.md §2.1 "Adasum ops"; math
   // parity with cpu_backend.cpp's fp64 oracle).
-  void adasum(std::vector<TensorEntry>& entries) {
-    if ((size_ & (size_ - 1)) != 0)
-      throw std::runtime_error("adasum requires a power-of-two world size");
+  void adasum(std::vector<TensorEntry>& entries, SetComm& sc) {
+    if ((sc.size & (sc.size - 1)) != 0)
+      throw std::runtime_error("adasum requires a power-of-two process-set size");
     for (auto& e : entries) {
       int dtype = static_cast<int>(e.in.scalar_type());
       size_t esize = e.in.element_size();
@@ -502,9 +552,9 @@ class RcclBackend : public GpuBackend {
       auto dt = nccl_dtype(e.in.scalar_type());
 
       int64_t off = 0, len = n;
-      for (int stride = 1; stride < size_; stride <<= 1) {
-        int peer = rank_ ^ stride;
-        bool a_side = (rank_ & stride) == 0;
+      for (int stride = 1; stride < sc.size; stride <<= 1) {
+        int peer = sc.rank ^ stride;
+        bool a_side = (sc.rank & stride) == 0;
         int64_t len1 = (len + 1) / 2, len2 = len - len1;
         int64_t keep_off = a_side ? off : off + len1;
         int64_t keep_len = a_side ? len1 : len2;
@@ -513,11 +563,11 @@ class RcclBackend : public GpuBackend {
         // 1) swap halves with the level peer.
         nccl_check(ncclGroupStart(), "ncclGroupStart");
         if (send_len > 0)
-          nccl_check(ncclSend(b + send_off * esize, send_len, dt, peer, comm_,
+          nccl_check(ncclSend(b + send_off * esize, send_len, dt, peer, sc.comm,
                               stream_),
                      "ncclSend(adasum half)");
         if (keep_len > 0)
-          nccl_check(ncclRecv(scratch, keep_len, dt, peer, comm_, stream_),
+          nccl_check(ncclRecv(scratch, keep_len, dt, peer, sc.comm, stream_),
                      "ncclRecv(adasum half)");
         nccl_check(ncclGroupEnd(), "ncclGroupEnd");
         // 2) local partial dot triple (canonical a/b order).
@@ -528,11 +578,11 @@ class RcclBackend : public GpuBackend {
                              dots_mine_, a_side, stream_);
         // 3) butterfly-sum the triple across the 2*stride-rank group.
         for (int j = 1; j <= stride; j <<= 1) {
-          int dpeer = rank_ ^ j;
+          int dpeer = sc.rank ^ j;
           nccl_check(ncclGroupStart(), "ncclGroupStart");
-          nccl_check(ncclSend(dots_mine_, 3, ncclFloat64, dpeer, comm_, stream_),
+          nccl_check(ncclSend(dots_mine_, 3, ncclFloat64, dpeer, sc.comm, stream_),
                      "ncclSend(dots)");
-          nccl_check(ncclRecv(dots_peer_, 3, ncclFloat64, dpeer, comm_, stream_),
+          nccl_check(ncclRecv(dots_peer_, 3, ncclFloat64, dpeer, sc.comm, stream_),
                      "ncclRecv(dots)");
           nccl_check(ncclGroupEnd(), "ncclGroupEnd");
           launch_add3(dots_mine_, dots_peer_, stream_);
@@ -549,7 +599,7 @@ class RcclBackend : public GpuBackend {
       // rank's offsets) via direct p2p.
       auto segment_of = [&](int rk) {
         int64_t o = 0, l = n;
-        for (int stride = 1; stride < size_; stride <<= 1) {
+        for (int stride = 1; stride < sc.size; stride <<= 1) {
           int64_t l1 = (l + 1) / 2;
           if ((rk & stride) == 0) l = l1;
           else { o += l1; l -= l1; }
@@ -557,14 +607,14 @@ class RcclBackend : public GpuBackend {
         return std::make_pair(o, l);
       };
       nccl_check(ncclGroupStart(), "ncclGroupStart");
-      for (int p = 0; p < size_; ++p) {
-        if (p == rank_) continue;
+      for (int p = 0; p < sc.size; ++p) {
+        if (p == sc.rank) continue;
         if (len > 0)
-          nccl_check(ncclSend(b + off * esize, len, dt, p, comm_, stream_),
+          nccl_check(ncclSend(b + off * esize, len, dt, p, sc.comm, stream_),
                      "ncclSend(adasum gather)");
         auto [po, pl] = segment_of(p);
         if (pl > 0)
-          nccl_check(ncclRecv(b + po * esize, pl, dt, p, comm_, stream_),
+          nccl_check(ncclRecv(b + po * esize, pl, dt, p, sc.comm, stream_),
                      "ncclRecv(adasum gather)");
       }
       nccl_check(ncclGroupEnd(), "ncclGroupEnd");
@@ -619,7 +669,9 @@ class RcclBackend : public GpuBackend {
   ControlPlane* cp_;
   Timeline* tl_;
   int device_ = 0;
-  ncclComm_t comm_ = nullptr;
+  bool device_init_ = false;
+  std::unordered_map<int, SetComm> comms_;      // by process-set id
+  std::unordered_set<int> relayed_sets_;        // rank-0 relay bookkeeping
   hipStream_t stream_ = nullptr;
   void* fusebuf_ = nullptr;
   size_t fuse_bytes_ = 0;

@@ -15,9 +15,10 @@ from axon import mpi_ops
 from axon.basics import rank
 
 
-def broadcast_parameters(params, root_rank=0):
+def broadcast_parameters(params, root_rank=0, process_set=None):
     """Broadcast a state_dict or list of (name, tensor) pairs from
-    root_rank to all ranks (rank-0 weights win at startup)."""
+    root_rank to all ranks (rank-0 weights win at startup). With a
+    process_set, broadcasts within the set only (call from members only)."""
     if isinstance(params, dict):
         items = sorted(params.items())
     else:
@@ -29,12 +30,13 @@ def broadcast_parameters(params, root_rank=0):
         if not torch.is_tensor(t):
             continue  # non-tensor state_dict entries (e.g. num_batches_tracked ints are tensors; skip others)
         handles.append(mpi_ops.broadcast_async_(t.data, root_rank,
-                                                name=f"bparam.{name}"))
+                                                name=f"bparam.{name}",
+                                                process_set=process_set))
     for h in handles:
         mpi_ops.synchronize(h)
 
 
-def broadcast_object(obj, root_rank=0, name=None):
+def broadcast_object(obj, root_rank=0, name=None, process_set=None):
     """Broadcast an arbitrary picklable object (cloudpickle -> uint8 tensor
     -> broadcast)."""
     name = name or "broadcast_object"
@@ -43,26 +45,30 @@ def broadcast_object(obj, root_rank=0, name=None):
         sz = torch.tensor([len(buf)], dtype=torch.int64)
     else:
         sz = torch.zeros(1, dtype=torch.int64)
-    mpi_ops.broadcast_(sz, root_rank, name=f"{name}.size")
+    mpi_ops.broadcast_(sz, root_rank, name=f"{name}.size",
+                       process_set=process_set)
     n = int(sz.item())
     if rank() == root_rank:
         data = torch.frombuffer(bytearray(buf), dtype=torch.uint8).clone()
     else:
         data = torch.empty(n, dtype=torch.uint8)
-    mpi_ops.broadcast_(data, root_rank, name=f"{name}.data")
+    mpi_ops.broadcast_(data, root_rank, name=f"{name}.data",
+                       process_set=process_set)
     if rank() == root_rank:
         return obj
     return cloudpickle.loads(data.numpy().tobytes())
 
 
-def allgather_object(obj, name=None):
-    """Gather an arbitrary picklable object from every rank; returns the
-    list indexed by rank."""
+def allgather_object(obj, name=None, process_set=None):
+    """Gather an arbitrary picklable object from every rank (or set
+    member); returns the list indexed by (set-relative) rank."""
     name = name or "allgather_object"
     buf = cloudpickle.dumps(obj)
     data = torch.frombuffer(bytearray(buf), dtype=torch.uint8).clone()
-    gathered = mpi_ops.allgather(data, name=f"{name}.data")
-    sizes = mpi_ops.allgather(torch.tensor([data.numel()]), name=f"{name}.size")
+    gathered = mpi_ops.allgather(data, name=f"{name}.data",
+                                 process_set=process_set)
+    sizes = mpi_ops.allgather(torch.tensor([data.numel()]),
+                              name=f"{name}.size", process_set=process_set)
     out, off = [], 0
     for s in sizes.tolist():
         out.append(cloudpickle.loads(gathered[off:off + s].numpy().tobytes()))
@@ -111,7 +117,7 @@ def _rebuild_tensors(obj, device_hint=None):
     return obj
 
 
-def broadcast_optimizer_state(optimizer, root_rank=0):
+def broadcast_optimizer_state(optimizer, root_rank=0, process_set=None):
     """Broadcast optimizer state from root so all ranks start identically
     (SURVEY.md §3.3): the state-dict *structure* (param_groups, scalar
     entries) goes via broadcast_object; every tensor entry is then
@@ -124,7 +130,8 @@ def broadcast_optimizer_state(optimizer, root_rank=0):
 
     sd = optimizer.state_dict()
     meta = _strip_tensors(sd)
-    meta = broadcast_object(meta, root_rank, name="opt_state.meta")
+    meta = broadcast_object(meta, root_rank, name="opt_state.meta",
+                            process_set=process_set)
 
     if rank() != root_rank:
         # Map state ids -> parameter device so rebuilt tensors land on the
@@ -143,7 +150,8 @@ def broadcast_optimizer_state(optimizer, root_rank=0):
     handles = []
     for path, t in _walk_tensors(sd.get("state", {})):
         nm = "opt_state." + ".".join(str(x) for x in path)
-        handles.append(mpi_ops.broadcast_async_(t, root_rank, name=nm))
+        handles.append(mpi_ops.broadcast_async_(t, root_rank, name=nm,
+                                                process_set=process_set))
     for h in handles:
         mpi_ops.synchronize(h)
 

@@ -10,6 +10,7 @@ import threading
 import torch
 
 from axon.basics import _ensure_init, rank, size
+from axon.process_sets import resolve_id
 
 
 class ReduceOp:
@@ -34,15 +35,19 @@ _counters = {}
 _outputs = {}  # handle -> result payload returned by synchronize
 
 
-def _auto_name(kind, name):
+def _auto_name(kind, name, psid=0):
     if name is not None:
-        return f"{kind}.{name}"
-    # Deterministic per-op counter: every rank enqueues ops in the same
-    # program order, so the counter-derived names agree across ranks.
-    with _lock:
-        n = _counters.get(kind, 0)
-        _counters[kind] = n + 1
-    return f"{kind}.noname.{n}"
+        base = f"{kind}.{name}"
+    else:
+        # Deterministic per-op counter: every rank enqueues ops in the same
+        # program order, so the counter-derived names agree across ranks.
+        with _lock:
+            n = _counters.get(kind, 0)
+            _counters[kind] = n + 1
+        base = f"{kind}.noname.{n}"
+    # Scope names per process set so the same user name can be in flight on
+    # different sets without colliding in the controller table.
+    return base if psid == 0 else f"ps{psid}:{base}"
 
 
 def _register(handle, payload):
@@ -85,55 +90,58 @@ def _resolve_op(op, average):
 
 def allreduce_async(tensor, average=None, name=None, op=None,
                     prescale_factor=1.0, postscale_factor=1.0,
-                    compression=None):
+                    compression=None, process_set=None):
     """Out-of-place async allreduce; returns a handle."""
     red = _resolve_op(op, average)
+    psid = resolve_id(process_set)
     t = tensor if _dense(tensor) else tensor.contiguous()
     output = torch.empty_like(t)
     core = _ensure_init()
-    h = core.allreduce_async(t, output, _auto_name("allreduce", name), red,
-                             prescale_factor, postscale_factor,
-                             _wire(compression))
+    h = core.allreduce_async(t, output, _auto_name("allreduce", name, psid),
+                             red, prescale_factor, postscale_factor,
+                             _wire(compression), psid)
     return _register(h, output)
 
 
 def allreduce(tensor, average=None, name=None, op=None, prescale_factor=1.0,
-              postscale_factor=1.0, compression=None):
+              postscale_factor=1.0, compression=None, process_set=None):
     return synchronize(allreduce_async(tensor, average, name, op,
                                        prescale_factor, postscale_factor,
-                                       compression))
+                                       compression, process_set))
 
 
 def allreduce_async_(tensor, average=None, name=None, op=None,
                      prescale_factor=1.0, postscale_factor=1.0,
-                     compression=None):
+                     compression=None, process_set=None):
     """In-place async allreduce (the hot path for gradients)."""
     if not _dense(tensor):
         raise ValueError("allreduce_ requires a dense tensor")
     red = _resolve_op(op, average)
+    psid = resolve_id(process_set)
     core = _ensure_init()
-    h = core.allreduce_async(tensor, tensor, _auto_name("allreduce", name), red,
+    h = core.allreduce_async(tensor, tensor,
+                             _auto_name("allreduce", name, psid), red,
                              prescale_factor, postscale_factor,
-                             _wire(compression))
+                             _wire(compression), psid)
     return _register(h, tensor)
 
 
 def allreduce_(tensor, average=None, name=None, op=None, prescale_factor=1.0,
-               postscale_factor=1.0, compression=None):
+               postscale_factor=1.0, compression=None, process_set=None):
     return synchronize(allreduce_async_(tensor, average, name, op,
                                         prescale_factor, postscale_factor,
-                                        compression))
+                                        compression, process_set))
 
 
 def grouped_allreduce_async(tensors, average=None, name=None, op=None,
                             prescale_factor=1.0, postscale_factor=1.0,
-                            compression=None):
+                            compression=None, process_set=None):
     """Allreduce a list of tensors as one logical group. The group is
     negotiated in a single cycle, so the controller fuses it into as few
     collectives as the fusion threshold allows."""
     gname = _auto_name("grouped", name)
     return [allreduce_async(t, average, f"{gname}.{i}", op, prescale_factor,
-                            postscale_factor, compression)
+                            postscale_factor, compression, process_set)
             for i, t in enumerate(tensors)]
 
 
@@ -143,10 +151,10 @@ def grouped_allreduce(tensors, **kw):
 
 def grouped_allreduce_async_(tensors, average=None, name=None, op=None,
                              prescale_factor=1.0, postscale_factor=1.0,
-                             compression=None):
+                             compression=None, process_set=None):
     gname = _auto_name("grouped", name)
     return [allreduce_async_(t, average, f"{gname}.{i}", op, prescale_factor,
-                             postscale_factor, compression)
+                             postscale_factor, compression, process_set)
             for i, t in enumerate(tensors)]
 
 
@@ -156,50 +164,56 @@ def grouped_allreduce_(tensors, **kw):
 
 # --------------------------------------------------------------- broadcast
 
-def broadcast_async(tensor, root_rank, name=None):
+def broadcast_async(tensor, root_rank, name=None, process_set=None):
+    psid = resolve_id(process_set)
     t = tensor if _dense(tensor) else tensor.contiguous()
     output = torch.empty_like(t)
     core = _ensure_init()
-    h = core.broadcast_async(t, output, root_rank, _auto_name("broadcast", name))
+    h = core.broadcast_async(t, output, root_rank,
+                             _auto_name("broadcast", name, psid), psid)
     return _register(h, output)
 
 
-def broadcast(tensor, root_rank, name=None):
-    return synchronize(broadcast_async(tensor, root_rank, name))
+def broadcast(tensor, root_rank, name=None, process_set=None):
+    return synchronize(broadcast_async(tensor, root_rank, name, process_set))
 
 
-def broadcast_async_(tensor, root_rank, name=None):
+def broadcast_async_(tensor, root_rank, name=None, process_set=None):
     if not _dense(tensor):
         raise ValueError("broadcast_ requires a dense tensor")
+    psid = resolve_id(process_set)
     core = _ensure_init()
     h = core.broadcast_async(tensor, tensor, root_rank,
-                             _auto_name("broadcast", name))
+                             _auto_name("broadcast", name, psid), psid)
     return _register(h, tensor)
 
 
-def broadcast_(tensor, root_rank, name=None):
-    return synchronize(broadcast_async_(tensor, root_rank, name))
+def broadcast_(tensor, root_rank, name=None, process_set=None):
+    return synchronize(broadcast_async_(tensor, root_rank, name, process_set))
 
 
 # --------------------------------------------------------------- allgather
 
-def allgather_async(tensor, name=None):
+def allgather_async(tensor, name=None, process_set=None):
+    psid = resolve_id(process_set)
     t = tensor.contiguous()
     output = t.new_empty(0)
     core = _ensure_init()
-    h = core.allgather_async(t, output, _auto_name("allgather", name))
+    h = core.allgather_async(t, output, _auto_name("allgather", name, psid),
+                             psid)
     return _register(h, output)
 
 
-def allgather(tensor, name=None):
-    return synchronize(allgather_async(tensor, name))
+def allgather(tensor, name=None, process_set=None):
+    return synchronize(allgather_async(tensor, name, process_set))
 
 
 # ---------------------------------------------------------------- alltoall
 
-def alltoall_async(tensor, splits=None, name=None):
+def alltoall_async(tensor, splits=None, name=None, process_set=None):
+    psid = resolve_id(process_set)
     t = tensor.contiguous()
-    n = size()
+    n = size() if psid == 0 else len(_ensure_init().ps_ranks(psid))
     if splits is None:
         rows = t.shape[0] if t.dim() > 0 else t.numel()
         if rows % n != 0:
@@ -212,12 +226,13 @@ def alltoall_async(tensor, splits=None, name=None):
     received_splits = torch.empty(n, dtype=torch.int64)
     core = _ensure_init()
     h = core.alltoall_async(t, output, splits_list, received_splits,
-                            _auto_name("alltoall", name))
+                            _auto_name("alltoall", name, psid), psid)
     return _register(h, (output, received_splits))
 
 
-def alltoall(tensor, splits=None, name=None):
-    out, received = synchronize(alltoall_async(tensor, splits, name))
+def alltoall(tensor, splits=None, name=None, process_set=None):
+    out, received = synchronize(alltoall_async(tensor, splits, name,
+                                               process_set))
     if splits is None:
         return out
     return out, received
@@ -226,21 +241,24 @@ def alltoall(tensor, splits=None, name=None):
 # ----------------------------------------------------------- reducescatter
 
 def reducescatter_async(tensor, op=ReduceOp.Average, name=None,
-                        prescale_factor=1.0, postscale_factor=1.0):
+                        prescale_factor=1.0, postscale_factor=1.0,
+                        process_set=None):
     if op not in (ReduceOp.Average, ReduceOp.Sum):
         raise ValueError("reducescatter supports Average and Sum")
+    psid = resolve_id(process_set)
     t = tensor.contiguous()
     output = t.new_empty(0)
     core = _ensure_init()
-    h = core.reducescatter_async(t, output, _auto_name("reducescatter", name),
-                                 op, prescale_factor, postscale_factor)
+    h = core.reducescatter_async(t, output,
+                                 _auto_name("reducescatter", name, psid),
+                                 op, prescale_factor, postscale_factor, psid)
     return _register(h, output)
 
 
 def reducescatter(tensor, op=ReduceOp.Average, name=None, prescale_factor=1.0,
-                  postscale_factor=1.0):
+                  postscale_factor=1.0, process_set=None):
     return synchronize(reducescatter_async(tensor, op, name, prescale_factor,
-                                           postscale_factor))
+                                           postscale_factor, process_set))
 
 
 # ------------------------------------------------------------ join/barrier

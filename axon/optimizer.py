@@ -20,11 +20,15 @@ from axon.mpi_ops import ReduceOp
 
 class _DistributedOptimizer:
     def _distributed_init(self, named_parameters, compression,
-                          backward_passes_per_step, op, gradient_predivide_factor):
+                          backward_passes_per_step, op,
+                          gradient_predivide_factor, process_set):
         self._compression = compression
         self._bpps = backward_passes_per_step
         self._op = op
         self._predivide = gradient_predivide_factor
+        self._process_set = process_set
+        self._ps_size = (mpi_ops.size() if process_set is None
+                         else len(process_set.ranks))
 
         if named_parameters is not None:
             name_of = {id(p): n for n, p in list(named_parameters)}
@@ -42,7 +46,8 @@ class _DistributedOptimizer:
         self._passes = {id(p): 0 for p in self._dist_params}
         self._hook_handles = []
         self._require_sync = True
-        if mpi_ops.size() > 1:
+        if self._ps_size > 1 and (self._process_set is None or
+                                  self._process_set.included()):
             self._register_hooks()
 
     def _register_hooks(self):
@@ -66,11 +71,14 @@ class _DistributedOptimizer:
             handle = mpi_ops.allreduce_async_(
                 p.grad, op=ReduceOp.Sum, name=name,
                 prescale_factor=1.0 / self._predivide,
-                postscale_factor=self._predivide / mpi_ops.size(),
-                compression=self._compression)
+                postscale_factor=self._predivide / self._ps_size,
+                compression=self._compression,
+                process_set=self._process_set)
         else:
             handle = mpi_ops.allreduce_async_(
-                p.grad, op=self._op, name=name, compression=self._compression)
+                p.grad, op=self._op, name=name,
+                compression=self._compression,
+                process_set=self._process_set)
         self._handles[id(p)] = handle
 
     def synchronize(self):
@@ -91,7 +99,7 @@ class _DistributedOptimizer:
             self._require_sync = prev
 
     def step(self, closure=None):
-        if self._require_sync and mpi_ops.size() > 1:
+        if self._require_sync and self._ps_size > 1:
             self.synchronize()
         return super().step(closure)
 
@@ -106,13 +114,13 @@ class _DistributedOptimizer:
 def DistributedOptimizer(optimizer, named_parameters=None,
                          compression=Compression.none,
                          backward_passes_per_step=1, op=ReduceOp.Average,
-                         gradient_predivide_factor=1.0):
-    """Wrap a torch optimizer with gradient averaging over all ranks
-    (hvd.DistributedOptimizer API)."""
+                         gradient_predivide_factor=1.0, process_set=None):
+    """Wrap a torch optimizer with gradient averaging over all ranks — or
+    over a registered ProcessSet's members (hvd.DistributedOptimizer API)."""
     cls = type(optimizer.__class__.__name__, (_DistributedOptimizer,
                                               optimizer.__class__), {})
     optimizer.__class__ = cls
     optimizer._distributed_init(named_parameters, compression,
                                 backward_passes_per_step, op,
-                                gradient_predivide_factor)
+                                gradient_predivide_factor, process_set)
     return optimizer

@@ -183,6 +183,17 @@ def test_fused_sgd_matches_torch_gpu(hvd):
         assert torch.allclose(p, q, rtol=1e-5, atol=1e-6)
 
 
+def test_process_set_gpu(hvd):
+    """Per-set RCCL communicator creation (ws=1: set {0} gets its own comm
+    distinct from the global one)."""
+    ps = hvd.add_process_set([0])
+    t = torch.arange(1000, dtype=torch.float32, device=DEV)
+    out = hvd.allreduce(t, op=hvd.Sum, process_set=ps, name="psg")
+    assert torch.allclose(out, t)
+    out2 = hvd.allreduce(t * 2, op=hvd.Sum, name="psg_global")
+    assert torch.allclose(out2, t * 2)
+
+
 def test_fused_sgd_channels_last_gpu(hvd):
     """channels_last params/grads (ResNet NHWC path): the flat-storage
     kernels must accept matching non-contiguous dense layouts."""
