@@ -211,11 +211,13 @@ bool do_poll(uint64_t handle) {
 }  // namespace
 
 void register_optim(pybind11::module_& m);  // optim_ops.cpp
+void register_bn(pybind11::module_& m);     // bn_ops.cpp
 
 }  // namespace axon
 
 PYBIND11_MODULE(_core, m) {
   axon::register_optim(m);
+  axon::register_bn(m);
   m.doc() = "axon core: MI355X-native ring-allreduce data-parallel engine";
   m.def("init", &axon::initialize, py::arg("rank"), py::arg("size"),
         py::arg("local_rank"), py::arg("control_path"),

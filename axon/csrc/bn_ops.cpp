@@ -1,0 +1,101 @@
+// Torch-facing launchers for the fused NHWC BatchNorm kernels
+// (kernels/batchnorm.hip) — replaces ATen's channels_last batch-norm
+// kernels, the largest single GPU-time consumer in the ResNet-50 config
+// (profiles/README.md). Runs on the CURRENT torch stream so it composes
+// with autograd like any other op.
+#include <torch/extension.h>
+
+#include <ATen/hip/HIPContext.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+
+#include "kernels/kernels.h"
+
+namespace axon {
+namespace {
+
+bool has(const at::Tensor& t) { return t.defined() && t.numel() > 0; }
+void* optr(const at::Tensor& t) { return has(t) ? t.data_ptr() : nullptr; }
+
+void check_nhwc(const at::Tensor& t, const char* what) {
+  TORCH_CHECK(t.is_cuda(), what, " must be a GPU tensor");
+  TORCH_CHECK(t.dim() == 4, what, " must be 4-D (NCHW logical)");
+  TORCH_CHECK(t.is_contiguous(at::MemoryFormat::ChannelsLast),
+              what, " must be channels_last contiguous");
+}
+
+std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
+                               const at::Tensor& beta,
+                               at::Tensor running_mean, at::Tensor running_var,
+                               double momentum, double eps, bool training,
+                               bool relu) {
+  check_nhwc(x, "bn input");
+  const int C = static_cast<int>(x.size(1));
+  const int64_t rows = x.numel() / C;
+  auto fopt = at::TensorOptions().dtype(at::kFloat).device(x.device());
+  at::Tensor save_mean = at::empty({C}, fopt);
+  at::Tensor save_rstd = at::empty({C}, fopt);
+  at::Tensor ab = at::empty({2 * C}, fopt);
+  at::Tensor y = at::empty_like(x);
+  at::Tensor scratch;
+  if (training) scratch = at::zeros({2 * C}, fopt);
+
+  int pdt = static_cast<int>(has(gamma) ? gamma.scalar_type()
+                             : has(running_mean) ? running_mean.scalar_type()
+                                                 : x.scalar_type());
+  TORCH_CHECK(training || (has(running_mean) && has(running_var)),
+              "bn eval mode needs running stats");
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(x.get_device());
+  launch_bn_fwd(static_cast<int>(x.scalar_type()), pdt, x.data_ptr(),
+                y.data_ptr(), rows, C, optr(gamma), optr(beta),
+                optr(running_mean), optr(running_var),
+                static_cast<float>(momentum), static_cast<float>(eps),
+                training, relu,
+                training ? scratch.data_ptr<float>() : nullptr,
+                save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
+                ab.data_ptr<float>(), stream.stream());
+  return {y, save_mean, save_rstd};
+}
+
+std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy_in,
+                               const at::Tensor& y, const at::Tensor& gamma,
+                               const at::Tensor& save_mean,
+                               const at::Tensor& save_rstd, bool relu) {
+  check_nhwc(x, "bn input");
+  at::Tensor dy = dy_in.is_contiguous(at::MemoryFormat::ChannelsLast)
+                      ? dy_in
+                      : dy_in.contiguous(at::MemoryFormat::ChannelsLast);
+  const int C = static_cast<int>(x.size(1));
+  const int64_t rows = x.numel() / C;
+  auto fopt = at::TensorOptions().dtype(at::kFloat).device(x.device());
+  at::Tensor scratch = at::zeros({2 * C}, fopt);
+  at::Tensor dgamma = at::empty({C}, fopt);
+  at::Tensor dbeta = at::empty({C}, fopt);
+  at::Tensor coef = at::empty({3 * C}, fopt);
+  at::Tensor dx = at::empty_like(x);
+  if (relu) TORCH_CHECK(has(y), "bn relu backward needs the fwd output");
+
+  int pdt = static_cast<int>(has(gamma) ? gamma.scalar_type()
+                                        : x.scalar_type());
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(x.get_device());
+  launch_bn_bwd(static_cast<int>(x.scalar_type()), pdt, x.data_ptr(),
+                dy.data_ptr(), relu ? y.data_ptr() : nullptr, dx.data_ptr(),
+                rows, C, optr(gamma),
+                save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(), relu,
+                scratch.data_ptr<float>(), dgamma.data_ptr<float>(),
+                dbeta.data_ptr<float>(), coef.data_ptr<float>(),
+                stream.stream());
+  return {dx, dgamma, dbeta};
+}
+
+}  // namespace
+
+void register_bn(pybind11::module_& m) {
+  m.def("bn_fwd", &bn_fwd, py::arg("x"), py::arg("gamma"), py::arg("beta"),
+        py::arg("running_mean"), py::arg("running_var"), py::arg("momentum"),
+        py::arg("eps"), py::arg("training"), py::arg("relu") = false);
+  m.def("bn_bwd", &bn_bwd, py::arg("x"), py::arg("dy"), py::arg("y"),
+        py::arg("gamma"), py::arg("save_mean"), py::arg("save_rstd"),
+        py::arg("relu") = false);
+}
+
+}  // namespace axon

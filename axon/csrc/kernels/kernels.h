@@ -47,4 +47,22 @@ void launch_add3(double* mine, const double* peer, hipStream_t stream);
 void launch_adasum_apply(void* kept, const void* recv, int64_t n, int dtype,
                          const double* dots3, bool a_side, hipStream_t stream);
 
+// Fused NHWC BatchNorm (batchnorm.hip): x/dy/y/dx are [rows, C] channels-
+// last streams; scratch is a zeroed fp32[2C] accumulator; save_mean/
+// save_rstd/ab/coef are fp32[C]/[C]/[2C]/[3C]. dtype args are
+// at::ScalarType codes (float/half/bfloat16). relu fuses the activation
+// into apply (fwd) and masks dy by y>0 (bwd).
+void launch_bn_fwd(int x_dtype, int param_dtype, const void* x, void* y,
+                   int64_t rows, int C, const void* gamma, const void* beta,
+                   void* running_mean, void* running_var, float momentum,
+                   float eps, bool training, bool relu, float* scratch,
+                   float* save_mean, float* save_rstd, float* ab,
+                   hipStream_t stream);
+void launch_bn_bwd(int x_dtype, int param_dtype, const void* x, const void* dy,
+                   const void* y, void* dx, int64_t rows, int C,
+                   const void* gamma, const float* save_mean,
+                   const float* save_rstd, bool relu, float* scratch,
+                   float* dgamma, float* dbeta, float* coef,
+                   hipStream_t stream);
+
 }  // namespace axon

@@ -7,27 +7,38 @@ architecture: He et al., Deep Residual Learning for Image Recognition.
 import torch
 import torch.nn as nn
 
+from axon.ops import FusedBatchNorm2d
+
 
 class Bottleneck(nn.Module):
     expansion = 4
 
-    def __init__(self, in_ch, width, stride=1, downsample=None):
+    def __init__(self, in_ch, width, stride=1, downsample=None,
+                 norm_layer=FusedBatchNorm2d):
         super().__init__()
         out_ch = width * self.expansion
+        fuses = norm_layer is FusedBatchNorm2d
         self.conv1 = nn.Conv2d(in_ch, width, 1, bias=False)
-        self.bn1 = nn.BatchNorm2d(width)
+        self.bn1 = norm_layer(width, fuse_relu=True) if fuses             else norm_layer(width)
         self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1,
                                bias=False)
-        self.bn2 = nn.BatchNorm2d(width)
+        self.bn2 = norm_layer(width, fuse_relu=True) if fuses             else norm_layer(width)
         self.conv3 = nn.Conv2d(width, out_ch, 1, bias=False)
-        self.bn3 = nn.BatchNorm2d(out_ch)
+        self.bn3 = norm_layer(out_ch)
         self.relu = nn.ReLU(inplace=True)
+        self._bn_fuses_relu = fuses
         self.downsample = downsample
 
     def forward(self, x):
         identity = x
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.relu(self.bn2(self.conv2(out)))
+        # bn1/bn2 fold their trailing ReLU into the fused NHWC BN kernel
+        # (axon/ops/batchnorm.py); the post-residual ReLU cannot fuse.
+        out = self.bn1(self.conv1(x))
+        if not self._bn_fuses_relu:
+            out = self.relu(out)
+        out = self.bn2(self.conv2(out))
+        if not self._bn_fuses_relu:
+            out = self.relu(out)
         out = self.bn3(self.conv3(out))
         if self.downsample is not None:
             identity = self.downsample(x)
@@ -35,12 +46,12 @@ class Bottleneck(nn.Module):
 
 
 class ResNet(nn.Module):
-    def __init__(self, layers, num_classes=1000, norm_layer=nn.BatchNorm2d):
+    def __init__(self, layers, num_classes=1000, norm_layer=FusedBatchNorm2d):
         super().__init__()
         self._norm = norm_layer
         self.in_ch = 64
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = norm_layer(64)
+        self.bn1 = norm_layer(64, fuse_relu=True)             if norm_layer is FusedBatchNorm2d else norm_layer(64)
         self.relu = nn.ReLU(inplace=True)
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         self.layer1 = self._make_layer(64, layers[0])
@@ -70,14 +81,18 @@ class ResNet(nn.Module):
                 nn.Conv2d(self.in_ch, out_ch, 1, stride=stride, bias=False),
                 self._norm(out_ch),
             )
-        layers = [Bottleneck(self.in_ch, width, stride, downsample)]
+        layers = [Bottleneck(self.in_ch, width, stride, downsample,
+                             norm_layer=self._norm)]
         self.in_ch = out_ch
         for _ in range(1, blocks):
-            layers.append(Bottleneck(self.in_ch, width))
+            layers.append(Bottleneck(self.in_ch, width, norm_layer=self._norm))
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.bn1(self.conv1(x))
+        if not getattr(self.bn1, "fuse_relu", False):
+            x = self.relu(x)
+        x = self.maxpool(x)
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         x = self.avgpool(x).flatten(1)
         return self.fc(x)

@@ -28,6 +28,13 @@ def _child(rank, size, sock, fn, args, q):
     os.environ["AXON_CONTROL_SOCKET"] = sock
     os.environ.setdefault("AXON_START_TIMEOUT", "60")
     try:
+        import torch
+
+        # Forked children inherit libgomp state from the parent; any OMP
+        # parallel region would then hang waiting on worker threads that do
+        # not exist post-fork. Single-thread the child before its first
+        # parallel region — these tests only touch tiny tensors.
+        torch.set_num_threads(1)
         import axon as hvd
 
         hvd.init()

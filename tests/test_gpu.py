@@ -183,6 +183,74 @@ def test_fused_sgd_matches_torch_gpu(hvd):
         assert torch.allclose(p, q, rtol=1e-5, atol=1e-6)
 
 
+@pytest.mark.parametrize("dtype,tol", [(torch.float32, 2e-4),
+                                       (torch.bfloat16, 3e-2)])
+@pytest.mark.parametrize("C,HW", [(64, 14), (128, 9), (256, 7), (96, 5)])
+def test_fused_batchnorm_matches_torch(hvd, dtype, tol, C, HW):
+    """Fused NHWC BN (kernels/batchnorm.hip) vs stock nn.BatchNorm2d:
+    fwd output, input/weight/bias grads, and running stats."""
+    from axon.ops import FusedBatchNorm2d
+
+    torch.manual_seed(C)
+    ref = torch.nn.BatchNorm2d(C).to(DEV)
+    fused = FusedBatchNorm2d(C).to(DEV)
+    fused.load_state_dict(ref.state_dict())
+    if dtype != torch.float32:
+        ref = ref.to(dtype)
+        fused = fused.to(dtype)
+    x = torch.randn(8, C, HW, HW, device=DEV, dtype=dtype) * 3 + 1
+    x1 = x.clone().requires_grad_(True)
+    x2 = x.clone().to(memory_format=torch.channels_last).requires_grad_(True)
+    y1 = ref(x1)
+    y2 = fused(x2)
+    assert torch.allclose(y1, y2.to(y1.dtype), rtol=tol, atol=tol), \
+        (y1 - y2).abs().max()
+    g = torch.randn_like(y1)
+    y1.backward(g)
+    y2.backward(g.to(memory_format=torch.channels_last))
+    assert torch.allclose(x1.grad, x2.grad, rtol=tol, atol=tol), \
+        (x1.grad - x2.grad).abs().max()
+    assert torch.allclose(ref.weight.grad, fused.weight.grad,
+                          rtol=tol, atol=tol * 10)
+    assert torch.allclose(ref.bias.grad, fused.bias.grad,
+                          rtol=tol, atol=tol * 10)
+    assert torch.allclose(ref.running_mean.float(),
+                          fused.running_mean.float(), rtol=tol, atol=tol)
+    assert torch.allclose(ref.running_var.float(),
+                          fused.running_var.float(), rtol=tol, atol=tol)
+    # eval mode (inference path, no grad)
+    ref.eval()
+    fused.eval()
+    with torch.no_grad():
+        ye1 = ref(x)
+        ye2 = fused(x.to(memory_format=torch.channels_last))
+    assert torch.allclose(ye1, ye2.to(ye1.dtype), rtol=tol, atol=tol)
+
+
+def test_fused_batchnorm_relu(hvd):
+    """fuse_relu folds the activation into fwd apply and masks dy in bwd."""
+    from axon.ops import FusedBatchNorm2d
+
+    torch.manual_seed(3)
+    C = 64
+    ref = torch.nn.BatchNorm2d(C).to(DEV)
+    fused = FusedBatchNorm2d(C, fuse_relu=True).to(DEV)
+    fused.load_state_dict(ref.state_dict())
+    x = torch.randn(4, C, 10, 10, device=DEV)
+    x1 = x.clone().requires_grad_(True)
+    x2 = x.clone().to(memory_format=torch.channels_last).requires_grad_(True)
+    y1 = torch.relu(ref(x1))
+    y2 = fused(x2)
+    assert torch.allclose(y1, y2, rtol=1e-4, atol=1e-5)
+    g = torch.randn_like(y1)
+    y1.backward(g)
+    y2.backward(g.to(memory_format=torch.channels_last))
+    assert torch.allclose(x1.grad, x2.grad, rtol=1e-4, atol=1e-5), \
+        (x1.grad - x2.grad).abs().max()
+    assert torch.allclose(ref.weight.grad, fused.weight.grad,
+                          rtol=1e-4, atol=1e-4)
+
+
 def test_process_set_gpu(hvd):
     """Per-set RCCL communicator creation (ws=1: set {0} gets its own comm
     distinct from the global one)."""
