@@ -27,8 +27,13 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
                                const at::Tensor& beta,
                                at::Tensor running_mean, at::Tensor running_var,
                                double momentum, double eps, bool training,
-                               bool relu) {
+                               bool relu, const at::Tensor& add) {
   check_nhwc(x, "bn input");
+  if (has(add)) {
+    check_nhwc(add, "bn residual");
+    TORCH_CHECK(add.sizes() == x.sizes() && add.scalar_type() == x.scalar_type(),
+                "bn residual shape/dtype mismatch");
+  }
   const int C = static_cast<int>(x.size(1));
   const int64_t rows = x.numel() / C;
   auto fopt = at::TensorOptions().dtype(at::kFloat).device(x.device());
@@ -36,8 +41,12 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
   at::Tensor save_rstd = at::empty({C}, fopt);
   at::Tensor ab = at::empty({2 * C}, fopt);
   at::Tensor y = at::empty_like(x);
-  at::Tensor scratch;
-  if (training) scratch = at::zeros({2 * C}, fopt);
+  at::Tensor scratch, partial;
+  if (training) {
+    scratch = at::empty({2 * C}, fopt);
+    partial = at::empty({static_cast<int64_t>(bn_partial_rows(rows, C)) * 2 * C},
+                        fopt);
+  }
 
   int pdt = static_cast<int>(has(gamma) ? gamma.scalar_type()
                              : has(running_mean) ? running_mean.scalar_type()
@@ -46,10 +55,11 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
               "bn eval mode needs running stats");
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(x.get_device());
   launch_bn_fwd(static_cast<int>(x.scalar_type()), pdt, x.data_ptr(),
-                y.data_ptr(), rows, C, optr(gamma), optr(beta),
+                y.data_ptr(), optr(add), rows, C, optr(gamma), optr(beta),
                 optr(running_mean), optr(running_var),
                 static_cast<float>(momentum), static_cast<float>(eps),
                 training, relu,
+                training ? partial.data_ptr<float>() : nullptr,
                 training ? scratch.data_ptr<float>() : nullptr,
                 save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
                 ab.data_ptr<float>(), stream.stream());
@@ -59,7 +69,8 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
 std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy_in,
                                const at::Tensor& y, const at::Tensor& gamma,
                                const at::Tensor& save_mean,
-                               const at::Tensor& save_rstd, bool relu) {
+                               const at::Tensor& save_rstd, bool relu,
+                               bool want_dadd) {
   check_nhwc(x, "bn input");
   at::Tensor dy = dy_in.is_contiguous(at::MemoryFormat::ChannelsLast)
                       ? dy_in
@@ -67,11 +78,15 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy_in,
   const int C = static_cast<int>(x.size(1));
   const int64_t rows = x.numel() / C;
   auto fopt = at::TensorOptions().dtype(at::kFloat).device(x.device());
-  at::Tensor scratch = at::zeros({2 * C}, fopt);
+  at::Tensor partial =
+      at::empty({static_cast<int64_t>(bn_partial_rows(rows, C)) * 2 * C}, fopt);
+  at::Tensor scratch = at::empty({2 * C}, fopt);
   at::Tensor dgamma = at::empty({C}, fopt);
   at::Tensor dbeta = at::empty({C}, fopt);
   at::Tensor coef = at::empty({3 * C}, fopt);
   at::Tensor dx = at::empty_like(x);
+  at::Tensor dadd;
+  if (want_dadd) dadd = at::empty_like(x);
   if (relu) TORCH_CHECK(has(y), "bn relu backward needs the fwd output");
 
   int pdt = static_cast<int>(has(gamma) ? gamma.scalar_type()
@@ -79,23 +94,66 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy_in,
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(x.get_device());
   launch_bn_bwd(static_cast<int>(x.scalar_type()), pdt, x.data_ptr(),
                 dy.data_ptr(), relu ? y.data_ptr() : nullptr, dx.data_ptr(),
-                rows, C, optr(gamma),
+                want_dadd ? dadd.data_ptr() : nullptr, rows, C, optr(gamma),
                 save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(), relu,
-                scratch.data_ptr<float>(), dgamma.data_ptr<float>(),
-                dbeta.data_ptr<float>(), coef.data_ptr<float>(),
-                stream.stream());
-  return {dx, dgamma, dbeta};
+                partial.data_ptr<float>(), scratch.data_ptr<float>(),
+                dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                coef.data_ptr<float>(), stream.stream());
+  if (!want_dadd) dadd = at::Tensor();
+  return {dx, dgamma, dbeta, dadd};
+}
+
+// ------------------------------------------------------------- RMSNorm
+
+std::vector<at::Tensor> rms_fwd(const at::Tensor& x, const at::Tensor& w,
+                                double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "rmsnorm input must be "
+              "contiguous GPU tensor");
+  const int H = static_cast<int>(x.size(-1));
+  TORCH_CHECK(w.is_contiguous() && w.numel() == H &&
+                  w.scalar_type() == x.scalar_type(),
+              "rmsnorm weight mismatch");
+  TORCH_CHECK(H % 2 == 0 && H <= 16384, "rmsnorm fast path needs even H <= "
+              "16384 (fallback in Python)");
+  const int64_t rows = x.numel() / H;
+  auto fopt = at::TensorOptions().dtype(at::kFloat).device(x.device());
+  at::Tensor rstd = at::empty({rows}, fopt);
+  at::Tensor y = at::empty_like(x);
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(x.get_device());
+  launch_rms_fwd(static_cast<int>(x.scalar_type()), x.data_ptr(), w.data_ptr(),
+                 y.data_ptr(), rstd.data_ptr<float>(), rows, H,
+                 static_cast<float>(eps), stream.stream());
+  return {y, rstd};
+}
+
+std::vector<at::Tensor> rms_bwd(const at::Tensor& x, const at::Tensor& dy_in,
+                                const at::Tensor& w, const at::Tensor& rstd) {
+  const int H = static_cast<int>(x.size(-1));
+  const int64_t rows = x.numel() / H;
+  at::Tensor dy = dy_in.is_contiguous() ? dy_in : dy_in.contiguous();
+  auto fopt = at::TensorOptions().dtype(at::kFloat).device(x.device());
+  at::Tensor dw = at::zeros({H}, fopt);
+  at::Tensor dx = at::empty_like(x);
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(x.get_device());
+  launch_rms_bwd(static_cast<int>(x.scalar_type()), x.data_ptr(), dy.data_ptr(),
+                 w.data_ptr(), rstd.data_ptr<float>(), dx.data_ptr(),
+                 dw.data_ptr<float>(), rows, H, stream.stream());
+  return {dx, dw};
 }
 
 }  // namespace
 
 void register_bn(pybind11::module_& m) {
+  m.def("rms_fwd", &rms_fwd, py::arg("x"), py::arg("w"), py::arg("eps"));
+  m.def("rms_bwd", &rms_bwd, py::arg("x"), py::arg("dy"), py::arg("w"),
+        py::arg("rstd"));
   m.def("bn_fwd", &bn_fwd, py::arg("x"), py::arg("gamma"), py::arg("beta"),
         py::arg("running_mean"), py::arg("running_var"), py::arg("momentum"),
-        py::arg("eps"), py::arg("training"), py::arg("relu") = false);
+        py::arg("eps"), py::arg("training"), py::arg("relu") = false,
+        py::arg("add") = at::Tensor());
   m.def("bn_bwd", &bn_bwd, py::arg("x"), py::arg("dy"), py::arg("y"),
         py::arg("gamma"), py::arg("save_mean"), py::arg("save_rstd"),
-        py::arg("relu") = false);
+        py::arg("relu") = false, py::arg("want_dadd") = false);
 }
 
 }  // namespace axon

@@ -56,11 +56,15 @@ struct alignas(sizeof(T) * V) BVec {
 };
 
 // ------------------------------------------------------ stats (fwd pass 1)
-// scratch[c] += sum(x[:, c]); scratch[C + c] += sum(x[:, c]^2)
+// partial[blockIdx.y][c] = sum over this block's rows of x[:, c];
+// partial[blockIdx.y][C + c] = sum of x[:, c]^2. No atomics: a second tiny
+// kernel (bn_reduce_partials) folds the nby partial rows — deterministic
+// and free of same-address atomic serialization (which measured 3-6x off
+// streaming SOL with ~2k blocks per channel).
 
 template <typename T, int VC>
 __global__ void bn_stats_kernel(const T* __restrict__ x, int64_t rows, int C,
-                                float* __restrict__ scratch, int rows_per_block) {
+                                float* __restrict__ partial, int rows_per_block) {
   const int lane = threadIdx.x % 64;
   const int wave = threadIdx.x / 64;
   const int c0 = blockIdx.x * 64 * VC + lane * VC;  // my first channel
@@ -91,7 +95,7 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, int64_t rows, int C,
     }
   }
 
-  // Cross-wave combine in LDS, then one atomic per channel per block.
+  // Cross-wave combine in LDS, then one coalesced partial-row store.
   __shared__ float lds[kWaves][64 * VC * 2];
 #pragma unroll
   for (int k = 0; k < VC; ++k) {
@@ -99,6 +103,7 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, int64_t rows, int C,
     lds[wave][64 * VC + lane * VC + k] = q[k];
   }
   __syncthreads();
+  float* out = partial + static_cast<int64_t>(blockIdx.y) * 2 * C;
   if (wave == 0) {
 #pragma unroll
     for (int k = 0; k < VC; ++k) {
@@ -110,10 +115,22 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, int64_t rows, int C,
         ss += lds[w][lane * VC + k];
         qq += lds[w][64 * VC + lane * VC + k];
       }
-      atomicAdd(&scratch[c], ss);
-      atomicAdd(&scratch[C + c], qq);
+      out[c] = ss;
+      out[C + c] = qq;
     }
   }
+}
+
+// Fold the nby partial rows into scratch[0..2C) (deterministic order).
+__global__ void bn_reduce_partials_kernel(const float* __restrict__ partial,
+                                          float* __restrict__ scratch, int twoC,
+                                          int nby) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= twoC) return;
+  float acc = 0.f;
+  for (int b = 0; b < nby; ++b)
+    acc += partial[static_cast<int64_t>(b) * twoC + c];
+  scratch[c] = acc;
 }
 
 // -------------------------------------------------- finalize (fwd, C-sized)
@@ -168,8 +185,9 @@ __global__ void bn_eval_coef_kernel(int C, float eps,
 // y = relu?(a[c] * x + b[c]) — one coalesced stream; per-channel coeffs are
 // broadcast reads served by L2/L1.
 
-template <typename T, int VC, bool RELU>
+template <typename T, int VC, bool RELU, bool ADD>
 __global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                const T* __restrict__ add,  // residual (ADD)
                                 int64_t nvec, int C,
                                 const float* __restrict__ ab) {
   const int cvec = C / VC;  // vectors per row (VC divides C by construction)
@@ -177,10 +195,13 @@ __global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ y,
        i += static_cast<int64_t>(gridDim.x) * blockDim.x) {
     int c0 = static_cast<int>(i % cvec) * VC;
     BVec<T, VC> xv = reinterpret_cast<const BVec<T, VC>*>(x)[i];
+    BVec<T, VC> av;
+    if (ADD) av = reinterpret_cast<const BVec<T, VC>*>(add)[i];
     BVec<T, VC> yv;
 #pragma unroll
     for (int k = 0; k < VC; ++k) {
       float out = fmaf(bf_of(xv.v[k]), ab[c0 + k], ab[C + c0 + k]);
+      if (ADD) out += bf_of(av.v[k]);
       if (RELU) out = fmaxf(out, 0.f);
       yv.v[k] = bt_of<T>(out);
     }
@@ -198,7 +219,7 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
                                      const T* __restrict__ y,  // RELU only
                                      int64_t rows, int C,
                                      const float* __restrict__ save_mean,
-                                     float* __restrict__ scratch,
+                                     float* __restrict__ partial,
                                      int rows_per_block) {
   const int lane = threadIdx.x % 64;
   const int wave = threadIdx.x / 64;
@@ -245,6 +266,7 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
     lds[wave][64 * VC + lane * VC + k] = sxdy[k];
   }
   __syncthreads();
+  float* out = partial + static_cast<int64_t>(blockIdx.y) * 2 * C;
   if (wave == 0) {
 #pragma unroll
     for (int k = 0; k < VC; ++k) {
@@ -256,8 +278,8 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
         a += lds[w][lane * VC + k];
         bsum += lds[w][64 * VC + lane * VC + k];
       }
-      atomicAdd(&scratch[c], a);
-      atomicAdd(&scratch[C + c], bsum);
+      out[c] = a;
+      out[C + c] = bsum;
     }
   }
 }
@@ -294,11 +316,13 @@ __global__ void bn_bwd_finalize_kernel(const float* __restrict__ scratch, int C,
 
 // --------------------------------------------------- backward apply (pass 2)
 
-template <typename T, int VC, bool RELU>
+template <typename T, int VC, bool RELU, bool DADD>
 __global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
                                     const T* __restrict__ dy,
                                     const T* __restrict__ y,  // RELU only
-                                    T* __restrict__ dx, int64_t nvec, int C,
+                                    T* __restrict__ dx,
+                                    T* __restrict__ dadd,  // DADD: dy_eff out
+                                    int64_t nvec, int C,
                                     const float* __restrict__ coef) {
   const int cvec = C / VC;
   for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
@@ -308,16 +332,18 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
     BVec<T, VC> dv = reinterpret_cast<const BVec<T, VC>*>(dy)[i];
     BVec<T, VC> yv;
     if (RELU) yv = reinterpret_cast<const BVec<T, VC>*>(y)[i];
-    BVec<T, VC> ov;
+    BVec<T, VC> ov, dav;
 #pragma unroll
     for (int k = 0; k < VC; ++k) {
       int c = c0 + k;
       float d = bf_of(dv.v[k]);
       if (RELU && bf_of(yv.v[k]) <= 0.f) d = 0.f;
+      if (DADD) dav.v[k] = bt_of<T>(d);
       ov.v[k] = bt_of<T>(fmaf(d, coef[c],
                               fmaf(bf_of(xv.v[k]), coef[C + c], coef[2 * C + c])));
     }
     reinterpret_cast<BVec<T, VC>*>(dx)[i] = ov;
+    if (DADD) reinterpret_cast<BVec<T, VC>*>(dadd)[i] = dav;
   }
 }
 
@@ -330,32 +356,42 @@ int pick_vc(int C) {
   return 1;  // 64-channel tiles with ragged-tail bounds checks
 }
 
-int stats_rows_per_block(int64_t rows, int ctiles) {
-  // Aim for >= ~1024 blocks total to fill 256 CUs across 8 XCDs, without
-  // making blocks so small the LDS/atomic epilogue dominates.
-  int64_t target_blocks = 2048 / (ctiles > 0 ? ctiles : 1) + 1;
-  int64_t rpb = (rows + target_blocks - 1) / target_blocks;
-  rpb = std::max<int64_t>(rpb, 64);
-  return static_cast<int>(std::min<int64_t>(rpb, rows));
+}  // namespace
+
+int bn_partial_rows(int64_t rows, int C) {
+  // grid.y for the reduction kernels: enough blocks to fill 256 CUs / 8
+  // XCDs (with grid.x = ctiles), few enough that the partial buffer and
+  // its fold stay trivial.
+  int vc = pick_vc(C);
+  int ctiles = (C + 64 * vc - 1) / (64 * vc);
+  int64_t target = 1024 / ctiles + 1;
+  int64_t max_rows_blocks = (rows + 63) / 64;
+  return static_cast<int>(std::max<int64_t>(1, std::min(target, max_rows_blocks)));
 }
 
+namespace {
+
 template <typename T, typename PT>
-void bn_fwd_impl(const void* x, void* y, int64_t rows, int C, const void* gamma,
-                 const void* beta, void* running_mean, void* running_var,
-                 float momentum, float eps, bool training, bool relu,
-                 float* scratch, float* save_mean, float* save_rstd, float* ab,
-                 hipStream_t stream) {
+void bn_fwd_impl(const void* x, void* y, const void* add, int64_t rows, int C,
+                 const void* gamma, const void* beta, void* running_mean,
+                 void* running_var, float momentum, float eps, bool training,
+                 bool relu, float* partial, float* scratch, float* save_mean,
+                 float* save_rstd, float* ab, hipStream_t stream) {
   int vc = pick_vc(C);
   if (training) {
     int ctiles = (C + 64 * vc - 1) / (64 * vc);
-    int rpb = stats_rows_per_block(rows, ctiles);
-    dim3 grid(ctiles, static_cast<unsigned>((rows + rpb - 1) / rpb));
+    int nby = bn_partial_rows(rows, C);
+    int rpb = static_cast<int>((rows + nby - 1) / nby);
+    nby = static_cast<int>((rows + rpb - 1) / rpb);
+    dim3 grid(ctiles, static_cast<unsigned>(nby));
     switch (vc) {
-      case 8: hipLaunchKernelGGL((bn_stats_kernel<T, 8>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, scratch, rpb); break;
-      case 4: hipLaunchKernelGGL((bn_stats_kernel<T, 4>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, scratch, rpb); break;
-      case 2: hipLaunchKernelGGL((bn_stats_kernel<T, 2>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, scratch, rpb); break;
-      default: hipLaunchKernelGGL((bn_stats_kernel<T, 1>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, scratch, rpb); break;
+      case 8: hipLaunchKernelGGL((bn_stats_kernel<T, 8>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, partial, rpb); break;
+      case 4: hipLaunchKernelGGL((bn_stats_kernel<T, 4>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, partial, rpb); break;
+      case 2: hipLaunchKernelGGL((bn_stats_kernel<T, 2>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, partial, rpb); break;
+      default: hipLaunchKernelGGL((bn_stats_kernel<T, 1>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, partial, rpb); break;
     }
+    hipLaunchKernelGGL(bn_reduce_partials_kernel, (2 * C + 255) / 256, 256, 0,
+                       stream, partial, scratch, 2 * C, nby);
     float inv_n = 1.f / static_cast<float>(rows);
     float unbias = rows > 1 ? static_cast<float>(rows) / (rows - 1) : 1.f;
     hipLaunchKernelGGL((bn_fwd_finalize_kernel<PT>), (C + 255) / 256, 256, 0,
@@ -370,59 +406,72 @@ void bn_fwd_impl(const void* x, void* y, int64_t rows, int C, const void* gamma,
                        static_cast<const PT*>(running_mean),
                        static_cast<const PT*>(running_var), ab);
   }
-  // Apply: VC must divide C for the flat-index channel math; pick_vc
-  // guarantees it except the vc==1 ragged case, which also divides.
   int64_t nvec = rows * C / vc;
   int blocks = static_cast<int>(std::min<int64_t>(
       (nvec + kThreads - 1) / kThreads, 16384));
-  auto launch_apply = [&](auto vctag, auto relutag) {
+  const bool has_add = add != nullptr;
+  auto launch_apply = [&](auto vctag, auto relutag, auto addtag) {
     constexpr int V = decltype(vctag)::value;
     constexpr bool R = decltype(relutag)::value;
-    hipLaunchKernelGGL((bn_apply_kernel<T, V, R>), blocks, kThreads, 0, stream,
-                       static_cast<const T*>(x), static_cast<T*>(y), nvec, C, ab);
+    constexpr bool A = decltype(addtag)::value;
+    hipLaunchKernelGGL((bn_apply_kernel<T, V, R, A>), blocks, kThreads, 0,
+                       stream, static_cast<const T*>(x), static_cast<T*>(y),
+                       static_cast<const T*>(add), nvec, C, ab);
   };
   using K1 = std::integral_constant<int, 1>;
   using K2 = std::integral_constant<int, 2>;
   using K4 = std::integral_constant<int, 4>;
   using K8 = std::integral_constant<int, 8>;
-  using RT = std::true_type;
-  using RF = std::false_type;
+  using BT = std::true_type;
+  using BF = std::false_type;
+  auto dispatch = [&](auto vctag) {
+    if (relu) {
+      has_add ? launch_apply(vctag, BT{}, BT{}) : launch_apply(vctag, BT{}, BF{});
+    } else {
+      has_add ? launch_apply(vctag, BF{}, BT{}) : launch_apply(vctag, BF{}, BF{});
+    }
+  };
   switch (vc) {
-    case 8: relu ? launch_apply(K8{}, RT{}) : launch_apply(K8{}, RF{}); break;
-    case 4: relu ? launch_apply(K4{}, RT{}) : launch_apply(K4{}, RF{}); break;
-    case 2: relu ? launch_apply(K2{}, RT{}) : launch_apply(K2{}, RF{}); break;
-    default: relu ? launch_apply(K1{}, RT{}) : launch_apply(K1{}, RF{}); break;
+    case 8: dispatch(K8{}); break;
+    case 4: dispatch(K4{}); break;
+    case 2: dispatch(K2{}); break;
+    default: dispatch(K1{}); break;
   }
 }
 
 template <typename T, typename PT>
 void bn_bwd_impl(const void* x, const void* dy, const void* y, void* dx,
-                 int64_t rows, int C, const void* gamma, const float* save_mean,
-                 const float* save_rstd, bool relu, float* scratch,
-                 float* dgamma, float* dbeta, float* coef, hipStream_t stream) {
+                 void* dadd, int64_t rows, int C, const void* gamma,
+                 const float* save_mean, const float* save_rstd, bool relu,
+                 float* partial, float* scratch, float* dgamma, float* dbeta,
+                 float* coef, hipStream_t stream) {
   int vc = pick_vc(C);
   int ctiles = (C + 64 * vc - 1) / (64 * vc);
-  int rpb = stats_rows_per_block(rows, ctiles);
-  dim3 grid(ctiles, static_cast<unsigned>((rows + rpb - 1) / rpb));
+  int nby = bn_partial_rows(rows, C);
+  int rpb = static_cast<int>((rows + nby - 1) / nby);
+  nby = static_cast<int>((rows + rpb - 1) / rpb);
+  dim3 grid(ctiles, static_cast<unsigned>(nby));
+  using K1 = std::integral_constant<int, 1>;
+  using K2 = std::integral_constant<int, 2>;
+  using K4 = std::integral_constant<int, 4>;
+  using K8 = std::integral_constant<int, 8>;
+  using BT = std::true_type;
+  using BF = std::false_type;
   auto launch_reduce = [&](auto vctag, auto relutag) {
     constexpr int V = decltype(vctag)::value;
     constexpr bool R = decltype(relutag)::value;
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, V, R>), grid, kThreads, 0,
                        stream, static_cast<const T*>(x), static_cast<const T*>(dy),
-                       static_cast<const T*>(y), rows, C, save_mean, scratch, rpb);
+                       static_cast<const T*>(y), rows, C, save_mean, partial, rpb);
   };
-  using K1 = std::integral_constant<int, 1>;
-  using K2 = std::integral_constant<int, 2>;
-  using K4 = std::integral_constant<int, 4>;
-  using K8 = std::integral_constant<int, 8>;
-  using RT = std::true_type;
-  using RF = std::false_type;
   switch (vc) {
-    case 8: relu ? launch_reduce(K8{}, RT{}) : launch_reduce(K8{}, RF{}); break;
-    case 4: relu ? launch_reduce(K4{}, RT{}) : launch_reduce(K4{}, RF{}); break;
-    case 2: relu ? launch_reduce(K2{}, RT{}) : launch_reduce(K2{}, RF{}); break;
-    default: relu ? launch_reduce(K1{}, RT{}) : launch_reduce(K1{}, RF{}); break;
+    case 8: relu ? launch_reduce(K8{}, BT{}) : launch_reduce(K8{}, BF{}); break;
+    case 4: relu ? launch_reduce(K4{}, BT{}) : launch_reduce(K4{}, BF{}); break;
+    case 2: relu ? launch_reduce(K2{}, BT{}) : launch_reduce(K2{}, BF{}); break;
+    default: relu ? launch_reduce(K1{}, BT{}) : launch_reduce(K1{}, BF{}); break;
   }
+  hipLaunchKernelGGL(bn_reduce_partials_kernel, (2 * C + 255) / 256, 256, 0,
+                     stream, partial, scratch, 2 * C, nby);
   hipLaunchKernelGGL((bn_bwd_finalize_kernel<PT>), (C + 255) / 256, 256, 0,
                      stream, scratch, C, 1.f / static_cast<float>(rows),
                      static_cast<const PT*>(gamma), save_mean, save_rstd,
@@ -430,65 +479,77 @@ void bn_bwd_impl(const void* x, const void* dy, const void* y, void* dx,
   int64_t nvec = rows * C / vc;
   int blocks = static_cast<int>(std::min<int64_t>(
       (nvec + kThreads - 1) / kThreads, 16384));
-  auto launch_apply = [&](auto vctag, auto relutag) {
+  const bool has_dadd = dadd != nullptr;
+  auto launch_apply = [&](auto vctag, auto relutag, auto dat) {
     constexpr int V = decltype(vctag)::value;
     constexpr bool R = decltype(relutag)::value;
-    hipLaunchKernelGGL((bn_bwd_apply_kernel<T, V, R>), blocks, kThreads, 0,
+    constexpr bool D = decltype(dat)::value;
+    hipLaunchKernelGGL((bn_bwd_apply_kernel<T, V, R, D>), blocks, kThreads, 0,
                        stream, static_cast<const T*>(x), static_cast<const T*>(dy),
-                       static_cast<const T*>(y), static_cast<T*>(dx), nvec, C, coef);
+                       static_cast<const T*>(y), static_cast<T*>(dx),
+                       static_cast<T*>(dadd), nvec, C, coef);
+  };
+  auto dispatch = [&](auto vctag) {
+    if (relu) {
+      has_dadd ? launch_apply(vctag, BT{}, BT{}) : launch_apply(vctag, BT{}, BF{});
+    } else {
+      has_dadd ? launch_apply(vctag, BF{}, BT{}) : launch_apply(vctag, BF{}, BF{});
+    }
   };
   switch (vc) {
-    case 8: relu ? launch_apply(K8{}, RT{}) : launch_apply(K8{}, RF{}); break;
-    case 4: relu ? launch_apply(K4{}, RT{}) : launch_apply(K4{}, RF{}); break;
-    case 2: relu ? launch_apply(K2{}, RT{}) : launch_apply(K2{}, RF{}); break;
-    default: relu ? launch_apply(K1{}, RT{}) : launch_apply(K1{}, RF{}); break;
+    case 8: dispatch(K8{}); break;
+    case 4: dispatch(K4{}); break;
+    case 2: dispatch(K2{}); break;
+    default: dispatch(K1{}); break;
   }
 }
 
 template <typename T>
-void bn_fwd_pt(int param_dtype, const void* x, void* y, int64_t rows, int C,
-               const void* gamma, const void* beta, void* rm, void* rv,
-               float momentum, float eps, bool training, bool relu,
-               float* scratch, float* save_mean, float* save_rstd, float* ab,
-               hipStream_t stream) {
+void bn_fwd_pt(int param_dtype, const void* x, void* y, const void* add,
+               int64_t rows, int C, const void* gamma, const void* beta,
+               void* rm, void* rv, float momentum, float eps, bool training,
+               bool relu, float* partial, float* scratch, float* save_mean,
+               float* save_rstd, float* ab, hipStream_t stream) {
   switch (param_dtype) {
     case kFloatId:
-      bn_fwd_impl<T, float>(x, y, rows, C, gamma, beta, rm, rv, momentum, eps,
-                            training, relu, scratch, save_mean, save_rstd, ab,
-                            stream);
+      bn_fwd_impl<T, float>(x, y, add, rows, C, gamma, beta, rm, rv, momentum,
+                            eps, training, relu, partial, scratch, save_mean,
+                            save_rstd, ab, stream);
       break;
     case kHalfId:
-      bn_fwd_impl<T, __half>(x, y, rows, C, gamma, beta, rm, rv, momentum, eps,
-                             training, relu, scratch, save_mean, save_rstd, ab,
-                             stream);
+      bn_fwd_impl<T, __half>(x, y, add, rows, C, gamma, beta, rm, rv, momentum,
+                             eps, training, relu, partial, scratch, save_mean,
+                             save_rstd, ab, stream);
       break;
     case kBF16Id:
-      bn_fwd_impl<T, __hip_bfloat16>(x, y, rows, C, gamma, beta, rm, rv,
-                                     momentum, eps, training, relu, scratch,
-                                     save_mean, save_rstd, ab, stream);
+      bn_fwd_impl<T, __hip_bfloat16>(x, y, add, rows, C, gamma, beta, rm, rv,
+                                     momentum, eps, training, relu, partial,
+                                     scratch, save_mean, save_rstd, ab, stream);
       break;
   }
 }
 
 template <typename T>
 void bn_bwd_pt(int param_dtype, const void* x, const void* dy, const void* y,
-               void* dx, int64_t rows, int C, const void* gamma,
+               void* dx, void* dadd, int64_t rows, int C, const void* gamma,
                const float* save_mean, const float* save_rstd, bool relu,
-               float* scratch, float* dgamma, float* dbeta, float* coef,
-               hipStream_t stream) {
+               float* partial, float* scratch, float* dgamma, float* dbeta,
+               float* coef, hipStream_t stream) {
   switch (param_dtype) {
     case kFloatId:
-      bn_bwd_impl<T, float>(x, dy, y, dx, rows, C, gamma, save_mean, save_rstd,
-                            relu, scratch, dgamma, dbeta, coef, stream);
+      bn_bwd_impl<T, float>(x, dy, y, dx, dadd, rows, C, gamma, save_mean,
+                            save_rstd, relu, partial, scratch, dgamma, dbeta,
+                            coef, stream);
       break;
     case kHalfId:
-      bn_bwd_impl<T, __half>(x, dy, y, dx, rows, C, gamma, save_mean, save_rstd,
-                             relu, scratch, dgamma, dbeta, coef, stream);
+      bn_bwd_impl<T, __half>(x, dy, y, dx, dadd, rows, C, gamma, save_mean,
+                             save_rstd, relu, partial, scratch, dgamma, dbeta,
+                             coef, stream);
       break;
     case kBF16Id:
-      bn_bwd_impl<T, __hip_bfloat16>(x, dy, y, dx, rows, C, gamma, save_mean,
-                                     save_rstd, relu, scratch, dgamma, dbeta,
-                                     coef, stream);
+      bn_bwd_impl<T, __hip_bfloat16>(x, dy, y, dx, dadd, rows, C, gamma,
+                                     save_mean, save_rstd, relu, partial,
+                                     scratch, dgamma, dbeta, coef, stream);
       break;
   }
 }
@@ -496,50 +557,53 @@ void bn_bwd_pt(int param_dtype, const void* x, const void* dy, const void* y,
 }  // namespace
 
 void launch_bn_fwd(int x_dtype, int param_dtype, const void* x, void* y,
-                   int64_t rows, int C, const void* gamma, const void* beta,
-                   void* running_mean, void* running_var, float momentum,
-                   float eps, bool training, bool relu, float* scratch,
-                   float* save_mean, float* save_rstd, float* ab,
-                   hipStream_t stream) {
+                   const void* add, int64_t rows, int C, const void* gamma,
+                   const void* beta, void* running_mean, void* running_var,
+                   float momentum, float eps, bool training, bool relu,
+                   float* partial, float* scratch, float* save_mean,
+                   float* save_rstd, float* ab, hipStream_t stream) {
   switch (x_dtype) {
     case kFloatId:
-      bn_fwd_pt<float>(param_dtype, x, y, rows, C, gamma, beta, running_mean,
-                       running_var, momentum, eps, training, relu, scratch,
-                       save_mean, save_rstd, ab, stream);
+      bn_fwd_pt<float>(param_dtype, x, y, add, rows, C, gamma, beta,
+                       running_mean, running_var, momentum, eps, training, relu,
+                       partial, scratch, save_mean, save_rstd, ab, stream);
       break;
     case kHalfId:
-      bn_fwd_pt<__half>(param_dtype, x, y, rows, C, gamma, beta, running_mean,
-                        running_var, momentum, eps, training, relu, scratch,
-                        save_mean, save_rstd, ab, stream);
+      bn_fwd_pt<__half>(param_dtype, x, y, add, rows, C, gamma, beta,
+                        running_mean, running_var, momentum, eps, training,
+                        relu, partial, scratch, save_mean, save_rstd, ab,
+                        stream);
       break;
     case kBF16Id:
-      bn_fwd_pt<__hip_bfloat16>(param_dtype, x, y, rows, C, gamma, beta,
+      bn_fwd_pt<__hip_bfloat16>(param_dtype, x, y, add, rows, C, gamma, beta,
                                 running_mean, running_var, momentum, eps,
-                                training, relu, scratch, save_mean, save_rstd,
-                                ab, stream);
+                                training, relu, partial, scratch, save_mean,
+                                save_rstd, ab, stream);
       break;
   }
 }
 
 void launch_bn_bwd(int x_dtype, int param_dtype, const void* x, const void* dy,
-                   const void* y, void* dx, int64_t rows, int C,
+                   const void* y, void* dx, void* dadd, int64_t rows, int C,
                    const void* gamma, const float* save_mean,
-                   const float* save_rstd, bool relu, float* scratch,
-                   float* dgamma, float* dbeta, float* coef,
+                   const float* save_rstd, bool relu, float* partial,
+                   float* scratch, float* dgamma, float* dbeta, float* coef,
                    hipStream_t stream) {
   switch (x_dtype) {
     case kFloatId:
-      bn_bwd_pt<float>(param_dtype, x, dy, y, dx, rows, C, gamma, save_mean,
-                       save_rstd, relu, scratch, dgamma, dbeta, coef, stream);
+      bn_bwd_pt<float>(param_dtype, x, dy, y, dx, dadd, rows, C, gamma,
+                       save_mean, save_rstd, relu, partial, scratch, dgamma,
+                       dbeta, coef, stream);
       break;
     case kHalfId:
-      bn_bwd_pt<__half>(param_dtype, x, dy, y, dx, rows, C, gamma, save_mean,
-                        save_rstd, relu, scratch, dgamma, dbeta, coef, stream);
+      bn_bwd_pt<__half>(param_dtype, x, dy, y, dx, dadd, rows, C, gamma,
+                        save_mean, save_rstd, relu, partial, scratch, dgamma,
+                        dbeta, coef, stream);
       break;
     case kBF16Id:
-      bn_bwd_pt<__hip_bfloat16>(param_dtype, x, dy, y, dx, rows, C, gamma,
-                                save_mean, save_rstd, relu, scratch, dgamma,
-                                dbeta, coef, stream);
+      bn_bwd_pt<__hip_bfloat16>(param_dtype, x, dy, y, dx, dadd, rows, C,
+                                gamma, save_mean, save_rstd, relu, partial,
+                                scratch, dgamma, dbeta, coef, stream);
       break;
   }
 }

@@ -48,21 +48,34 @@ void launch_adasum_apply(void* kept, const void* recv, int64_t n, int dtype,
                          const double* dots3, bool a_side, hipStream_t stream);
 
 // Fused NHWC BatchNorm (batchnorm.hip): x/dy/y/dx are [rows, C] channels-
-// last streams; scratch is a zeroed fp32[2C] accumulator; save_mean/
-// save_rstd/ab/coef are fp32[C]/[C]/[2C]/[3C]. dtype args are
-// at::ScalarType codes (float/half/bfloat16). relu fuses the activation
-// into apply (fwd) and masks dy by y>0 (bwd).
+// last streams. partial is fp32[bn_partial_rows(rows,C) * 2C] (uninit ok),
+// scratch fp32[2C]; save_mean/save_rstd/ab/coef are fp32[C]/[C]/[2C]/[3C].
+// dtype args are at::ScalarType codes (float/half/bfloat16). relu fuses
+// the activation into apply (fwd) and masks dy by y>0 (bwd). add (fwd) /
+// dadd (bwd) fuse a residual: y = relu(a*x+b+add), dadd = relu-masked dy.
+int bn_partial_rows(int64_t rows, int C);
 void launch_bn_fwd(int x_dtype, int param_dtype, const void* x, void* y,
-                   int64_t rows, int C, const void* gamma, const void* beta,
-                   void* running_mean, void* running_var, float momentum,
-                   float eps, bool training, bool relu, float* scratch,
-                   float* save_mean, float* save_rstd, float* ab,
-                   hipStream_t stream);
+                   const void* add, int64_t rows, int C, const void* gamma,
+                   const void* beta, void* running_mean, void* running_var,
+                   float momentum, float eps, bool training, bool relu,
+                   float* partial, float* scratch, float* save_mean,
+                   float* save_rstd, float* ab, hipStream_t stream);
 void launch_bn_bwd(int x_dtype, int param_dtype, const void* x, const void* dy,
-                   const void* y, void* dx, int64_t rows, int C,
+                   const void* y, void* dx, void* dadd, int64_t rows, int C,
                    const void* gamma, const float* save_mean,
-                   const float* save_rstd, bool relu, float* scratch,
-                   float* dgamma, float* dbeta, float* coef,
+                   const float* save_rstd, bool relu, float* partial,
+                   float* scratch, float* dgamma, float* dbeta, float* coef,
                    hipStream_t stream);
+
+// Fused RMSNorm (rmsnorm.hip): x/y/dx are [rows, H] contiguous, w is [H]
+// (same dtype as x), rstd fp32[rows], dw a ZEROED fp32[H] accumulator.
+// Caller guarantees H % 2 == 0 and H <= 16384 (register/LDS bounds) —
+// the Python wrapper falls back to eager otherwise.
+void launch_rms_fwd(int dtype, const void* x, const void* w, void* y,
+                    float* rstd, int64_t rows, int H, float eps,
+                    hipStream_t stream);
+void launch_rms_bwd(int dtype, const void* x, const void* dy, const void* w,
+                    const float* rstd, void* dx, float* dw, int64_t rows,
+                    int H, hipStream_t stream);
 
 }  // namespace axon

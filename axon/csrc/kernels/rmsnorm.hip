@@ -1,0 +1,243 @@
+// Fused RMSNorm for CDNA4 — the Llama elementwise hot spot: the eager
+// formula (cast→square→mean→rsqrt→mul→mul→cast) costs ~7 kernels and 4+
+// passes over x per call; this is one kernel and ONE global read of x
+// (row staged through LDS) per direction. rocprof context:
+// profiles/README.md (llama: elementwise ≈12% of step time).
+//
+//   fwd: y = x * w * rsqrt(mean(x^2) + eps); saves per-row rstd (fp32).
+//   bwd: dx = rstd*w*dy - x * rstd^3/H * sum(dy*w*x)
+//        dw = sum_rows(dy * x * rstd)  (fp32 scratch, atomics per block)
+//
+// Layout: x is [rows, H], H innermost and contiguous. One block per row
+// (fwd) / per row-batch (bwd); 256 threads, 16-byte vector IO, fp32
+// accumulation, LDS tree for the cross-wave reduction.
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+#include <hip/hip_runtime.h>
+
+#include <algorithm>
+
+#include "kernels.h"
+
+namespace axon {
+
+namespace {
+
+constexpr int kThreads = 256;
+constexpr int kWaves = 4;
+constexpr int kHalfId = 5, kFloatId = 6, kBF16Id = 15;
+
+template <typename T>
+__device__ __forceinline__ float rf_of(T x);
+template <>
+__device__ __forceinline__ float rf_of(float x) { return x; }
+template <>
+__device__ __forceinline__ float rf_of(__half x) { return __half2float(x); }
+template <>
+__device__ __forceinline__ float rf_of(__hip_bfloat16 x) { return __bfloat162float(x); }
+
+template <typename T>
+__device__ __forceinline__ T rt_of(float x);
+template <>
+__device__ __forceinline__ float rt_of(float x) { return x; }
+template <>
+__device__ __forceinline__ __half rt_of(float x) { return __float2half(x); }
+template <>
+__device__ __forceinline__ __hip_bfloat16 rt_of(float x) { return __float2bfloat16(x); }
+
+template <typename T, int V>
+struct alignas(sizeof(T) * V) RVec {
+  T v[V];
+};
+
+// Block-wide sum of one float per thread; returns the total to all threads.
+// red[] must be __shared__ float[kWaves].
+__device__ float block_sum(float v, float* red) {
+  // wave-level tree via xor shuffles (wave64)
+  for (int off = 32; off > 0; off >>= 1)
+    v += __shfl_xor(v, off, 64);
+  const int lane = threadIdx.x % 64, wave = threadIdx.x / 64;
+  if (lane == 0) red[wave] = v;
+  __syncthreads();
+  float total = red[0];
+#pragma unroll
+  for (int w = 1; w < kWaves; ++w) total += red[w];
+  return total;
+}
+
+// ------------------------------------------------------------------ forward
+// One block per row; row staged in LDS (bytes = H * sizeof(T)).
+
+template <typename T, int VC>
+__global__ void rms_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                               T* __restrict__ y, float* __restrict__ rstd_out,
+                               int64_t rows, int H, float eps) {
+  extern __shared__ unsigned char smem[];
+  T* row = reinterpret_cast<T*>(smem);
+  __shared__ float red[kWaves];
+  const int hv = H / VC;
+
+  for (int64_t r = blockIdx.x; r < rows; r += gridDim.x) {
+    const RVec<T, VC>* xr = reinterpret_cast<const RVec<T, VC>*>(x + r * H);
+    float ss = 0.f;
+    for (int i = threadIdx.x; i < hv; i += kThreads) {
+      RVec<T, VC> xv = xr[i];
+      reinterpret_cast<RVec<T, VC>*>(row)[i] = xv;  // stage for pass 2
+#pragma unroll
+      for (int k = 0; k < VC; ++k) {
+        float f = rf_of(xv.v[k]);
+        ss += f * f;
+      }
+    }
+    __syncthreads();  // red[] reuse across rows
+    float total = block_sum(ss, red);
+    float rstd = rsqrtf(total / H + eps);
+    if (threadIdx.x == 0 && rstd_out) rstd_out[r] = rstd;
+
+    RVec<T, VC>* yr = reinterpret_cast<RVec<T, VC>*>(y + r * H);
+    const RVec<T, VC>* wr = reinterpret_cast<const RVec<T, VC>*>(w);
+    for (int i = threadIdx.x; i < hv; i += kThreads) {
+      RVec<T, VC> xv = reinterpret_cast<const RVec<T, VC>*>(row)[i];
+      RVec<T, VC> wv = wr[i];
+      RVec<T, VC> yv;
+#pragma unroll
+      for (int k = 0; k < VC; ++k)
+        yv.v[k] = rt_of<T>(rf_of(xv.v[k]) * rstd * rf_of(wv.v[k]));
+      yr[i] = yv;
+    }
+    __syncthreads();
+  }
+}
+
+// ----------------------------------------------------------------- backward
+// Each block owns a contiguous row range; x/dy staged per row in LDS;
+// per-thread dw partials accumulate across the block's rows in registers
+// (thread t owns columns {t*VC + j*kThreads*VC}), one atomicAdd sweep at
+// the end. kMaxDwChunks bounds the register array: H <= 32768 at VC=8.
+
+constexpr int kMaxDwChunks = 8;  // bounds dw registers: H <= 16384 at VC=8
+
+template <typename T, int VC>
+__global__ void rms_bwd_kernel(const T* __restrict__ x, const T* __restrict__ dy,
+                               const T* __restrict__ w,
+                               const float* __restrict__ rstd_in,
+                               T* __restrict__ dx, float* __restrict__ dw,
+                               int64_t rows, int H, int rows_per_block) {
+  extern __shared__ unsigned char smem[];
+  T* xrow = reinterpret_cast<T*>(smem);
+  T* drow = reinterpret_cast<T*>(smem + static_cast<size_t>(H) * sizeof(T));
+  __shared__ float red[kWaves];
+  const int hv = H / VC;
+  const int chunks = (hv + kThreads - 1) / kThreads;
+
+  float dwacc[kMaxDwChunks][VC];
+  for (int j = 0; j < chunks; ++j)
+#pragma unroll
+    for (int k = 0; k < VC; ++k) dwacc[j][k] = 0.f;
+
+  const int64_t r0 = static_cast<int64_t>(blockIdx.x) * rows_per_block;
+  const int64_t r1 = r0 + rows_per_block > rows ? rows : r0 + rows_per_block;
+  const RVec<T, VC>* wr = reinterpret_cast<const RVec<T, VC>*>(w);
+
+  for (int64_t r = r0; r < r1; ++r) {
+    const RVec<T, VC>* xr = reinterpret_cast<const RVec<T, VC>*>(x + r * H);
+    const RVec<T, VC>* dr = reinterpret_cast<const RVec<T, VC>*>(dy + r * H);
+    float rstd = rstd_in[r];
+    float dot = 0.f;
+    for (int i = threadIdx.x, j = 0; i < hv; i += kThreads, ++j) {
+      RVec<T, VC> xv = xr[i], dv = dr[i], wv = wr[i];
+      reinterpret_cast<RVec<T, VC>*>(xrow)[i] = xv;
+      reinterpret_cast<RVec<T, VC>*>(drow)[i] = dv;
+#pragma unroll
+      for (int k = 0; k < VC; ++k) {
+        float xf = rf_of(xv.v[k]), df = rf_of(dv.v[k]);
+        dot += df * rf_of(wv.v[k]) * xf;
+        dwacc[j][k] += df * xf * rstd;
+      }
+    }
+    __syncthreads();
+    float total = block_sum(dot, red);
+    float c2 = total * rstd * rstd * rstd / H;
+
+    RVec<T, VC>* oxr = reinterpret_cast<RVec<T, VC>*>(dx + r * H);
+    for (int i = threadIdx.x; i < hv; i += kThreads) {
+      RVec<T, VC> xv = reinterpret_cast<const RVec<T, VC>*>(xrow)[i];
+      RVec<T, VC> dv = reinterpret_cast<const RVec<T, VC>*>(drow)[i];
+      RVec<T, VC> wv = wr[i];
+      RVec<T, VC> ov;
+#pragma unroll
+      for (int k = 0; k < VC; ++k)
+        ov.v[k] = rt_of<T>(rstd * rf_of(wv.v[k]) * rf_of(dv.v[k]) -
+                           rf_of(xv.v[k]) * c2);
+      oxr[i] = ov;
+    }
+    __syncthreads();
+  }
+
+  for (int i = threadIdx.x, j = 0; i < hv; i += kThreads, ++j)
+#pragma unroll
+    for (int k = 0; k < VC; ++k) atomicAdd(&dw[i * VC + k], dwacc[j][k]);
+}
+
+int rms_vc(int H, size_t /*tsize*/) {
+  if (H % 8 == 0) return 8;
+  if (H % 4 == 0) return 4;
+  if (H % 2 == 0) return 2;
+  return 1;
+}
+
+template <typename T>
+void rms_fwd_t(const void* x, const void* w, void* y, float* rstd, int64_t rows,
+               int H, float eps, hipStream_t stream) {
+  int vc = rms_vc(H, sizeof(T));
+  size_t lds = static_cast<size_t>(H) * sizeof(T);
+  int blocks = static_cast<int>(std::min<int64_t>(rows, 16384));
+  switch (vc) {
+    case 8: hipLaunchKernelGGL((rms_fwd_kernel<T, 8>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(w), static_cast<T*>(y), rstd, rows, H, eps); break;
+    case 4: hipLaunchKernelGGL((rms_fwd_kernel<T, 4>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(w), static_cast<T*>(y), rstd, rows, H, eps); break;
+    case 2: hipLaunchKernelGGL((rms_fwd_kernel<T, 2>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(w), static_cast<T*>(y), rstd, rows, H, eps); break;
+    default: hipLaunchKernelGGL((rms_fwd_kernel<T, 1>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(w), static_cast<T*>(y), rstd, rows, H, eps); break;
+  }
+}
+
+template <typename T>
+void rms_bwd_t(const void* x, const void* dy, const void* w, const float* rstd,
+               void* dx, float* dw, int64_t rows, int H, hipStream_t stream) {
+  int vc = rms_vc(H, sizeof(T));
+  // Bound the per-thread dw register array: chunks <= kMaxDwChunks.
+  while ((H / vc + kThreads - 1) / kThreads > kMaxDwChunks && vc < 8) vc *= 2;
+  size_t lds = 2 * static_cast<size_t>(H) * sizeof(T);
+  // ~8 rows per block balances dw atomic traffic vs fill.
+  int rpb = static_cast<int>(std::max<int64_t>(1, rows / 2048));
+  int blocks = static_cast<int>((rows + rpb - 1) / rpb);
+  switch (vc) {
+    case 8: hipLaunchKernelGGL((rms_bwd_kernel<T, 8>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), rstd, static_cast<T*>(dx), dw, rows, H, rpb); break;
+    case 4: hipLaunchKernelGGL((rms_bwd_kernel<T, 4>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), rstd, static_cast<T*>(dx), dw, rows, H, rpb); break;
+    case 2: hipLaunchKernelGGL((rms_bwd_kernel<T, 2>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), rstd, static_cast<T*>(dx), dw, rows, H, rpb); break;
+    default: hipLaunchKernelGGL((rms_bwd_kernel<T, 1>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), rstd, static_cast<T*>(dx), dw, rows, H, rpb); break;
+  }
+}
+
+}  // namespace
+
+void launch_rms_fwd(int dtype, const void* x, const void* w, void* y,
+                    float* rstd, int64_t rows, int H, float eps,
+                    hipStream_t stream) {
+  switch (dtype) {
+    case kFloatId: rms_fwd_t<float>(x, w, y, rstd, rows, H, eps, stream); break;
+    case kHalfId: rms_fwd_t<__half>(x, w, y, rstd, rows, H, eps, stream); break;
+    case kBF16Id: rms_fwd_t<__hip_bfloat16>(x, w, y, rstd, rows, H, eps, stream); break;
+  }
+}
+
+void launch_rms_bwd(int dtype, const void* x, const void* dy, const void* w,
+                    const float* rstd, void* dx, float* dw, int64_t rows, int H,
+                    hipStream_t stream) {
+  switch (dtype) {
+    case kFloatId: rms_bwd_t<float>(x, dy, w, rstd, dx, dw, rows, H, stream); break;
+    case kHalfId: rms_bwd_t<__half>(x, dy, w, rstd, dx, dw, rows, H, stream); break;
+    case kBF16Id: rms_bwd_t<__hip_bfloat16>(x, dy, w, rstd, dx, dw, rows, H, stream); break;
+  }
+}
+
+}  // namespace axon

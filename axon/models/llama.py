@@ -13,6 +13,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from axon.ops.rmsnorm import FusedRMSNorm
+
 
 @dataclass
 class LlamaConfig:
@@ -42,17 +44,9 @@ def llama_tiny_config():
                        ffn=128, max_seq=128)
 
 
-class RMSNorm(nn.Module):
-    def __init__(self, dim, eps):
-        super().__init__()
-        self.weight = nn.Parameter(torch.ones(dim))
-        self.eps = eps
-
-    def forward(self, x):
-        dt = x.dtype
-        x = x.float()
-        x = x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + self.eps)
-        return (x * self.weight.float()).to(dt)
+# RMSNorm: the fused CDNA4 kernel (axon/ops/rmsnorm.py) on GPU, eager fp32
+# formula elsewhere — identical math, same state dict.
+RMSNorm = FusedRMSNorm
 
 
 def _rope_cache(dim, seq, theta, device, dtype):

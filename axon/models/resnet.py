@@ -31,18 +31,22 @@ class Bottleneck(nn.Module):
 
     def forward(self, x):
         identity = x
-        # bn1/bn2 fold their trailing ReLU into the fused NHWC BN kernel
-        # (axon/ops/batchnorm.py); the post-residual ReLU cannot fuse.
+        # bn1/bn2 fold their trailing ReLU into the fused NHWC BN kernel;
+        # bn3 additionally folds the residual add + final ReLU
+        # (axon/ops/batchnorm.py forward_add_relu) — the whole bottleneck
+        # tail is one pass over the activations.
         out = self.bn1(self.conv1(x))
         if not self._bn_fuses_relu:
             out = self.relu(out)
         out = self.bn2(self.conv2(out))
         if not self._bn_fuses_relu:
             out = self.relu(out)
-        out = self.bn3(self.conv3(out))
+        out = self.conv3(out)
         if self.downsample is not None:
             identity = self.downsample(x)
-        return self.relu(out + identity)
+        if self._bn_fuses_relu:
+            return self.bn3.forward_add_relu(out, identity)
+        return self.relu(self.bn3(out) + identity)
 
 
 class ResNet(nn.Module):

@@ -18,19 +18,26 @@ _SUPPORTED = (torch.float32, torch.float16, torch.bfloat16)
 
 
 class _BatchNormFunc(torch.autograd.Function):
+    """Training-mode fused BN; optionally also fuses a trailing residual
+    add (bn3 + identity in a ResNet bottleneck) and/or ReLU into the same
+    apply pass. `add=None` means no residual input."""
+
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var, momentum,
-                eps, relu):
+                eps, relu, add):
         from axon import _core
 
+        empty = torch.Tensor()
         y, save_mean, save_rstd = _core.bn_fwd(
-            x, weight if weight is not None else torch.Tensor(),
-            bias if bias is not None else torch.Tensor(),
-            running_mean if running_mean is not None else torch.Tensor(),
-            running_var if running_var is not None else torch.Tensor(),
-            momentum, eps, True, relu)
+            x, weight if weight is not None else empty,
+            bias if bias is not None else empty,
+            running_mean if running_mean is not None else empty,
+            running_var if running_var is not None else empty,
+            momentum, eps, True, relu,
+            add if add is not None else empty)
         ctx.relu = relu
         ctx.has_weight = weight is not None
+        ctx.has_add = add is not None
         ctx.save_for_backward(x,
                               weight if weight is not None else None,
                               save_mean, save_rstd,
@@ -42,15 +49,21 @@ class _BatchNormFunc(torch.autograd.Function):
         from axon import _core
 
         x, weight, save_mean, save_rstd, y = ctx.saved_tensors
-        dx, dgamma, dbeta = _core.bn_bwd(
-            x, dy, y if y is not None else torch.Tensor(),
-            weight if weight is not None else torch.Tensor(),
-            save_mean, save_rstd, ctx.relu)
+        empty = torch.Tensor()
+        dx, dgamma, dbeta, dadd = _core.bn_bwd(
+            x, dy, y if y is not None else empty,
+            weight if weight is not None else empty,
+            save_mean, save_rstd, ctx.relu, ctx.has_add)
         dw = db = None
         if ctx.has_weight:
             dw = dgamma.to(weight.dtype)
             db = dbeta.to(weight.dtype)
-        return dx, dw, db, None, None, None, None, None
+        if not ctx.has_add:
+            dadd = None
+        elif not ctx.relu:
+            # without a relu mask d(add) is just dy (pass-through)
+            dadd = dy
+        return dx, dw, db, None, None, None, None, None, dadd
 
 
 class FusedBatchNorm2d(nn.BatchNorm2d):
@@ -69,19 +82,22 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
                 x.is_contiguous(memory_format=torch.channels_last) and
                 x.size(1) == self.num_features)
 
+    def _train_momentum(self):
+        momentum = self.momentum if self.momentum is not None else 0.1
+        if self.track_running_stats and self.num_batches_tracked is not None:
+            self.num_batches_tracked.add_(1)
+            if self.momentum is None:  # cumulative moving average
+                momentum = 1.0 / float(self.num_batches_tracked)
+        return momentum
+
     def forward(self, x):
         if self._fast_path(x):
             if self.training:
-                momentum = self.momentum if self.momentum is not None else 0.1
-                if self.track_running_stats and \
-                        self.num_batches_tracked is not None:
-                    self.num_batches_tracked.add_(1)
-                    if self.momentum is None:  # cumulative moving average
-                        momentum = 1.0 / float(self.num_batches_tracked)
                 rm = self.running_mean if self.track_running_stats else None
                 rv = self.running_var if self.track_running_stats else None
                 return _BatchNormFunc.apply(x, self.weight, self.bias, rm, rv,
-                                            momentum, self.eps, self.fuse_relu)
+                                            self._train_momentum(), self.eps,
+                                            self.fuse_relu, None)
             if self.track_running_stats and not torch.is_grad_enabled():
                 from axon import _core
 
@@ -94,3 +110,16 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
                 return y
         y = super().forward(x)
         return F.relu(y, inplace=True) if self.fuse_relu else y
+
+    def forward_add_relu(self, x, residual):
+        """y = relu(bn(x) + residual) in ONE fused pass (the bottleneck
+        tail). Falls back to unfused ops off the fast path."""
+        if self._fast_path(x) and self.training and \
+                residual.is_contiguous(memory_format=torch.channels_last):
+            rm = self.running_mean if self.track_running_stats else None
+            rv = self.running_var if self.track_running_stats else None
+            return _BatchNormFunc.apply(x, self.weight, self.bias, rm, rv,
+                                        self._train_momentum(), self.eps,
+                                        True, residual)
+        assert not self.fuse_relu  # bn3-style norm: relu applied here, once
+        return F.relu(self.forward(x) + residual, inplace=True)
