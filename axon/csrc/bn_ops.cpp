@@ -141,9 +141,57 @@ std::vector<at::Tensor> rms_bwd(const at::Tensor& x, const at::Tensor& dy_in,
   return {dx, dw};
 }
 
+at::Tensor rope_apply(const at::Tensor& x, const at::Tensor& cos_t,
+                      const at::Tensor& sin_t, bool backward) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4,
+              "rope input must be contiguous [B,S,H,D] GPU tensor");
+  const int D = static_cast<int>(x.size(3));
+  const int H = static_cast<int>(x.size(2));
+  const int S = static_cast<int>(x.size(1));
+  TORCH_CHECK(D % 2 == 0, "rope head dim must be even");
+  TORCH_CHECK(cos_t.is_contiguous() && sin_t.is_contiguous() &&
+                  cos_t.numel() == static_cast<int64_t>(S) * (D / 2) &&
+                  cos_t.scalar_type() == x.scalar_type() &&
+                  sin_t.scalar_type() == x.scalar_type(),
+              "rope cos/sin must be [S, D/2] in the input dtype");
+  at::Tensor y = at::empty_like(x);
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(x.get_device());
+  launch_rope(static_cast<int>(x.scalar_type()), x.data_ptr(), y.data_ptr(),
+              cos_t.data_ptr(), sin_t.data_ptr(), x.numel(), D / 2, H, S,
+              backward, stream.stream());
+  return y;
+}
+
+at::Tensor swiglu_fwd(const at::Tensor& a, const at::Tensor& g) {
+  TORCH_CHECK(a.is_cuda() && a.is_contiguous() && g.is_contiguous() &&
+                  a.sizes() == g.sizes() && a.scalar_type() == g.scalar_type(),
+              "swiglu inputs must be matching contiguous GPU tensors");
+  at::Tensor z = at::empty_like(a);
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(a.get_device());
+  launch_swiglu_fwd(static_cast<int>(a.scalar_type()), a.data_ptr(),
+                    g.data_ptr(), z.data_ptr(), a.numel(), stream.stream());
+  return z;
+}
+
+std::vector<at::Tensor> swiglu_bwd(const at::Tensor& dz_in, const at::Tensor& a,
+                                   const at::Tensor& g) {
+  at::Tensor dz = dz_in.is_contiguous() ? dz_in : dz_in.contiguous();
+  at::Tensor da = at::empty_like(a);
+  at::Tensor dg = at::empty_like(g);
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(a.get_device());
+  launch_swiglu_bwd(static_cast<int>(a.scalar_type()), dz.data_ptr(),
+                    a.data_ptr(), g.data_ptr(), da.data_ptr(), dg.data_ptr(),
+                    a.numel(), stream.stream());
+  return {da, dg};
+}
+
 }  // namespace
 
 void register_bn(pybind11::module_& m) {
+  m.def("rope", &rope_apply, py::arg("x"), py::arg("cos"), py::arg("sin"),
+        py::arg("backward") = false);
+  m.def("swiglu_fwd", &swiglu_fwd, py::arg("a"), py::arg("g"));
+  m.def("swiglu_bwd", &swiglu_bwd, py::arg("dz"), py::arg("a"), py::arg("g"));
   m.def("rms_fwd", &rms_fwd, py::arg("x"), py::arg("w"), py::arg("eps"));
   m.def("rms_bwd", &rms_bwd, py::arg("x"), py::arg("dy"), py::arg("w"),
         py::arg("rstd"));

@@ -78,4 +78,15 @@ void launch_rms_bwd(int dtype, const void* x, const void* dy, const void* w,
                     const float* rstd, void* dx, float* dw, int64_t rows,
                     int H, hipStream_t stream);
 
+// Rotary embedding (elementwise.hip): x/y are [B,S,H,D] contiguous (numel
+// elements), cos/sin [S, D/2] in the same dtype; backward = transposed
+// rotation. SwiGLU: z = silu(a)*g, all tensors same-shape contiguous.
+void launch_rope(int dtype, const void* x, void* y, const void* cos_t,
+                 const void* sin_t, int64_t numel, int half_d, int heads,
+                 int seq, bool backward, hipStream_t stream);
+void launch_swiglu_fwd(int dtype, const void* a, const void* g, void* z,
+                       int64_t numel, hipStream_t stream);
+void launch_swiglu_bwd(int dtype, const void* dz, const void* a, const void* g,
+                       void* da, void* dg, int64_t numel, hipStream_t stream);
+
 }  // namespace axon

@@ -14,6 +14,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from axon.ops.rmsnorm import FusedRMSNorm
+from axon.ops.rotary import rope, swiglu
 
 
 @dataclass
@@ -56,15 +57,8 @@ def _rope_cache(dim, seq, theta, device, dtype):
     return torch.cos(freqs).to(dtype), torch.sin(freqs).to(dtype)
 
 
-def _apply_rope(x, cos, sin):
-    # x: (b, h, s, d); cos/sin: (s, d/2)
-    x1, x2 = x[..., 0::2], x[..., 1::2]
-    c = cos.view(1, 1, cos.shape[0], -1)
-    s = sin.view(1, 1, sin.shape[0], -1)
-    out = torch.empty_like(x)
-    out[..., 0::2] = x1 * c - x2 * s
-    out[..., 1::2] = x2 * c + x1 * s
-    return out
+# rotary applied via the fused CDNA4 kernel (axon/ops/rotary.py) on the
+# pre-transpose (b, s, h, d) layout; eager strided fallback off-GPU.
 
 
 class LlamaAttention(nn.Module):
@@ -80,13 +74,12 @@ class LlamaAttention(nn.Module):
 
     def forward(self, x, cos, sin):
         b, s, _ = x.shape
-        q = self.wq(x).view(b, s, self.heads, self.head_dim).transpose(1, 2)
-        k = self.wk(x).view(b, s, self.kv_heads, self.head_dim).transpose(1, 2)
-        v = self.wv(x).view(b, s, self.kv_heads, self.head_dim).transpose(1, 2)
-        q = _apply_rope(q, cos, sin)
-        k = _apply_rope(k, cos, sin)
-        o = F.scaled_dot_product_attention(q, k, v, is_causal=True,
-                                           enable_gqa=True)
+        q = rope(self.wq(x).view(b, s, self.heads, self.head_dim), cos, sin)
+        k = rope(self.wk(x).view(b, s, self.kv_heads, self.head_dim), cos, sin)
+        v = self.wv(x).view(b, s, self.kv_heads, self.head_dim)
+        o = F.scaled_dot_product_attention(
+            q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2),
+            is_causal=True, enable_gqa=True)
         return self.wo(o.transpose(1, 2).reshape(b, s, -1))
 
 
@@ -103,7 +96,7 @@ class LlamaBlock(nn.Module):
     def forward(self, x, cos, sin):
         x = x + self.attn(self.attn_norm(x), cos, sin)
         h = self.ffn_norm(x)
-        return x + self.w2(F.silu(self.w1(h)) * self.w3(h))
+        return x + self.w2(swiglu(self.w1(h), self.w3(h)))
 
 
 class LlamaForCausalLM(nn.Module):

@@ -397,3 +397,42 @@ def test_llama_tiny_step_gpu(hvd):
         losses.append(loss.item())
     assert all(torch.isfinite(torch.tensor(losses)))
     assert losses[-1] < losses[0]
+
+
+def test_fused_rope_matches_eager(hvd):
+    from axon.ops.rotary import _rope_eager, rope
+
+    torch.manual_seed(9)
+    b, s, h, d = 2, 33, 4, 64
+    cos = torch.randn(s, d // 2, device=DEV).clamp(-1, 1)
+    sin = torch.randn(s, d // 2, device=DEV).clamp(-1, 1)
+    x = torch.randn(b, s, h, d, device=DEV)
+    x1 = x.clone().requires_grad_(True)
+    x2 = x.clone().requires_grad_(True)
+    y1 = _rope_eager(x1, cos, sin)
+    y2 = rope(x2, cos, sin)
+    assert torch.allclose(y1, y2, rtol=1e-5, atol=1e-6), (y1 - y2).abs().max()
+    g = torch.randn_like(y1)
+    y1.backward(g)
+    y2.backward(g)
+    assert torch.allclose(x1.grad, x2.grad, rtol=1e-5, atol=1e-6)
+
+
+def test_fused_swiglu_matches_eager(hvd):
+    import torch.nn.functional as F
+
+    from axon.ops.rotary import swiglu
+
+    torch.manual_seed(10)
+    a = torch.randn(1000, 48, device=DEV) * 2
+    g = torch.randn(1000, 48, device=DEV)
+    a1, g1 = a.clone().requires_grad_(True), g.clone().requires_grad_(True)
+    a2, g2 = a.clone().requires_grad_(True), g.clone().requires_grad_(True)
+    z1 = F.silu(a1) * g1
+    z2 = swiglu(a2, g2)
+    assert torch.allclose(z1, z2, rtol=1e-5, atol=1e-6)
+    gr = torch.randn_like(z1)
+    z1.backward(gr)
+    z2.backward(gr)
+    assert torch.allclose(a1.grad, a2.grad, rtol=1e-5, atol=1e-5)
+    assert torch.allclose(g1.grad, g2.grad, rtol=1e-5, atol=1e-6)
