@@ -197,8 +197,8 @@ void register_bn(pybind11::module_& m) {
         py::arg("rstd"));
   m.def("bn_fwd", &bn_fwd, py::arg("x"), py::arg("gamma"), py::arg("beta"),
         py::arg("running_mean"), py::arg("running_var"), py::arg("momentum"),
-        py::arg("eps"), py::arg("training"), py::arg("relu") = false,
-        py::arg("add") = at::Tensor());
+        py::arg("eps"), py::arg("training"), py::arg("relu"),
+        py::arg("add"));
   m.def("bn_bwd", &bn_bwd, py::arg("x"), py::arg("dy"), py::arg("y"),
         py::arg("gamma"), py::arg("save_mean"), py::arg("save_rstd"),
         py::arg("relu") = false, py::arg("want_dadd") = false);

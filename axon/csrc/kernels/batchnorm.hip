@@ -122,15 +122,25 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, int64_t rows, int C,
 }
 
 // Fold the nby partial rows into scratch[0..2C) (deterministic order).
+// Each block owns 64 consecutive columns; its 4 waves sum interleaved
+// partial rows with fully coalesced 256 B loads, then combine in LDS —
+// a one-thread-per-column fold is near-serial for small C (the fold was
+// costing tens of µs per BN call that way).
 __global__ void bn_reduce_partials_kernel(const float* __restrict__ partial,
                                           float* __restrict__ scratch, int twoC,
                                           int nby) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= twoC) return;
+  const int lane = threadIdx.x % 64;
+  const int wave = threadIdx.x / 64;
+  const int c = blockIdx.x * 64 + lane;
   float acc = 0.f;
-  for (int b = 0; b < nby; ++b)
-    acc += partial[static_cast<int64_t>(b) * twoC + c];
-  scratch[c] = acc;
+  if (c < twoC)
+    for (int b = wave; b < nby; b += kWaves)
+      acc += partial[static_cast<int64_t>(b) * twoC + c];
+  __shared__ float lds[kWaves][64];
+  lds[wave][lane] = acc;
+  __syncthreads();
+  if (wave == 0 && c < twoC)
+    scratch[c] = lds[0][lane] + lds[1][lane] + lds[2][lane] + lds[3][lane];
 }
 
 // -------------------------------------------------- finalize (fwd, C-sized)
@@ -390,7 +400,7 @@ void bn_fwd_impl(const void* x, void* y, const void* add, int64_t rows, int C,
       case 2: hipLaunchKernelGGL((bn_stats_kernel<T, 2>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, partial, rpb); break;
       default: hipLaunchKernelGGL((bn_stats_kernel<T, 1>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, partial, rpb); break;
     }
-    hipLaunchKernelGGL(bn_reduce_partials_kernel, (2 * C + 255) / 256, 256, 0,
+    hipLaunchKernelGGL(bn_reduce_partials_kernel, (2 * C + 63) / 64, 256, 0,
                        stream, partial, scratch, 2 * C, nby);
     float inv_n = 1.f / static_cast<float>(rows);
     float unbias = rows > 1 ? static_cast<float>(rows) / (rows - 1) : 1.f;
@@ -470,7 +480,7 @@ void bn_bwd_impl(const void* x, const void* dy, const void* y, void* dx,
     case 2: relu ? launch_reduce(K2{}, BT{}) : launch_reduce(K2{}, BF{}); break;
     default: relu ? launch_reduce(K1{}, BT{}) : launch_reduce(K1{}, BF{}); break;
   }
-  hipLaunchKernelGGL(bn_reduce_partials_kernel, (2 * C + 255) / 256, 256, 0,
+  hipLaunchKernelGGL(bn_reduce_partials_kernel, (2 * C + 63) / 64, 256, 0,
                      stream, partial, scratch, 2 * C, nby);
   hipLaunchKernelGGL((bn_bwd_finalize_kernel<PT>), (C + 255) / 256, 256, 0,
                      stream, scratch, C, 1.f / static_cast<float>(rows),

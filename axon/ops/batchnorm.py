@@ -106,7 +106,7 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
                     x, self.weight if self.weight is not None else empty,
                     self.bias if self.bias is not None else empty,
                     self.running_mean, self.running_var, 0.0, self.eps,
-                    False, self.fuse_relu)
+                    False, self.fuse_relu, empty)
                 return y
         y = super().forward(x)
         return F.relu(y, inplace=True) if self.fuse_relu else y
