@@ -68,6 +68,7 @@ from axon.process_sets import (  # noqa: F401,E402
     process_set_ranks,
     remove_process_set,
 )
+from axon import elastic  # noqa: F401,E402  (hvd.elastic.run / State)
 
 
 def nccl_built():

@@ -6,11 +6,18 @@ per MI355X via HIP_VISIBLE_DEVICES, passes the unix-socket rendezvous path,
 waits on children and kills the whole tree on the first failure
 (fail-fast). No ssh, no cloud, no mpirun.
 
+`--elastic` switches to generation-based membership (SURVEY.md §5.3c,
+axon/elastic.py): rank failures respawn the worker (up to --max-restarts
+per worker) or shrink the world (down to --min-np); survivors recover from
+their last in-memory `state.commit()`.
+
 Usage:
     axonrun -np 8 python train.py [args...]
+    axonrun -np 8 --elastic --min-np 4 python train_elastic.py
     python -m axon.runner.run -np 8 python train.py
 """
 import argparse
+import json
 import os
 import signal
 import subprocess
@@ -34,12 +41,22 @@ def main(argv=None):
     parser.add_argument("--cycle-time-ms", type=float, default=None,
                         help="override AXON_CYCLE_TIME (ms)")
     parser.add_argument("--verbose", "-v", action="store_true")
+    parser.add_argument("--elastic", action="store_true",
+                        help="generation-based membership: recover from rank "
+                             "failures (see axon/elastic.py)")
+    parser.add_argument("--min-np", type=int, default=1,
+                        help="elastic: abort if the world would shrink below")
+    parser.add_argument("--max-restarts", type=int, default=1,
+                        help="elastic: respawns allowed per worker before it "
+                             "is evicted (world shrinks)")
     parser.add_argument("command", nargs=argparse.REMAINDER,
                         help="command to launch per rank")
     args = parser.parse_args(argv)
 
     if not args.command:
         parser.error("no command given")
+    if args.elastic:
+        return run_elastic(args)
     np_ = args.num_proc
     sock = os.path.join(tempfile.gettempdir(), f"axon_{uuid.uuid4().hex[:12]}.sock")
 
@@ -106,6 +123,111 @@ def main(argv=None):
             os.unlink(sock)
         except OSError:
             pass
+    return exit_code
+
+
+def run_elastic(args):
+    np_ = args.num_proc
+    base = os.path.join(tempfile.gettempdir(), f"axon_{uuid.uuid4().hex[:12]}")
+    genfile = base + ".gen"
+    gen = 0
+    members = list(range(np_))  # worker ids, membership order = rank order
+    restarts = {w: 0 for w in members}
+
+    def write_genfile():
+        doc = {"gen": gen, "size": len(members),
+               "socket": f"{base}.g{gen}.sock",
+               "ranks": {str(w): i for i, w in enumerate(members)}}
+        tmp = genfile + ".tmp"
+        with open(tmp, "w") as f:
+            json.dump(doc, f)
+        os.replace(tmp, genfile)
+        if args.verbose:
+            print(f"[axonrun] generation {gen}: size={len(members)} "
+                  f"members={members}", file=sys.stderr)
+
+    procs = {}
+
+    def spawn(worker):
+        env = dict(os.environ)
+        env["AXON_WORKER_ID"] = str(worker)
+        env["AXON_ELASTIC_GENFILE"] = genfile
+        if not args.no_pin:
+            env["HIP_VISIBLE_DEVICES"] = str(worker)
+            env["AXON_PINNED"] = "1"
+        if args.timeline:
+            env["AXON_TIMELINE"] = args.timeline
+        if args.fusion_threshold_mb is not None:
+            env["AXON_FUSION_THRESHOLD"] = str(args.fusion_threshold_mb << 20)
+        if args.cycle_time_ms is not None:
+            env["AXON_CYCLE_TIME"] = str(args.cycle_time_ms)
+        procs[worker] = subprocess.Popen(args.command, env=env,
+                                         start_new_session=True)
+
+    def kill_all(sig=signal.SIGTERM):
+        for p in procs.values():
+            if p.poll() is None:
+                try:
+                    os.killpg(os.getpgid(p.pid), sig)
+                except (ProcessLookupError, PermissionError):
+                    pass
+
+    signal.signal(signal.SIGTERM, lambda s_, f: kill_all(s_))
+    signal.signal(signal.SIGINT, lambda s_, f: kill_all(s_))
+
+    write_genfile()
+    for w in members:
+        spawn(w)
+
+    exit_code = 0
+    try:
+        while True:
+            running = [w for w in members if procs[w].poll() is None]
+            done_ok = [w for w in members if procs[w].poll() == 0]
+            dead = [w for w in members
+                    if procs[w].poll() not in (None, 0)]
+            if done_ok and not dead:
+                if not running:
+                    break  # every member finished cleanly
+            if dead:
+                for w in dead:
+                    rc = procs[w].poll()
+                    print(f"[axonrun] worker {w} exited with code {rc}",
+                          file=sys.stderr)
+                respawn, evict = [], []
+                for w in dead:
+                    if restarts[w] < args.max_restarts:
+                        restarts[w] += 1
+                        respawn.append(w)
+                    else:
+                        evict.append(w)
+                members = [w for w in members if w not in evict]
+                if len(members) < args.min_np:
+                    print(f"[axonrun] world would shrink below --min-np "
+                          f"{args.min_np} — aborting", file=sys.stderr)
+                    exit_code = 1
+                    kill_all()
+                    break
+                gen += 1
+                write_genfile()
+                for w in respawn:
+                    spawn(w)
+                if args.verbose and evict:
+                    print(f"[axonrun] evicted workers {evict}", file=sys.stderr)
+            time.sleep(0.1)
+    finally:
+        time.sleep(0.2)
+        kill_all(signal.SIGKILL)
+        for suffix in [".gen", ".gen.tmp"]:
+            try:
+                os.unlink(base + suffix)
+            except OSError:
+                pass
+        for g in range(gen + 1):
+            try:
+                os.unlink(f"{base}.g{g}.sock")
+            except OSError:
+                pass
     return exit_code
 
 

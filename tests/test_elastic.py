@@ -1,0 +1,58 @@
+"""Elastic recovery: fault-injection through the real launcher (SURVEY.md
+§5.3c): a worker is killed mid-training; the job must continue with the
+surviving ranks from the last in-memory commit."""
+import json
+import os
+import subprocess
+import sys
+import tempfile
+import uuid
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+SCRIPT = os.path.join(REPO, "tests", "elastic_script.py")
+
+
+def _run_elastic(np_, extra, out):
+    cmd = [sys.executable, "-m", "axon.runner.run", "-np", str(np_),
+           "--elastic", "--no-pin", "-v", *extra,
+           sys.executable, SCRIPT, out]
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["OMP_NUM_THREADS"] = "1"
+    return subprocess.run(cmd, env=env, capture_output=True, text=True,
+                          timeout=180)
+
+
+def test_elastic_shrink_on_failure(tmp_path):
+    """max-restarts 0: worker 2's death shrinks the world 3 -> 2; the
+    survivors finish all steps with identical weights."""
+    out = str(tmp_path / f"el_{uuid.uuid4().hex[:8]}")
+    r = _run_elastic(3, ["--min-np", "2", "--max-restarts", "0"], out)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert os.path.exists(out + ".died")  # fault actually injected
+    assert not os.path.exists(out + ".w2.json")  # evicted worker wrote nothing
+    res = {}
+    for w in (0, 1):
+        with open(f"{out}.w{w}.json") as f:
+            res[w] = json.load(f)
+    assert res[0]["size"] == 2 and res[1]["size"] == 2
+    assert res[0]["step"] == 12 and res[1]["step"] == 12
+    assert abs(res[0]["wsum"] - res[1]["wsum"]) < 1e-9
+    assert "generation 1" in r.stderr
+
+
+def test_elastic_respawn_on_failure(tmp_path):
+    """max-restarts 1: the dead worker is respawned, world stays 3, and
+    the respawned (state-less) worker resyncs from rank 0."""
+    out = str(tmp_path / f"el_{uuid.uuid4().hex[:8]}")
+    r = _run_elastic(3, ["--min-np", "3", "--max-restarts", "1"], out)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert os.path.exists(out + ".died")
+    res = {}
+    for w in (0, 1, 2):
+        with open(f"{out}.w{w}.json") as f:
+            res[w] = json.load(f)
+    for w in (0, 1, 2):
+        assert res[w]["size"] == 3
+        assert res[w]["step"] == 12
+    assert abs(res[0]["wsum"] - res[2]["wsum"]) < 1e-9
