@@ -132,12 +132,15 @@ std::vector<at::Tensor> rms_bwd(const at::Tensor& x, const at::Tensor& dy_in,
   const int64_t rows = x.numel() / H;
   at::Tensor dy = dy_in.is_contiguous() ? dy_in : dy_in.contiguous();
   auto fopt = at::TensorOptions().dtype(at::kFloat).device(x.device());
-  at::Tensor dw = at::zeros({H}, fopt);
+  at::Tensor dw = at::empty({H}, fopt);
+  at::Tensor dwpart = at::empty(
+      {static_cast<int64_t>(H) * rms_bwd_blocks(rows, nullptr)}, fopt);
   at::Tensor dx = at::empty_like(x);
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(x.get_device());
   launch_rms_bwd(static_cast<int>(x.scalar_type()), x.data_ptr(), dy.data_ptr(),
                  w.data_ptr(), rstd.data_ptr<float>(), dx.data_ptr(),
-                 dw.data_ptr<float>(), rows, H, stream.stream());
+                 dwpart.data_ptr<float>(), dw.data_ptr<float>(), rows, H,
+                 stream.stream());
   return {dx, dw};
 }
 

@@ -56,11 +56,12 @@ struct alignas(sizeof(T) * V) BVec {
 };
 
 // ------------------------------------------------------ stats (fwd pass 1)
-// partial[blockIdx.y][c] = sum over this block's rows of x[:, c];
-// partial[blockIdx.y][C + c] = sum of x[:, c]^2. No atomics: a second tiny
-// kernel (bn_reduce_partials) folds the nby partial rows — deterministic
-// and free of same-address atomic serialization (which measured 3-6x off
-// streaming SOL with ~2k blocks per channel).
+// partial is COLUMN-major [2C][nby]: partial[c*nby + blockIdx.y] = this
+// block's row-range sum of x[:, c]; partial[(C+c)*nby + blockIdx.y] the
+// sum of squares. No atomics (same-address atomicAdd measured 3-6x off
+// streaming SOL with ~2k blocks per channel), and the per-channel partials
+// are CONTIGUOUS so the fold kernel reduces them with coalesced vector
+// loads — deterministic both stages.
 
 template <typename T, int VC>
 __global__ void bn_stats_kernel(const T* __restrict__ x, int64_t rows, int C,
@@ -103,7 +104,7 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, int64_t rows, int C,
     lds[wave][64 * VC + lane * VC + k] = q[k];
   }
   __syncthreads();
-  float* out = partial + static_cast<int64_t>(blockIdx.y) * 2 * C;
+  const int nby = gridDim.y;
   if (wave == 0) {
 #pragma unroll
     for (int k = 0; k < VC; ++k) {
@@ -115,32 +116,30 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, int64_t rows, int C,
         ss += lds[w][lane * VC + k];
         qq += lds[w][64 * VC + lane * VC + k];
       }
-      out[c] = ss;
-      out[C + c] = qq;
+      partial[static_cast<int64_t>(c) * nby + blockIdx.y] = ss;
+      partial[static_cast<int64_t>(C + c) * nby + blockIdx.y] = qq;
     }
   }
 }
 
-// Fold the nby partial rows into scratch[0..2C) (deterministic order).
-// Each block owns 64 consecutive columns; its 4 waves sum interleaved
-// partial rows with fully coalesced 256 B loads, then combine in LDS —
-// a one-thread-per-column fold is near-serial for small C (the fold was
-// costing tens of µs per BN call that way).
+// Fold each channel's contiguous nby partials into scratch[c]
+// (deterministic order). One block per channel: 256 threads stream the
+// column with coalesced loads, then a shuffle+LDS tree.
 __global__ void bn_reduce_partials_kernel(const float* __restrict__ partial,
-                                          float* __restrict__ scratch, int twoC,
+                                          float* __restrict__ scratch,
                                           int nby) {
-  const int lane = threadIdx.x % 64;
-  const int wave = threadIdx.x / 64;
-  const int c = blockIdx.x * 64 + lane;
+  const float* col = partial + static_cast<int64_t>(blockIdx.x) * nby;
   float acc = 0.f;
-  if (c < twoC)
-    for (int b = wave; b < nby; b += kWaves)
-      acc += partial[static_cast<int64_t>(b) * twoC + c];
-  __shared__ float lds[kWaves][64];
-  lds[wave][lane] = acc;
+  for (int b = threadIdx.x; b < nby; b += static_cast<int>(blockDim.x))
+    acc += col[b];
+  for (int off = 32; off > 0; off >>= 1)
+    acc += __shfl_xor(acc, off, 64);
+  __shared__ float red[kWaves];
+  const int lane = threadIdx.x % 64, wave = threadIdx.x / 64;
+  if (lane == 0) red[wave] = acc;
   __syncthreads();
-  if (wave == 0 && c < twoC)
-    scratch[c] = lds[0][lane] + lds[1][lane] + lds[2][lane] + lds[3][lane];
+  if (threadIdx.x == 0)
+    scratch[blockIdx.x] = red[0] + red[1] + red[2] + red[3];
 }
 
 // -------------------------------------------------- finalize (fwd, C-sized)
@@ -276,7 +275,7 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
     lds[wave][64 * VC + lane * VC + k] = sxdy[k];
   }
   __syncthreads();
-  float* out = partial + static_cast<int64_t>(blockIdx.y) * 2 * C;
+  const int nby = gridDim.y;
   if (wave == 0) {
 #pragma unroll
     for (int k = 0; k < VC; ++k) {
@@ -288,8 +287,8 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
         a += lds[w][lane * VC + k];
         bsum += lds[w][64 * VC + lane * VC + k];
       }
-      out[c] = a;
-      out[C + c] = bsum;
+      partial[static_cast<int64_t>(c) * nby + blockIdx.y] = a;
+      partial[static_cast<int64_t>(C + c) * nby + blockIdx.y] = bsum;
     }
   }
 }
@@ -400,8 +399,8 @@ void bn_fwd_impl(const void* x, void* y, const void* add, int64_t rows, int C,
       case 2: hipLaunchKernelGGL((bn_stats_kernel<T, 2>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, partial, rpb); break;
       default: hipLaunchKernelGGL((bn_stats_kernel<T, 1>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, partial, rpb); break;
     }
-    hipLaunchKernelGGL(bn_reduce_partials_kernel, (2 * C + 63) / 64, 256, 0,
-                       stream, partial, scratch, 2 * C, nby);
+    hipLaunchKernelGGL(bn_reduce_partials_kernel, 2 * C, 256, 0,
+                       stream, partial, scratch, nby);
     float inv_n = 1.f / static_cast<float>(rows);
     float unbias = rows > 1 ? static_cast<float>(rows) / (rows - 1) : 1.f;
     hipLaunchKernelGGL((bn_fwd_finalize_kernel<PT>), (C + 255) / 256, 256, 0,
@@ -480,8 +479,8 @@ void bn_bwd_impl(const void* x, const void* dy, const void* y, void* dx,
     case 2: relu ? launch_reduce(K2{}, BT{}) : launch_reduce(K2{}, BF{}); break;
     default: relu ? launch_reduce(K1{}, BT{}) : launch_reduce(K1{}, BF{}); break;
   }
-  hipLaunchKernelGGL(bn_reduce_partials_kernel, (2 * C + 63) / 64, 256, 0,
-                     stream, partial, scratch, 2 * C, nby);
+  hipLaunchKernelGGL(bn_reduce_partials_kernel, 2 * C, 256, 0,
+                     stream, partial, scratch, nby);
   hipLaunchKernelGGL((bn_bwd_finalize_kernel<PT>), (C + 255) / 256, 256, 0,
                      stream, scratch, C, 1.f / static_cast<float>(rows),
                      static_cast<const PT*>(gamma), save_mean, save_rstd,

@@ -68,15 +68,17 @@ void launch_bn_bwd(int x_dtype, int param_dtype, const void* x, const void* dy,
                    hipStream_t stream);
 
 // Fused RMSNorm (rmsnorm.hip): x/y/dx are [rows, H] contiguous, w is [H]
-// (same dtype as x), rstd fp32[rows], dw a ZEROED fp32[H] accumulator.
+// (same dtype as x), rstd fp32[rows]; dwpart is an uninitialized fp32
+// [H * rms_bwd_blocks(rows)] partial buffer, dw fp32[H] (overwritten).
 // Caller guarantees H % 2 == 0 and H <= 16384 (register/LDS bounds) —
 // the Python wrapper falls back to eager otherwise.
 void launch_rms_fwd(int dtype, const void* x, const void* w, void* y,
                     float* rstd, int64_t rows, int H, float eps,
                     hipStream_t stream);
+int rms_bwd_blocks(int64_t rows, int* rows_per_block);
 void launch_rms_bwd(int dtype, const void* x, const void* dy, const void* w,
-                    const float* rstd, void* dx, float* dw, int64_t rows,
-                    int H, hipStream_t stream);
+                    const float* rstd, void* dx, float* dwpart, float* dw,
+                    int64_t rows, int H, hipStream_t stream);
 
 // Rotary embedding (elementwise.hip): x/y are [B,S,H,D] contiguous (numel
 // elements), cos/sin [S, D/2] in the same dtype; backward = transposed

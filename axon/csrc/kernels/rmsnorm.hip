@@ -112,8 +112,11 @@ __global__ void rms_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
 // ----------------------------------------------------------------- backward
 // Each block owns a contiguous row range; x/dy staged per row in LDS;
 // per-thread dw partials accumulate across the block's rows in registers
-// (thread t owns columns {t*VC + j*kThreads*VC}), one atomicAdd sweep at
-// the end. kMaxDwChunks bounds the register array: H <= 32768 at VC=8.
+// (thread t owns columns {t*VC + j*kThreads*VC}). Each block then stores
+// its partials into a COLUMN-major [H][nblocks] buffer — no atomics
+// (same-address atomicAdd from ~2k blocks measured 40x off SOL) — and a
+// per-column fold kernel produces dw. kMaxDwChunks bounds the register
+// array: H <= 16384 at VC=8.
 
 constexpr int kMaxDwChunks = 8;  // bounds dw registers: H <= 16384 at VC=8
 
@@ -121,7 +124,7 @@ template <typename T, int VC>
 __global__ void rms_bwd_kernel(const T* __restrict__ x, const T* __restrict__ dy,
                                const T* __restrict__ w,
                                const float* __restrict__ rstd_in,
-                               T* __restrict__ dx, float* __restrict__ dw,
+                               T* __restrict__ dx, float* __restrict__ dwpart,
                                int64_t rows, int H, int rows_per_block) {
   extern __shared__ unsigned char smem[];
   T* xrow = reinterpret_cast<T*>(smem);
@@ -174,10 +177,44 @@ __global__ void rms_bwd_kernel(const T* __restrict__ x, const T* __restrict__ dy
     __syncthreads();
   }
 
+  const int nblocks = gridDim.x;
   for (int i = threadIdx.x, j = 0; i < hv; i += kThreads, ++j)
 #pragma unroll
-    for (int k = 0; k < VC; ++k) atomicAdd(&dw[i * VC + k], dwacc[j][k]);
+    for (int k = 0; k < VC; ++k)
+      dwpart[static_cast<int64_t>(i * VC + k) * nblocks + blockIdx.x] =
+          dwacc[j][k];
 }
+
+// Fold each column's contiguous nblocks partials into dw[col].
+__global__ void rms_fold_kernel(const float* __restrict__ dwpart,
+                                float* __restrict__ dw, int nblocks) {
+  const float* col = dwpart + static_cast<int64_t>(blockIdx.x) * nblocks;
+  float acc = 0.f;
+  for (int b = threadIdx.x; b < nblocks; b += static_cast<int>(blockDim.x))
+    acc += col[b];
+  for (int off = 32; off > 0; off >>= 1)
+    acc += __shfl_xor(acc, off, 64);
+  __shared__ float red[kWaves];
+  const int lane = threadIdx.x % 64, wave = threadIdx.x / 64;
+  if (lane == 0) red[wave] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0)
+    dw[blockIdx.x] = red[0] + red[1] + red[2] + red[3];
+}
+
+}  // namespace
+
+// Number of row-blocks the backward uses (callers size the dw partial
+// buffer as fp32[H * rms_bwd_blocks(rows, ...)]). ~1024 blocks fills the
+// chip; each handles >= 4 rows so the per-block epilogue stays amortized.
+int rms_bwd_blocks(int64_t rows, int* rows_per_block) {
+  int64_t rpb = std::max<int64_t>(4, (rows + 1023) / 1024);
+  int blocks = static_cast<int>((rows + rpb - 1) / rpb);
+  if (rows_per_block) *rows_per_block = static_cast<int>(rpb);
+  return blocks;
+}
+
+namespace {
 
 int rms_vc(int H, size_t /*tsize*/) {
   if (H % 8 == 0) return 8;
@@ -202,20 +239,22 @@ void rms_fwd_t(const void* x, const void* w, void* y, float* rstd, int64_t rows,
 
 template <typename T>
 void rms_bwd_t(const void* x, const void* dy, const void* w, const float* rstd,
-               void* dx, float* dw, int64_t rows, int H, hipStream_t stream) {
+               void* dx, float* dwpart, float* dw, int64_t rows, int H,
+               hipStream_t stream) {
   int vc = rms_vc(H, sizeof(T));
   // Bound the per-thread dw register array: chunks <= kMaxDwChunks.
   while ((H / vc + kThreads - 1) / kThreads > kMaxDwChunks && vc < 8) vc *= 2;
   size_t lds = 2 * static_cast<size_t>(H) * sizeof(T);
-  // ~8 rows per block balances dw atomic traffic vs fill.
-  int rpb = static_cast<int>(std::max<int64_t>(1, rows / 2048));
-  int blocks = static_cast<int>((rows + rpb - 1) / rpb);
+  int rpb = 0;
+  int blocks = rms_bwd_blocks(rows, &rpb);
   switch (vc) {
-    case 8: hipLaunchKernelGGL((rms_bwd_kernel<T, 8>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), rstd, static_cast<T*>(dx), dw, rows, H, rpb); break;
-    case 4: hipLaunchKernelGGL((rms_bwd_kernel<T, 4>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), rstd, static_cast<T*>(dx), dw, rows, H, rpb); break;
-    case 2: hipLaunchKernelGGL((rms_bwd_kernel<T, 2>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), rstd, static_cast<T*>(dx), dw, rows, H, rpb); break;
-    default: hipLaunchKernelGGL((rms_bwd_kernel<T, 1>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), rstd, static_cast<T*>(dx), dw, rows, H, rpb); break;
+    case 8: hipLaunchKernelGGL((rms_bwd_kernel<T, 8>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), rstd, static_cast<T*>(dx), dwpart, rows, H, rpb); break;
+    case 4: hipLaunchKernelGGL((rms_bwd_kernel<T, 4>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), rstd, static_cast<T*>(dx), dwpart, rows, H, rpb); break;
+    case 2: hipLaunchKernelGGL((rms_bwd_kernel<T, 2>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), rstd, static_cast<T*>(dx), dwpart, rows, H, rpb); break;
+    default: hipLaunchKernelGGL((rms_bwd_kernel<T, 1>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), rstd, static_cast<T*>(dx), dwpart, rows, H, rpb); break;
   }
+  hipLaunchKernelGGL(rms_fold_kernel, H, kThreads, 0, stream, dwpart, dw,
+                     blocks);
 }
 
 }  // namespace
@@ -231,12 +270,12 @@ void launch_rms_fwd(int dtype, const void* x, const void* w, void* y,
 }
 
 void launch_rms_bwd(int dtype, const void* x, const void* dy, const void* w,
-                    const float* rstd, void* dx, float* dw, int64_t rows, int H,
-                    hipStream_t stream) {
+                    const float* rstd, void* dx, float* dwpart, float* dw,
+                    int64_t rows, int H, hipStream_t stream) {
   switch (dtype) {
-    case kFloatId: rms_bwd_t<float>(x, dy, w, rstd, dx, dw, rows, H, stream); break;
-    case kHalfId: rms_bwd_t<__half>(x, dy, w, rstd, dx, dw, rows, H, stream); break;
-    case kBF16Id: rms_bwd_t<__hip_bfloat16>(x, dy, w, rstd, dx, dw, rows, H, stream); break;
+    case kFloatId: rms_bwd_t<float>(x, dy, w, rstd, dx, dwpart, dw, rows, H, stream); break;
+    case kHalfId: rms_bwd_t<__half>(x, dy, w, rstd, dx, dwpart, dw, rows, H, stream); break;
+    case kBF16Id: rms_bwd_t<__hip_bfloat16>(x, dy, w, rstd, dx, dwpart, dw, rows, H, stream); break;
   }
 }
 
