@@ -63,12 +63,19 @@ struct alignas(sizeof(T) * V) BVec {
 // are CONTIGUOUS so the fold kernel reduces them with coalesced vector
 // loads — deterministic both stages.
 
-template <typename T, int VC>
+// W = rows covered by ONE wave load iteration (64*VC = C*W when W > 1):
+// for small C the wave's 64 VC-wide lanes span W whole rows, keeping the
+// per-wave transaction at 64*VC elements (1 KiB at VC=8/bf16) instead of
+// a narrow 2*C-byte strip. Lane layout: flat = lane*VC; its channels are
+// flat % C .. +VC-1 (fixed per lane) at row offset flat / C.
+template <typename T, int VC, int W>
 __global__ void bn_stats_kernel(const T* __restrict__ x, int64_t rows, int C,
                                 float* __restrict__ partial, int rows_per_block) {
   const int lane = threadIdx.x % 64;
   const int wave = threadIdx.x / 64;
-  const int c0 = blockIdx.x * 64 * VC + lane * VC;  // my first channel
+  const int flat = lane * VC;
+  const int row_off = W > 1 ? flat / C : 0;
+  const int c0 = W > 1 ? flat % C : blockIdx.x * 64 * VC + flat;
   const int64_t r0 = static_cast<int64_t>(blockIdx.y) * rows_per_block;
   const int64_t r1 = r0 + rows_per_block > rows ? rows : r0 + rows_per_block;
 
@@ -76,7 +83,17 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, int64_t rows, int C,
 #pragma unroll
   for (int k = 0; k < VC; ++k) s[k] = q[k] = 0.f;
 
-  if (c0 + VC <= C) {
+  if (W > 1) {
+    for (int64_t r = r0 + wave * W + row_off; r < r1; r += kWaves * W) {
+      BVec<T, VC> xv = *reinterpret_cast<const BVec<T, VC>*>(x + r * C + c0);
+#pragma unroll
+      for (int k = 0; k < VC; ++k) {
+        float v = bf_of(xv.v[k]);
+        s[k] += v;
+        q[k] += v * v;
+      }
+    }
+  } else if (c0 + VC <= C) {
     for (int64_t r = r0 + wave; r < r1; r += kWaves) {
       BVec<T, VC> xv = *reinterpret_cast<const BVec<T, VC>*>(x + r * C + c0);
 #pragma unroll
@@ -96,26 +113,29 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, int64_t rows, int C,
     }
   }
 
-  // Cross-wave combine in LDS, then one coalesced partial-row store.
+  // Cross-wave (and, for W>1, cross-row-group) combine in LDS, then one
+  // coalesced partial store.
   __shared__ float lds[kWaves][64 * VC * 2];
 #pragma unroll
   for (int k = 0; k < VC; ++k) {
-    lds[wave][lane * VC + k] = s[k];
-    lds[wave][64 * VC + lane * VC + k] = q[k];
+    lds[wave][flat + k] = s[k];
+    lds[wave][64 * VC + flat + k] = q[k];
   }
   __syncthreads();
   const int nby = gridDim.y;
-  if (wave == 0) {
+  if (wave == 0 && flat < (W > 1 ? C : 64 * VC)) {
 #pragma unroll
     for (int k = 0; k < VC; ++k) {
       int c = c0 + k;
       if (c >= C) break;
       float ss = 0.f, qq = 0.f;
 #pragma unroll
-      for (int w = 0; w < kWaves; ++w) {
-        ss += lds[w][lane * VC + k];
-        qq += lds[w][64 * VC + lane * VC + k];
-      }
+      for (int w = 0; w < kWaves; ++w)
+#pragma unroll
+        for (int m = 0; m < W; ++m) {
+          ss += lds[w][m * C + flat + k];
+          qq += lds[w][64 * VC + m * C + flat + k];
+        }
       partial[static_cast<int64_t>(c) * nby + blockIdx.y] = ss;
       partial[static_cast<int64_t>(C + c) * nby + blockIdx.y] = qq;
     }
@@ -222,7 +242,7 @@ __global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ y,
 // scratch[c] += sum(dy_eff[:, c]); scratch[C+c] += sum(dy_eff * (x - mean)).
 // RELU fusion: dy_eff = (y > 0) ? dy : 0 (y = the module's fused output).
 
-template <typename T, int VC, bool RELU>
+template <typename T, int VC, int W, bool RELU>
 __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
                                      const T* __restrict__ dy,
                                      const T* __restrict__ y,  // RELU only
@@ -232,7 +252,9 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
                                      int rows_per_block) {
   const int lane = threadIdx.x % 64;
   const int wave = threadIdx.x / 64;
-  const int c0 = blockIdx.x * 64 * VC + lane * VC;
+  const int flat = lane * VC;
+  const int row_off = W > 1 ? flat / C : 0;
+  const int c0 = W > 1 ? flat % C : blockIdx.x * 64 * VC + flat;
   const int64_t r0 = static_cast<int64_t>(blockIdx.y) * rows_per_block;
   const int64_t r1 = r0 + rows_per_block > rows ? rows : r0 + rows_per_block;
 
@@ -243,7 +265,21 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
     mean[k] = (c0 + k < C) ? save_mean[c0 + k] : 0.f;
   }
 
-  if (c0 + VC <= C) {
+  if (W > 1) {
+    for (int64_t r = r0 + wave * W + row_off; r < r1; r += kWaves * W) {
+      BVec<T, VC> xv = *reinterpret_cast<const BVec<T, VC>*>(x + r * C + c0);
+      BVec<T, VC> dv = *reinterpret_cast<const BVec<T, VC>*>(dy + r * C + c0);
+      BVec<T, VC> yv;
+      if (RELU) yv = *reinterpret_cast<const BVec<T, VC>*>(y + r * C + c0);
+#pragma unroll
+      for (int k = 0; k < VC; ++k) {
+        float d = bf_of(dv.v[k]);
+        if (RELU && bf_of(yv.v[k]) <= 0.f) d = 0.f;
+        sdy[k] += d;
+        sxdy[k] += d * (bf_of(xv.v[k]) - mean[k]);
+      }
+    }
+  } else if (c0 + VC <= C) {
     for (int64_t r = r0 + wave; r < r1; r += kWaves) {
       BVec<T, VC> xv = *reinterpret_cast<const BVec<T, VC>*>(x + r * C + c0);
       BVec<T, VC> dv = *reinterpret_cast<const BVec<T, VC>*>(dy + r * C + c0);
@@ -271,22 +307,24 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
   __shared__ float lds[kWaves][64 * VC * 2];
 #pragma unroll
   for (int k = 0; k < VC; ++k) {
-    lds[wave][lane * VC + k] = sdy[k];
-    lds[wave][64 * VC + lane * VC + k] = sxdy[k];
+    lds[wave][flat + k] = sdy[k];
+    lds[wave][64 * VC + flat + k] = sxdy[k];
   }
   __syncthreads();
   const int nby = gridDim.y;
-  if (wave == 0) {
+  if (wave == 0 && flat < (W > 1 ? C : 64 * VC)) {
 #pragma unroll
     for (int k = 0; k < VC; ++k) {
       int c = c0 + k;
       if (c >= C) break;
       float a = 0.f, bsum = 0.f;
 #pragma unroll
-      for (int w = 0; w < kWaves; ++w) {
-        a += lds[w][lane * VC + k];
-        bsum += lds[w][64 * VC + lane * VC + k];
-      }
+      for (int w = 0; w < kWaves; ++w)
+#pragma unroll
+        for (int m = 0; m < W; ++m) {
+          a += lds[w][m * C + flat + k];
+          bsum += lds[w][64 * VC + m * C + flat + k];
+        }
       partial[static_cast<int64_t>(c) * nby + blockIdx.y] = a;
       partial[static_cast<int64_t>(C + c) * nby + blockIdx.y] = bsum;
     }
@@ -365,6 +403,21 @@ int pick_vc(int C) {
   return 1;  // 64-channel tiles with ragged-tail bounds checks
 }
 
+// Rows per wave iteration for the reduce kernels: small C gets W whole
+// rows per 1 KiB wave transaction (64*8 == C*W).
+int pick_w(int C) {
+  if (C % 8 == 0 && C <= 512 && 512 % C == 0) return 512 / C;
+  return 1;
+}
+
+// Vector width for the APPLY kernels (flat elementwise): only needs VC|C.
+int pick_apply_vc(int C) {
+  if (C % 8 == 0) return 8;
+  if (C % 4 == 0) return 4;
+  if (C % 2 == 0) return 2;
+  return 1;
+}
+
 }  // namespace
 
 int bn_partial_rows(int64_t rows, int C) {
@@ -386,18 +439,22 @@ void bn_fwd_impl(const void* x, void* y, const void* add, int64_t rows, int C,
                  void* running_var, float momentum, float eps, bool training,
                  bool relu, float* partial, float* scratch, float* save_mean,
                  float* save_rstd, float* ab, hipStream_t stream) {
-  int vc = pick_vc(C);
+  const int w = pick_w(C);
+  const int vc = w > 1 ? 8 : pick_vc(C);
   if (training) {
-    int ctiles = (C + 64 * vc - 1) / (64 * vc);
+    int ctiles = w > 1 ? 1 : (C + 64 * vc - 1) / (64 * vc);
     int nby = bn_partial_rows(rows, C);
     int rpb = static_cast<int>((rows + nby - 1) / nby);
     nby = static_cast<int>((rows + rpb - 1) / rpb);
     dim3 grid(ctiles, static_cast<unsigned>(nby));
-    switch (vc) {
-      case 8: hipLaunchKernelGGL((bn_stats_kernel<T, 8>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, partial, rpb); break;
-      case 4: hipLaunchKernelGGL((bn_stats_kernel<T, 4>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, partial, rpb); break;
-      case 2: hipLaunchKernelGGL((bn_stats_kernel<T, 2>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, partial, rpb); break;
-      default: hipLaunchKernelGGL((bn_stats_kernel<T, 1>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, partial, rpb); break;
+    switch (w > 1 ? 8 + w : vc) {
+      case 16: hipLaunchKernelGGL((bn_stats_kernel<T, 8, 8>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, partial, rpb); break;
+      case 12: hipLaunchKernelGGL((bn_stats_kernel<T, 8, 4>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, partial, rpb); break;
+      case 10: hipLaunchKernelGGL((bn_stats_kernel<T, 8, 2>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, partial, rpb); break;
+      case 8: hipLaunchKernelGGL((bn_stats_kernel<T, 8, 1>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, partial, rpb); break;
+      case 4: hipLaunchKernelGGL((bn_stats_kernel<T, 4, 1>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, partial, rpb); break;
+      case 2: hipLaunchKernelGGL((bn_stats_kernel<T, 2, 1>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, partial, rpb); break;
+      default: hipLaunchKernelGGL((bn_stats_kernel<T, 1, 1>), grid, kThreads, 0, stream, static_cast<const T*>(x), rows, C, partial, rpb); break;
     }
     hipLaunchKernelGGL(bn_reduce_partials_kernel, 2 * C, 256, 0,
                        stream, partial, scratch, nby);
@@ -415,7 +472,8 @@ void bn_fwd_impl(const void* x, void* y, const void* add, int64_t rows, int C,
                        static_cast<const PT*>(running_mean),
                        static_cast<const PT*>(running_var), ab);
   }
-  int64_t nvec = rows * C / vc;
+  const int vca = pick_apply_vc(C);
+  int64_t nvec = rows * C / vca;
   int blocks = static_cast<int>(std::min<int64_t>(
       (nvec + kThreads - 1) / kThreads, 16384));
   const bool has_add = add != nullptr;
@@ -440,7 +498,7 @@ void bn_fwd_impl(const void* x, void* y, const void* add, int64_t rows, int C,
       has_add ? launch_apply(vctag, BF{}, BT{}) : launch_apply(vctag, BF{}, BF{});
     }
   };
-  switch (vc) {
+  switch (vca) {
     case 8: dispatch(K8{}); break;
     case 4: dispatch(K4{}); break;
     case 2: dispatch(K2{}); break;
@@ -454,8 +512,9 @@ void bn_bwd_impl(const void* x, const void* dy, const void* y, void* dx,
                  const float* save_mean, const float* save_rstd, bool relu,
                  float* partial, float* scratch, float* dgamma, float* dbeta,
                  float* coef, hipStream_t stream) {
-  int vc = pick_vc(C);
-  int ctiles = (C + 64 * vc - 1) / (64 * vc);
+  const int w = pick_w(C);
+  const int vc = w > 1 ? 8 : pick_vc(C);
+  int ctiles = w > 1 ? 1 : (C + 64 * vc - 1) / (64 * vc);
   int nby = bn_partial_rows(rows, C);
   int rpb = static_cast<int>((rows + nby - 1) / nby);
   nby = static_cast<int>((rows + rpb - 1) / rpb);
@@ -466,18 +525,22 @@ void bn_bwd_impl(const void* x, const void* dy, const void* y, void* dx,
   using K8 = std::integral_constant<int, 8>;
   using BT = std::true_type;
   using BF = std::false_type;
-  auto launch_reduce = [&](auto vctag, auto relutag) {
+  auto launch_reduce = [&](auto vctag, auto wtag, auto relutag) {
     constexpr int V = decltype(vctag)::value;
+    constexpr int WW = decltype(wtag)::value;
     constexpr bool R = decltype(relutag)::value;
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, V, R>), grid, kThreads, 0,
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, V, WW, R>), grid, kThreads, 0,
                        stream, static_cast<const T*>(x), static_cast<const T*>(dy),
                        static_cast<const T*>(y), rows, C, save_mean, partial, rpb);
   };
-  switch (vc) {
-    case 8: relu ? launch_reduce(K8{}, BT{}) : launch_reduce(K8{}, BF{}); break;
-    case 4: relu ? launch_reduce(K4{}, BT{}) : launch_reduce(K4{}, BF{}); break;
-    case 2: relu ? launch_reduce(K2{}, BT{}) : launch_reduce(K2{}, BF{}); break;
-    default: relu ? launch_reduce(K1{}, BT{}) : launch_reduce(K1{}, BF{}); break;
+  switch (w > 1 ? 8 + w : vc) {
+    case 16: relu ? launch_reduce(K8{}, K8{}, BT{}) : launch_reduce(K8{}, K8{}, BF{}); break;
+    case 12: relu ? launch_reduce(K8{}, K4{}, BT{}) : launch_reduce(K8{}, K4{}, BF{}); break;
+    case 10: relu ? launch_reduce(K8{}, K2{}, BT{}) : launch_reduce(K8{}, K2{}, BF{}); break;
+    case 8: relu ? launch_reduce(K8{}, K1{}, BT{}) : launch_reduce(K8{}, K1{}, BF{}); break;
+    case 4: relu ? launch_reduce(K4{}, K1{}, BT{}) : launch_reduce(K4{}, K1{}, BF{}); break;
+    case 2: relu ? launch_reduce(K2{}, K1{}, BT{}) : launch_reduce(K2{}, K1{}, BF{}); break;
+    default: relu ? launch_reduce(K1{}, K1{}, BT{}) : launch_reduce(K1{}, K1{}, BF{}); break;
   }
   hipLaunchKernelGGL(bn_reduce_partials_kernel, 2 * C, 256, 0,
                      stream, partial, scratch, nby);
@@ -485,7 +548,8 @@ void bn_bwd_impl(const void* x, const void* dy, const void* y, void* dx,
                      stream, scratch, C, 1.f / static_cast<float>(rows),
                      static_cast<const PT*>(gamma), save_mean, save_rstd,
                      dgamma, dbeta, coef);
-  int64_t nvec = rows * C / vc;
+  const int vca = pick_apply_vc(C);
+  int64_t nvec = rows * C / vca;
   int blocks = static_cast<int>(std::min<int64_t>(
       (nvec + kThreads - 1) / kThreads, 16384));
   const bool has_dadd = dadd != nullptr;
@@ -505,7 +569,7 @@ void bn_bwd_impl(const void* x, const void* dy, const void* y, void* dx,
       has_dadd ? launch_apply(vctag, BF{}, BT{}) : launch_apply(vctag, BF{}, BF{});
     }
   };
-  switch (vc) {
+  switch (vca) {
     case 8: dispatch(K8{}); break;
     case 4: dispatch(K4{}); break;
     case 2: dispatch(K2{}); break;

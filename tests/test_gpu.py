@@ -185,7 +185,8 @@ def test_fused_sgd_matches_torch_gpu(hvd):
 
 @pytest.mark.parametrize("dtype,tol", [(torch.float32, 2e-4),
                                        (torch.bfloat16, 3e-2)])
-@pytest.mark.parametrize("C,HW", [(64, 14), (128, 9), (256, 7), (96, 5)])
+@pytest.mark.parametrize("C,HW", [(64, 14), (128, 9), (256, 7), (96, 5),
+                                  (512, 4)])
 def test_fused_batchnorm_matches_torch(hvd, dtype, tol, C, HW):
     """Fused NHWC BN (kernels/batchnorm.hip) vs stock nn.BatchNorm2d:
     fwd output, input/weight/bias grads, and running stats."""
