@@ -45,6 +45,9 @@ def build_resnet(args, device, dtype):
     from axon.models.resnet import resnet50
     from axon.optim import FusedSGD
 
+    # MIOpen immediate-find: benchmark mode picks the fastest conv algo per
+    # shape during warmup (shapes are static here).
+    torch.backends.cudnn.benchmark = True
     torch.manual_seed(1234)
     model = resnet50().to(device=device, dtype=dtype)
     if device.type == "cuda":
@@ -65,7 +68,7 @@ def build_resnet(args, device, dtype):
     y = torch.randint(0, 1000, (batch,), generator=g).to(device)
 
     def step():
-        opt.zero_grad(set_to_none=False)
+        opt.zero_grad(set_to_none=True)
         loss = F.cross_entropy(model(x).float(), y)
         loss.backward()
         opt.step()
@@ -95,7 +98,7 @@ def build_bert(args, device, dtype):
     labels[mask.to(device)] = -100
 
     def step():
-        opt.zero_grad(set_to_none=False)
+        opt.zero_grad(set_to_none=True)
         loss = model.mlm_loss(ids, labels)
         loss.backward()
         opt.step()
@@ -124,7 +127,7 @@ def build_llama(args, device, dtype, which="llama3-8b"):
     ids = torch.randint(0, cfgm.vocab_size, (batch, seq), generator=g).to(device)
 
     def step():
-        opt.zero_grad(set_to_none=False)
+        opt.zero_grad(set_to_none=True)
         loss = model.lm_loss(ids)
         loss.backward()
         opt.step()
