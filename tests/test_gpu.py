@@ -437,3 +437,46 @@ def test_fused_swiglu_matches_eager(hvd):
     z2.backward(gr)
     assert torch.allclose(a1.grad, a2.grad, rtol=1e-5, atol=1e-5)
     assert torch.allclose(g1.grad, g2.grad, rtol=1e-5, atol=1e-6)
+
+
+def test_syncbn_gpu_ws1(hvd):
+    """SyncBatchNorm on GPU (ws=1: must equal plain BN)."""
+    from axon.sync_batch_norm import SyncBatchNorm
+
+    torch.manual_seed(21)
+    ref = torch.nn.BatchNorm2d(16).to(DEV)
+    sbn = SyncBatchNorm(16).to(DEV)
+    sbn.load_state_dict(ref.state_dict())
+    x = torch.randn(4, 16, 6, 6, device=DEV)
+    x1 = x.clone().requires_grad_(True)
+    x2 = x.clone().requires_grad_(True)
+    y1, y2 = ref(x1), sbn(x2)
+    assert torch.allclose(y1, y2, rtol=1e-4, atol=1e-5)
+    g = torch.randn_like(y1)
+    y1.backward(g)
+    y2.backward(g)
+    assert torch.allclose(x1.grad, x2.grad, rtol=1e-4, atol=1e-5)
+    assert torch.allclose(ref.running_mean, sbn.running_mean,
+                          rtol=1e-4, atol=1e-5)
+
+
+def test_bench_adasum_step_gpu(hvd):
+    """Config-5 sanity: a ResNet step trains under op=Adasum on GPU."""
+    import torch.nn.functional as F
+
+    from axon.models.resnet import resnet50
+    from axon.optim import FusedSGD
+
+    torch.manual_seed(2)
+    model = resnet50().to(DEV, torch.bfloat16)
+    opt = FusedSGD(model.parameters(), lr=0.05, momentum=0.9)
+    opt = hvd.DistributedOptimizer(opt, named_parameters=model.named_parameters(),
+                                   op=hvd.Adasum)
+    x = torch.randn(4, 3, 64, 64, device=DEV, dtype=torch.bfloat16)
+    y = torch.randint(0, 1000, (4,), device=DEV)
+    for _ in range(2):
+        opt.zero_grad()
+        loss = F.cross_entropy(model(x).float(), y)
+        loss.backward()
+        opt.step()
+    assert torch.isfinite(loss.float())
