@@ -81,6 +81,13 @@ class _DistributedOptimizer:
                 process_set=self._process_set)
         self._handles[id(p)] = handle
 
+    def set_backward_passes_per_step(self, passes):
+        """Change gradient-accumulation depth (hvd API): allreduce fires on
+        every `passes`-th backward for each parameter."""
+        self._bpps = passes
+        for p in self._dist_params:
+            self._passes[id(p)] = 0
+
     def synchronize(self):
         """Wait for every outstanding gradient allreduce."""
         for h in self._handles.values():
