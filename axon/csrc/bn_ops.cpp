@@ -144,6 +144,52 @@ std::vector<at::Tensor> rms_bwd(const at::Tensor& x, const at::Tensor& dy_in,
   return {dx, dw};
 }
 
+std::vector<at::Tensor> ln_fwd(const at::Tensor& x, const at::Tensor& w,
+                               const at::Tensor& b, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(),
+              "layernorm input must be contiguous GPU tensor");
+  const int H = static_cast<int>(x.size(-1));
+  TORCH_CHECK(w.is_contiguous() && b.is_contiguous() && w.numel() == H &&
+                  b.numel() == H && w.scalar_type() == x.scalar_type() &&
+                  b.scalar_type() == x.scalar_type(),
+              "layernorm weight/bias mismatch");
+  TORCH_CHECK(H % 2 == 0 && H <= 16384, "layernorm fast path needs even H "
+              "<= 16384 (fallback in Python)");
+  const int64_t rows = x.numel() / H;
+  auto fopt = at::TensorOptions().dtype(at::kFloat).device(x.device());
+  at::Tensor mean = at::empty({rows}, fopt);
+  at::Tensor rstd = at::empty({rows}, fopt);
+  at::Tensor y = at::empty_like(x);
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(x.get_device());
+  launch_ln_fwd(static_cast<int>(x.scalar_type()), x.data_ptr(), w.data_ptr(),
+                b.data_ptr(), y.data_ptr(), mean.data_ptr<float>(),
+                rstd.data_ptr<float>(), rows, H, static_cast<float>(eps),
+                stream.stream());
+  return {y, mean, rstd};
+}
+
+std::vector<at::Tensor> ln_bwd(const at::Tensor& x, const at::Tensor& dy_in,
+                               const at::Tensor& w, const at::Tensor& mean,
+                               const at::Tensor& rstd) {
+  const int H = static_cast<int>(x.size(-1));
+  const int64_t rows = x.numel() / H;
+  at::Tensor dy = dy_in.is_contiguous() ? dy_in : dy_in.contiguous();
+  auto fopt = at::TensorOptions().dtype(at::kFloat).device(x.device());
+  int nblocks = rms_bwd_blocks(rows, nullptr);
+  at::Tensor dw = at::empty({H}, fopt);
+  at::Tensor db = at::empty({H}, fopt);
+  at::Tensor dwpart = at::empty({static_cast<int64_t>(H) * nblocks}, fopt);
+  at::Tensor dbpart = at::empty({static_cast<int64_t>(H) * nblocks}, fopt);
+  at::Tensor dx = at::empty_like(x);
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(x.get_device());
+  launch_ln_bwd(static_cast<int>(x.scalar_type()), x.data_ptr(), dy.data_ptr(),
+                w.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                dx.data_ptr(), dwpart.data_ptr<float>(),
+                dbpart.data_ptr<float>(), dw.data_ptr<float>(),
+                db.data_ptr<float>(), rows, H, stream.stream());
+  return {dx, dw, db};
+}
+
 at::Tensor rope_apply(const at::Tensor& x, const at::Tensor& cos_t,
                       const at::Tensor& sin_t, bool backward) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4,
@@ -191,6 +237,10 @@ std::vector<at::Tensor> swiglu_bwd(const at::Tensor& dz_in, const at::Tensor& a,
 }  // namespace
 
 void register_bn(pybind11::module_& m) {
+  m.def("ln_fwd", &ln_fwd, py::arg("x"), py::arg("w"), py::arg("b"),
+        py::arg("eps"));
+  m.def("ln_bwd", &ln_bwd, py::arg("x"), py::arg("dy"), py::arg("w"),
+        py::arg("mean"), py::arg("rstd"));
   m.def("rope", &rope_apply, py::arg("x"), py::arg("cos"), py::arg("sin"),
         py::arg("backward") = false);
   m.def("swiglu_fwd", &swiglu_fwd, py::arg("a"), py::arg("g"));

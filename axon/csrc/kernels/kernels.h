@@ -80,6 +80,18 @@ void launch_rms_bwd(int dtype, const void* x, const void* dy, const void* w,
                     const float* rstd, void* dx, float* dwpart, float* dw,
                     int64_t rows, int H, hipStream_t stream);
 
+// Fused LayerNorm (layernorm.hip): x/y/dx [rows, H] contiguous; w/b [H]
+// in x's dtype; mean/rstd fp32[rows]; dwpart/dbpart uninitialized fp32
+// [H * rms_bwd_blocks(rows)], dw/db fp32[H] (overwritten). Same bounds as
+// RMSNorm (even H <= 16384; Python falls back otherwise).
+void launch_ln_fwd(int dtype, const void* x, const void* w, const void* b,
+                   void* y, float* mean, float* rstd, int64_t rows, int H,
+                   float eps, hipStream_t stream);
+void launch_ln_bwd(int dtype, const void* x, const void* dy, const void* w,
+                   const float* mean, const float* rstd, void* dx,
+                   float* dwpart, float* dbpart, float* dw, float* db,
+                   int64_t rows, int H, hipStream_t stream);
+
 // Rotary embedding (elementwise.hip): x/y are [B,S,H,D] contiguous (numel
 // elements), cos/sin [S, D/2] in the same dtype; backward = transposed
 // rotation. SwiGLU: z = silu(a)*g, all tensors same-shape contiguous.

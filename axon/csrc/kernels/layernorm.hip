@@ -1,0 +1,274 @@
+// Fused LayerNorm for CDNA4 (the BERT path; same design as rmsnorm.hip):
+// one kernel and one global read of x per direction, rows staged through
+// LDS, fp32 arithmetic, column-major partial buffers for dgamma/dbeta
+// (no atomics).
+//
+//   fwd: y = (x - mean) * rstd * w + b ; saves per-row mean/rstd (fp32)
+//   bwd: g = dy*w ; dx = rstd*(g - mean_r(g) - xhat*mean_r(g*xhat))
+//        dw = sum_rows(dy * xhat) ; db = sum_rows(dy)
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+#include <hip/hip_runtime.h>
+
+#include <algorithm>
+
+#include "kernels.h"
+
+namespace axon {
+
+int rms_bwd_blocks(int64_t rows, int* rows_per_block);  // rmsnorm.hip
+
+namespace {
+
+constexpr int kThreads = 256;
+constexpr int kWaves = 4;
+constexpr int kHalfId = 5, kFloatId = 6, kBF16Id = 15;
+constexpr int kMaxLnChunks = 4;  // dgamma/dbeta register arrays: H <= 8192@VC8
+
+template <typename T>
+__device__ __forceinline__ float lf_of(T x);
+template <>
+__device__ __forceinline__ float lf_of(float x) { return x; }
+template <>
+__device__ __forceinline__ float lf_of(__half x) { return __half2float(x); }
+template <>
+__device__ __forceinline__ float lf_of(__hip_bfloat16 x) { return __bfloat162float(x); }
+
+template <typename T>
+__device__ __forceinline__ T lt_of(float x);
+template <>
+__device__ __forceinline__ float lt_of(float x) { return x; }
+template <>
+__device__ __forceinline__ __half lt_of(float x) { return __float2half(x); }
+template <>
+__device__ __forceinline__ __hip_bfloat16 lt_of(float x) { return __float2bfloat16(x); }
+
+template <typename T, int V>
+struct alignas(sizeof(T) * V) LVec {
+  T v[V];
+};
+
+__device__ float ln_block_sum(float v, float* red) {
+  for (int off = 32; off > 0; off >>= 1)
+    v += __shfl_xor(v, off, 64);
+  const int lane = threadIdx.x % 64, wave = threadIdx.x / 64;
+  if (lane == 0) red[wave] = v;
+  __syncthreads();
+  float total = red[0];
+#pragma unroll
+  for (int w = 1; w < kWaves; ++w) total += red[w];
+  return total;
+}
+
+// ------------------------------------------------------------------ forward
+
+template <typename T, int VC>
+__global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                              const T* __restrict__ b, T* __restrict__ y,
+                              float* __restrict__ mean_out,
+                              float* __restrict__ rstd_out, int64_t rows, int H,
+                              float eps) {
+  extern __shared__ unsigned char smem[];
+  T* row = reinterpret_cast<T*>(smem);
+  __shared__ float red[kWaves];
+  const int hv = H / VC;
+
+  for (int64_t r = blockIdx.x; r < rows; r += gridDim.x) {
+    const LVec<T, VC>* xr = reinterpret_cast<const LVec<T, VC>*>(x + r * H);
+    float s = 0.f, q = 0.f;
+    for (int i = threadIdx.x; i < hv; i += kThreads) {
+      LVec<T, VC> xv = xr[i];
+      reinterpret_cast<LVec<T, VC>*>(row)[i] = xv;
+#pragma unroll
+      for (int k = 0; k < VC; ++k) {
+        float f = lf_of(xv.v[k]);
+        s += f;
+        q += f * f;
+      }
+    }
+    __syncthreads();
+    float tot_s = ln_block_sum(s, red);
+    __syncthreads();  // red[] reuse between the two reductions
+    float tot_q = ln_block_sum(q, red);
+    float mean = tot_s / H;
+    float var = fmaxf(tot_q / H - mean * mean, 0.f);
+    float rstd = rsqrtf(var + eps);
+    if (threadIdx.x == 0) {
+      if (mean_out) mean_out[r] = mean;
+      if (rstd_out) rstd_out[r] = rstd;
+    }
+
+    LVec<T, VC>* yr = reinterpret_cast<LVec<T, VC>*>(y + r * H);
+    const LVec<T, VC>* wr = reinterpret_cast<const LVec<T, VC>*>(w);
+    const LVec<T, VC>* br = reinterpret_cast<const LVec<T, VC>*>(b);
+    for (int i = threadIdx.x; i < hv; i += kThreads) {
+      LVec<T, VC> xv = reinterpret_cast<const LVec<T, VC>*>(row)[i];
+      LVec<T, VC> wv = wr[i], bv = br[i];
+      LVec<T, VC> yv;
+#pragma unroll
+      for (int k = 0; k < VC; ++k)
+        yv.v[k] = lt_of<T>(fmaf((lf_of(xv.v[k]) - mean) * rstd, lf_of(wv.v[k]),
+                                lf_of(bv.v[k])));
+      yr[i] = yv;
+    }
+    __syncthreads();
+  }
+}
+
+// ----------------------------------------------------------------- backward
+
+template <typename T, int VC>
+__global__ void ln_bwd_kernel(const T* __restrict__ x, const T* __restrict__ dy,
+                              const T* __restrict__ w,
+                              const float* __restrict__ mean_in,
+                              const float* __restrict__ rstd_in,
+                              T* __restrict__ dx, float* __restrict__ dwpart,
+                              float* __restrict__ dbpart, int64_t rows, int H,
+                              int rows_per_block) {
+  extern __shared__ unsigned char smem[];
+  T* xrow = reinterpret_cast<T*>(smem);
+  T* drow = reinterpret_cast<T*>(smem + static_cast<size_t>(H) * sizeof(T));
+  __shared__ float red[kWaves];
+  const int hv = H / VC;
+  const int chunks = (hv + kThreads - 1) / kThreads;
+
+  float dwacc[kMaxLnChunks][VC], dbacc[kMaxLnChunks][VC];
+  for (int j = 0; j < chunks; ++j)
+#pragma unroll
+    for (int k = 0; k < VC; ++k) dwacc[j][k] = dbacc[j][k] = 0.f;
+
+  const int64_t r0 = static_cast<int64_t>(blockIdx.x) * rows_per_block;
+  const int64_t r1 = r0 + rows_per_block > rows ? rows : r0 + rows_per_block;
+  const LVec<T, VC>* wr = reinterpret_cast<const LVec<T, VC>*>(w);
+
+  for (int64_t r = r0; r < r1; ++r) {
+    const LVec<T, VC>* xr = reinterpret_cast<const LVec<T, VC>*>(x + r * H);
+    const LVec<T, VC>* dr = reinterpret_cast<const LVec<T, VC>*>(dy + r * H);
+    float mean = mean_in[r], rstd = rstd_in[r];
+    float s1 = 0.f, s2 = 0.f;
+    for (int i = threadIdx.x, j = 0; i < hv; i += kThreads, ++j) {
+      LVec<T, VC> xv = xr[i], dv = dr[i], wv = wr[i];
+      reinterpret_cast<LVec<T, VC>*>(xrow)[i] = xv;
+      reinterpret_cast<LVec<T, VC>*>(drow)[i] = dv;
+#pragma unroll
+      for (int k = 0; k < VC; ++k) {
+        float xhat = (lf_of(xv.v[k]) - mean) * rstd;
+        float df = lf_of(dv.v[k]);
+        float g = df * lf_of(wv.v[k]);
+        s1 += g;
+        s2 += g * xhat;
+        dwacc[j][k] += df * xhat;
+        dbacc[j][k] += df;
+      }
+    }
+    __syncthreads();
+    float m1 = ln_block_sum(s1, red) / H;
+    __syncthreads();
+    float m2 = ln_block_sum(s2, red) / H;
+
+    LVec<T, VC>* oxr = reinterpret_cast<LVec<T, VC>*>(dx + r * H);
+    for (int i = threadIdx.x; i < hv; i += kThreads) {
+      LVec<T, VC> xv = reinterpret_cast<const LVec<T, VC>*>(xrow)[i];
+      LVec<T, VC> dv = reinterpret_cast<const LVec<T, VC>*>(drow)[i];
+      LVec<T, VC> wv = wr[i];
+      LVec<T, VC> ov;
+#pragma unroll
+      for (int k = 0; k < VC; ++k) {
+        float xhat = (lf_of(xv.v[k]) - mean) * rstd;
+        float g = lf_of(dv.v[k]) * lf_of(wv.v[k]);
+        ov.v[k] = lt_of<T>(rstd * (g - m1 - xhat * m2));
+      }
+      oxr[i] = ov;
+    }
+    __syncthreads();
+  }
+
+  const int nblocks = gridDim.x;
+  for (int i = threadIdx.x, j = 0; i < hv; i += kThreads, ++j)
+#pragma unroll
+    for (int k = 0; k < VC; ++k) {
+      dwpart[static_cast<int64_t>(i * VC + k) * nblocks + blockIdx.x] = dwacc[j][k];
+      dbpart[static_cast<int64_t>(i * VC + k) * nblocks + blockIdx.x] = dbacc[j][k];
+    }
+}
+
+__global__ void ln_fold_kernel(const float* __restrict__ part,
+                               float* __restrict__ out, int nblocks) {
+  const float* col = part + static_cast<int64_t>(blockIdx.x) * nblocks;
+  float acc = 0.f;
+  for (int b = threadIdx.x; b < nblocks; b += static_cast<int>(blockDim.x))
+    acc += col[b];
+  for (int off = 32; off > 0; off >>= 1)
+    acc += __shfl_xor(acc, off, 64);
+  __shared__ float red[kWaves];
+  const int lane = threadIdx.x % 64, wave = threadIdx.x / 64;
+  if (lane == 0) red[wave] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) out[blockIdx.x] = red[0] + red[1] + red[2] + red[3];
+}
+
+int ln_vc(int H) {
+  if (H % 8 == 0) return 8;
+  if (H % 4 == 0) return 4;
+  if (H % 2 == 0) return 2;
+  return 1;
+}
+
+template <typename T>
+void ln_fwd_t(const void* x, const void* w, const void* b, void* y,
+              float* mean, float* rstd, int64_t rows, int H, float eps,
+              hipStream_t stream) {
+  int vc = ln_vc(H);
+  size_t lds = static_cast<size_t>(H) * sizeof(T);
+  int blocks = static_cast<int>(std::min<int64_t>(rows, 16384));
+  switch (vc) {
+    case 8: hipLaunchKernelGGL((ln_fwd_kernel<T, 8>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(w), static_cast<const T*>(b), static_cast<T*>(y), mean, rstd, rows, H, eps); break;
+    case 4: hipLaunchKernelGGL((ln_fwd_kernel<T, 4>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(w), static_cast<const T*>(b), static_cast<T*>(y), mean, rstd, rows, H, eps); break;
+    case 2: hipLaunchKernelGGL((ln_fwd_kernel<T, 2>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(w), static_cast<const T*>(b), static_cast<T*>(y), mean, rstd, rows, H, eps); break;
+    default: hipLaunchKernelGGL((ln_fwd_kernel<T, 1>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(w), static_cast<const T*>(b), static_cast<T*>(y), mean, rstd, rows, H, eps); break;
+  }
+}
+
+template <typename T>
+void ln_bwd_t(const void* x, const void* dy, const void* w, const float* mean,
+              const float* rstd, void* dx, float* dwpart, float* dbpart,
+              float* dw, float* db, int64_t rows, int H, hipStream_t stream) {
+  int vc = ln_vc(H);
+  while ((H / vc + kThreads - 1) / kThreads > kMaxLnChunks && vc < 8) vc *= 2;
+  size_t lds = 2 * static_cast<size_t>(H) * sizeof(T);
+  int rpb = 0;
+  int blocks = rms_bwd_blocks(rows, &rpb);
+  switch (vc) {
+    case 8: hipLaunchKernelGGL((ln_bwd_kernel<T, 8>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), mean, rstd, static_cast<T*>(dx), dwpart, dbpart, rows, H, rpb); break;
+    case 4: hipLaunchKernelGGL((ln_bwd_kernel<T, 4>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), mean, rstd, static_cast<T*>(dx), dwpart, dbpart, rows, H, rpb); break;
+    case 2: hipLaunchKernelGGL((ln_bwd_kernel<T, 2>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), mean, rstd, static_cast<T*>(dx), dwpart, dbpart, rows, H, rpb); break;
+    default: hipLaunchKernelGGL((ln_bwd_kernel<T, 1>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), mean, rstd, static_cast<T*>(dx), dwpart, dbpart, rows, H, rpb); break;
+  }
+  hipLaunchKernelGGL(ln_fold_kernel, H, kThreads, 0, stream, dwpart, dw, blocks);
+  hipLaunchKernelGGL(ln_fold_kernel, H, kThreads, 0, stream, dbpart, db, blocks);
+}
+
+}  // namespace
+
+void launch_ln_fwd(int dtype, const void* x, const void* w, const void* b,
+                   void* y, float* mean, float* rstd, int64_t rows, int H,
+                   float eps, hipStream_t stream) {
+  switch (dtype) {
+    case kFloatId: ln_fwd_t<float>(x, w, b, y, mean, rstd, rows, H, eps, stream); break;
+    case kHalfId: ln_fwd_t<__half>(x, w, b, y, mean, rstd, rows, H, eps, stream); break;
+    case kBF16Id: ln_fwd_t<__hip_bfloat16>(x, w, b, y, mean, rstd, rows, H, eps, stream); break;
+  }
+}
+
+void launch_ln_bwd(int dtype, const void* x, const void* dy, const void* w,
+                   const float* mean, const float* rstd, void* dx,
+                   float* dwpart, float* dbpart, float* dw, float* db,
+                   int64_t rows, int H, hipStream_t stream) {
+  switch (dtype) {
+    case kFloatId: ln_bwd_t<float>(x, dy, w, mean, rstd, dx, dwpart, dbpart, dw, db, rows, H, stream); break;
+    case kHalfId: ln_bwd_t<__half>(x, dy, w, mean, rstd, dx, dwpart, dbpart, dw, db, rows, H, stream); break;
+    case kBF16Id: ln_bwd_t<__hip_bfloat16>(x, dy, w, mean, rstd, dx, dwpart, dbpart, dw, db, rows, H, stream); break;
+  }
+}
+
+}  // namespace axon

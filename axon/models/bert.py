@@ -9,6 +9,8 @@ from dataclasses import dataclass
 
 import torch
 import torch.nn as nn
+
+from axon.ops.layernorm import FusedLayerNorm
 import torch.nn.functional as F
 
 
@@ -60,10 +62,10 @@ class BertLayer(nn.Module):
     def __init__(self, cfg):
         super().__init__()
         self.attn = BertSelfAttention(cfg)
-        self.ln1 = nn.LayerNorm(cfg.hidden)
+        self.ln1 = FusedLayerNorm(cfg.hidden)
         self.fc1 = nn.Linear(cfg.hidden, cfg.ffn)
         self.fc2 = nn.Linear(cfg.ffn, cfg.hidden)
-        self.ln2 = nn.LayerNorm(cfg.hidden)
+        self.ln2 = FusedLayerNorm(cfg.hidden)
         self.drop = nn.Dropout(cfg.dropout)
 
     def forward(self, x, attn_mask=None):
@@ -81,10 +83,10 @@ class BertForPreTraining(nn.Module):
         self.tok = nn.Embedding(cfg.vocab_size, cfg.hidden)
         self.pos = nn.Embedding(cfg.max_seq, cfg.hidden)
         self.typ = nn.Embedding(cfg.type_vocab, cfg.hidden)
-        self.ln_emb = nn.LayerNorm(cfg.hidden)
+        self.ln_emb = FusedLayerNorm(cfg.hidden)
         self.layers = nn.ModuleList(BertLayer(cfg) for _ in range(cfg.layers))
         self.mlm_dense = nn.Linear(cfg.hidden, cfg.hidden)
-        self.mlm_ln = nn.LayerNorm(cfg.hidden)
+        self.mlm_ln = FusedLayerNorm(cfg.hidden)
         self.mlm_bias = nn.Parameter(torch.zeros(cfg.vocab_size))
         self.apply(self._init)
 

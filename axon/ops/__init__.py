@@ -1,2 +1,5 @@
 """Hand-written CDNA4 op modules (HIP kernels with torch autograd)."""
 from axon.ops.batchnorm import FusedBatchNorm2d  # noqa: F401
+from axon.ops.layernorm import FusedLayerNorm  # noqa: F401
+from axon.ops.rmsnorm import FusedRMSNorm  # noqa: F401
+from axon.ops.rotary import rope, swiglu  # noqa: F401

@@ -1,0 +1,45 @@
+"""Fused LayerNorm module (csrc/kernels/layernorm.hip): one kernel and one
+global pass over x per direction vs ATen's multi-kernel path. Drop-in for
+nn.LayerNorm over the last dimension; falls back off the fast path."""
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+_SUPPORTED = (torch.float32, torch.float16, torch.bfloat16)
+
+
+class _LayerNormFunc(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        from axon import _core
+
+        y, mean, rstd = _core.ln_fwd(x, weight, bias, eps)
+        ctx.save_for_backward(x, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        from axon import _core
+
+        x, weight, mean, rstd = ctx.saved_tensors
+        dx, dw, db = _core.ln_bwd(x, dy, weight, mean, rstd)
+        return dx, dw.to(weight.dtype), db.to(weight.dtype), None
+
+
+class FusedLayerNorm(nn.LayerNorm):
+    """nn.LayerNorm with the CDNA4 kernels (single normalized dim only)."""
+
+    def _fast_path(self, x):
+        return (x.is_cuda and x.dtype in _SUPPORTED and x.is_contiguous() and
+                len(self.normalized_shape) == 1 and
+                x.size(-1) == self.normalized_shape[0] and
+                self.weight is not None and self.bias is not None and
+                x.dtype == self.weight.dtype and
+                self.normalized_shape[0] % 2 == 0 and
+                self.normalized_shape[0] <= 16384)
+
+    def forward(self, x):
+        if self._fast_path(x):
+            return _LayerNormFunc.apply(x, self.weight, self.bias, self.eps)
+        return F.layer_norm(x, self.normalized_shape, self.weight, self.bias,
+                            self.eps)

@@ -480,3 +480,56 @@ def test_bench_adasum_step_gpu(hvd):
         loss.backward()
         opt.step()
     assert torch.isfinite(loss.float())
+
+
+@pytest.mark.parametrize("dtype,tol", [(torch.float32, 1e-5),
+                                       (torch.bfloat16, 2e-2)])
+@pytest.mark.parametrize("H", [256, 1024])
+def test_fused_layernorm_matches_torch(hvd, dtype, tol, H):
+    from axon.ops import FusedLayerNorm
+
+    torch.manual_seed(H)
+    ref = torch.nn.LayerNorm(H).to(DEV, dtype)
+    fused = FusedLayerNorm(H).to(DEV, dtype)
+    fused.load_state_dict(ref.state_dict())
+    with torch.no_grad():
+        for m in (ref, fused):
+            m.weight.copy_(torch.randn(H).to(dtype) * 0.1 + 1)
+            m.bias.copy_(torch.randn(H).to(dtype) * 0.1)
+    fused.load_state_dict(ref.state_dict())
+    x = torch.randn(57, H, device=DEV, dtype=dtype) * 2 + 1
+    x1 = x.clone().requires_grad_(True)
+    x2 = x.clone().requires_grad_(True)
+    y1, y2 = ref(x1), fused(x2)
+    assert torch.allclose(y1, y2, rtol=tol, atol=tol), (y1 - y2).abs().max()
+    g = torch.randn_like(y1)
+    y1.backward(g)
+    y2.backward(g)
+    assert torch.allclose(x1.grad, x2.grad, rtol=tol, atol=tol * 4), \
+        (x1.grad - x2.grad).abs().max()
+    assert torch.allclose(ref.weight.grad, fused.weight.grad,
+                          rtol=tol * 4, atol=tol * 40)
+    assert torch.allclose(ref.bias.grad, fused.bias.grad,
+                          rtol=tol * 4, atol=tol * 40)
+
+
+def test_bert_tiny_step_gpu(hvd):
+    from axon.models.bert import BertForPreTraining, bert_large_config
+    from axon.optim import FusedLAMB
+
+    cfg = bert_large_config()
+    cfg.layers, cfg.hidden, cfg.heads, cfg.ffn = 2, 128, 4, 256
+    torch.manual_seed(0)
+    model = BertForPreTraining(cfg).to(DEV, torch.bfloat16)
+    opt = FusedLAMB(model.parameters(), lr=1e-3)
+    ids = torch.randint(0, 1000, (4, 32), device=DEV)
+    labels = ids.clone()
+    losses = []
+    for _ in range(3):
+        opt.zero_grad()
+        loss = model.mlm_loss(ids, labels)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert losses[-1] < losses[0]
