@@ -5,6 +5,11 @@ Run: python -m axon.runner.run -np 8 python examples/train_llama.py
 """
 import argparse
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import torch
 
 import axon as hvd

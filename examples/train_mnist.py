@@ -3,6 +3,11 @@ Run: python -m axon.runner.run -np 2 --no-pin python examples/train_mnist.py
 (Synthetic MNIST-shaped data: this environment has no dataset downloads;
 swap _synthetic_loader for torchvision.datasets.MNIST when available.)
 """
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import torch
 import torch.nn.functional as F
 

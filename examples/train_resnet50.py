@@ -6,6 +6,11 @@ Adasum variant (config 5): add --adasum.
 import argparse
 import time
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import torch
 import torch.nn.functional as F
 
