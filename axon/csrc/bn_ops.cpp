@@ -36,6 +36,7 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
   }
   const int C = static_cast<int>(x.size(1));
   const int64_t rows = x.numel() / C;
+  TORCH_CHECK(rows > 0 && C > 0, "bn: empty input");
   auto fopt = at::TensorOptions().dtype(at::kFloat).device(x.device());
   at::Tensor save_mean = at::empty({C}, fopt);
   at::Tensor save_rstd = at::empty({C}, fopt);
@@ -116,6 +117,7 @@ std::vector<at::Tensor> rms_fwd(const at::Tensor& x, const at::Tensor& w,
   TORCH_CHECK(H % 2 == 0 && H <= 16384, "rmsnorm fast path needs even H <= "
               "16384 (fallback in Python)");
   const int64_t rows = x.numel() / H;
+  TORCH_CHECK(rows > 0, "rmsnorm: empty input");
   auto fopt = at::TensorOptions().dtype(at::kFloat).device(x.device());
   at::Tensor rstd = at::empty({rows}, fopt);
   at::Tensor y = at::empty_like(x);
@@ -156,6 +158,7 @@ std::vector<at::Tensor> ln_fwd(const at::Tensor& x, const at::Tensor& w,
   TORCH_CHECK(H % 2 == 0 && H <= 16384, "layernorm fast path needs even H "
               "<= 16384 (fallback in Python)");
   const int64_t rows = x.numel() / H;
+  TORCH_CHECK(rows > 0, "layernorm: empty input");
   auto fopt = at::TensorOptions().dtype(at::kFloat).device(x.device());
   at::Tensor mean = at::empty({rows}, fopt);
   at::Tensor rstd = at::empty({rows}, fopt);
@@ -204,6 +207,7 @@ at::Tensor rope_apply(const at::Tensor& x, const at::Tensor& cos_t,
                   sin_t.scalar_type() == x.scalar_type(),
               "rope cos/sin must be [S, D/2] in the input dtype");
   at::Tensor y = at::empty_like(x);
+  if (x.numel() == 0) return y;
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(x.get_device());
   launch_rope(static_cast<int>(x.scalar_type()), x.data_ptr(), y.data_ptr(),
               cos_t.data_ptr(), sin_t.data_ptr(), x.numel(), D / 2, H, S,
@@ -216,6 +220,7 @@ at::Tensor swiglu_fwd(const at::Tensor& a, const at::Tensor& g) {
                   a.sizes() == g.sizes() && a.scalar_type() == g.scalar_type(),
               "swiglu inputs must be matching contiguous GPU tensors");
   at::Tensor z = at::empty_like(a);
+  if (a.numel() == 0) return z;
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(a.get_device());
   launch_swiglu_fwd(static_cast<int>(a.scalar_type()), a.data_ptr(),
                     g.data_ptr(), z.data_ptr(), a.numel(), stream.stream());
