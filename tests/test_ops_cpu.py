@@ -435,3 +435,39 @@ def _steady_state_cache(rank, size):
 
 def test_steady_state_response_cache():
     run_distributed(_steady_state_cache, 4)
+
+
+def _ws8_smoke(rank, size):
+    import axon as hvd
+
+    t = torch.full((33,), float(rank + 1))
+    out = hvd.allreduce(t, op=hvd.Sum, name="ws8")
+    assert torch.allclose(out, torch.full((33,), float(sum(range(1, size + 1)))))
+    ps = hvd.add_process_set([0, 3, 5, 7])
+    if ps.included():
+        o = hvd.allreduce(t, average=True, process_set=ps, name="ws8ps")
+        expect = sum(r + 1 for r in ps.ranks) / ps.size()
+        assert torch.allclose(o, torch.full((33,), expect))
+    hvd.barrier(name="ws8.done")
+
+
+def test_world_size_8():
+    """Full-width bitmask paths in the controller (size 8 = one rank per
+    MI355X on the target node)."""
+    run_distributed(_ws8_smoke, 8)
+
+
+def _bf16_wire(rank, size):
+    import axon as hvd
+
+    t = torch.arange(64, dtype=torch.float32) * (rank + 1)
+    out = hvd.allreduce(t, op=hvd.Sum, compression=hvd.Compression.bf16,
+                        name="bf16wire")
+    expect = torch.arange(64, dtype=torch.float32).bfloat16().float() * \
+        sum(range(1, size + 1))
+    assert out.dtype == torch.float32
+    assert torch.allclose(out, expect, rtol=2e-2, atol=2e-2)
+
+
+def test_bf16_wire_compression():
+    run_distributed(_bf16_wire, 2)
