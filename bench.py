@@ -138,7 +138,7 @@ def build_llama(args, device, dtype, which="llama3-8b"):
     torch.manual_seed(1234)
     cfgm = llama3_8b_config() if which == "llama3-8b" else llama3_1b_config()
     model = LlamaForCausalLM(cfgm).to(device=device, dtype=dtype)
-    batch = args.batch or (4 if which == "llama3-8b" else 8)
+    batch = args.batch or (8 if which == "llama3-8b" else 8)
     seq = args.seq_len or 4096
     if args.no_fused_optimizer or device.type == "cpu":
         opt = torch.optim.AdamW(model.parameters(), lr=3e-4)
