@@ -137,9 +137,12 @@ def build_llama(args, device, dtype, which="llama3-8b"):
         _load_tunableop("llama")
     fa = os.environ.get("AXON_FA_LIB")
     if fa and torch.cuda.is_available():
-        # "ck" selects composable_kernel flash attention, "aotriton" the
-        # default Triton-AOT kernels (probe both; see profiles/README.md)
-        torch.backends.cuda.preferred_rocm_fa_library(fa)
+        # "ck" selects composable_kernel flash attention when torch is
+        # built with it (this image's torch is not — AOTriton only).
+        try:
+            torch.backends.cuda.preferred_rocm_fa_library(fa)
+        except RuntimeError as e:
+            print(f"[bench] AXON_FA_LIB={fa} unavailable: {e}", flush=True)
     torch.manual_seed(1234)
     cfgm = llama3_8b_config() if which == "llama3-8b" else llama3_1b_config()
     model = LlamaForCausalLM(cfgm).to(device=device, dtype=dtype)

@@ -11,8 +11,8 @@ import tempfile
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
-def _run(np_, script, extra_args=(), timeout=120):
-    env = dict(os.environ)
+def _run(np_, script, extra_args=(), timeout=120, env=None):
+    env = {**os.environ, **(env or {})}
     env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
     env["AXON_SHUTDOWN_TIMEOUT"] = "3"  # keep fail-fast tests quick
     return subprocess.run(
@@ -85,3 +85,57 @@ def test_timeline_written():
         # per-tensor lanes got labeled
         lanes = [e for e in events if e.get("name") == "thread_name"]
         assert any("allreduce.t0" in json.dumps(e) for e in lanes)
+
+
+def test_duplicate_inflight_name_rejected():
+    """Two async ops with the same name in flight must error loudly, not
+    corrupt the pending table."""
+    script = (
+        "import torch, axon as hvd\n"
+        "hvd.init()\n"
+        "h1 = hvd.allreduce_async_(torch.ones(4), name='dup')\n"
+        "try:\n"
+        "    h2 = hvd.allreduce_async_(torch.ones(4), name='dup')\n"
+        "    hvd.synchronize(h2)\n"
+        "    raise SystemExit('second submit should have failed')\n"
+        "except RuntimeError as e:\n"
+        "    assert 'unique name' in str(e) or 'in flight' in str(e), e\n"
+        "hvd.synchronize(h1)\n"
+        "hvd.shutdown()\n")
+    r = _run(2, script)
+    assert r.returncode == 0, r.stderr
+
+
+def test_shape_mismatch_reported():
+    """Mismatched allreduce shapes across ranks -> clean error on every
+    rank (controller validation), not a hang or corruption."""
+    script = (
+        "import os, torch, axon as hvd\n"
+        "hvd.init()\n"
+        "n = 4 if hvd.rank() == 0 else 5\n"
+        "try:\n"
+        "    hvd.allreduce(torch.ones(n), name='mm')\n"
+        "    raise SystemExit('mismatch not detected')\n"
+        "except RuntimeError as e:\n"
+        "    assert 'mismatch' in str(e), e\n"
+        "hvd.shutdown()\n")
+    r = _run(2, script)
+    assert r.returncode == 0, r.stderr
+
+
+def test_stall_inspector_warns():
+    """One rank submits, the other doesn't: the stall inspector must warn
+    within AXON_STALL_CHECK_TIME."""
+    script = (
+        "import os, time, torch, axon as hvd\n"
+        "hvd.init()\n"
+        "if hvd.rank() == 0:\n"
+        "    h = hvd.allreduce_async_(torch.ones(3), name='stalled')\n"
+        "time.sleep(2.5)\n"
+        "if hvd.rank() != 0:\n"
+        "    h = hvd.allreduce_async_(torch.ones(3), name='stalled')\n"
+        "hvd.synchronize(h)\n"
+        "hvd.shutdown()\n")
+    r = _run(2, script, env={"AXON_STALL_CHECK_TIME": "1"})
+    assert r.returncode == 0, r.stderr
+    assert "stall inspector" in r.stderr
