@@ -1,0 +1,71 @@
+"""Controller property tests (SURVEY.md §7 hard part #1): the identical-
+collective-order invariant must hold under randomized submission orders,
+interleavings, and op mixes. Every rank submits the same logical set of
+ops but in a different (seeded-random) order with random delays; the
+engine must complete all of them with correct results and no deadlock."""
+import random
+
+import torch
+
+from conftest import run_distributed
+
+
+def _random_order_ops(rank, size, seed):
+    import time
+
+    import axon as hvd
+
+    ops = []
+    rng = random.Random(seed)  # same op set on every rank
+    for i in range(12):
+        kind = rng.choice(["allreduce", "broadcast", "allgather"])
+        numel = rng.randint(1, 2000)
+        ops.append((i, kind, numel))
+
+    order = list(range(len(ops)))
+    # DIFFERENT order per rank: the controller must still agree globally.
+    random.Random(seed * 131 + rank).shuffle(order)
+
+    jitter = random.Random(seed * 17 + rank)
+    handles = {}
+    for idx in order:
+        i, kind, numel = ops[idx]
+        if jitter.random() < 0.4:
+            time.sleep(jitter.random() * 0.004)
+        t = torch.full((numel,), float(rank + 1))
+        if kind == "allreduce":
+            handles[i] = ("allreduce",
+                          hvd.allreduce_async(t, op=hvd.Sum, name=f"p{i}"), numel)
+        elif kind == "broadcast":
+            handles[i] = ("broadcast",
+                          hvd.broadcast_async(t, root_rank=i % size,
+                                              name=f"p{i}"), numel)
+        else:
+            handles[i] = ("allgather",
+                          hvd.allgather_async(t, name=f"p{i}"), numel)
+
+    for i, (kind, h, numel) in sorted(handles.items()):
+        out = hvd.synchronize(h)
+        if kind == "allreduce":
+            expect = float(sum(range(1, size + 1)))
+            assert torch.allclose(out, torch.full((numel,), expect)), (i, kind)
+        elif kind == "broadcast":
+            assert torch.allclose(out, torch.full((numel,), float(i % size + 1))), \
+                (i, kind)
+        else:
+            assert out.shape[0] == numel * size
+            for r in range(size):
+                assert torch.allclose(out[r * numel:(r + 1) * numel],
+                                      torch.full((numel,), float(r + 1))), (i, r)
+    hvd.barrier(name="prop.done")
+
+
+def test_randomized_submission_orders():
+    """5 seeds x ws=3: shuffled per-rank submission with jitter."""
+    for seed in range(5):
+        run_distributed(_random_order_ops, 3, args=(seed,))
+
+
+def test_randomized_submission_orders_ws4():
+    for seed in (7, 11):
+        run_distributed(_random_order_ops, 4, args=(seed,))
