@@ -114,8 +114,10 @@ std::vector<at::Tensor> rms_fwd(const at::Tensor& x, const at::Tensor& w,
   TORCH_CHECK(w.is_contiguous() && w.numel() == H &&
                   w.scalar_type() == x.scalar_type(),
               "rmsnorm weight mismatch");
-  TORCH_CHECK(H % 2 == 0 && H <= 16384, "rmsnorm fast path needs even H <= "
-              "16384 (fallback in Python)");
+  int vc_ = 0, ch_ = 0;
+  TORCH_CHECK(rms_bwd_pick(H, &vc_, &ch_),
+              "rmsnorm fast path supports H in {64,128,256,512,1024,2048,"
+              "4096} (fallback in Python)");
   const int64_t rows = x.numel() / H;
   TORCH_CHECK(rows > 0, "rmsnorm: empty input");
   auto fopt = at::TensorOptions().dtype(at::kFloat).device(x.device());
@@ -155,8 +157,10 @@ std::vector<at::Tensor> ln_fwd(const at::Tensor& x, const at::Tensor& w,
                   b.numel() == H && w.scalar_type() == x.scalar_type() &&
                   b.scalar_type() == x.scalar_type(),
               "layernorm weight/bias mismatch");
-  TORCH_CHECK(H % 2 == 0 && H <= 16384, "layernorm fast path needs even H "
-              "<= 16384 (fallback in Python)");
+  int vc_ = 0, ch_ = 0;
+  TORCH_CHECK(rms_bwd_pick(H, &vc_, &ch_),
+              "layernorm fast path supports H in {64,128,256,512,1024,2048,"
+              "4096} (fallback in Python)");
   const int64_t rows = x.numel() / H;
   TORCH_CHECK(rows > 0, "layernorm: empty input");
   auto fopt = at::TensorOptions().dtype(at::kFloat).device(x.device());

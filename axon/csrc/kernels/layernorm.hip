@@ -11,12 +11,14 @@
 #include <hip/hip_runtime.h>
 
 #include <algorithm>
+#include <type_traits>
 
 #include "kernels.h"
 
 namespace axon {
 
-int rms_bwd_blocks(int64_t rows, int* rows_per_block);  // rmsnorm.hip
+int rms_bwd_blocks(int64_t rows, int* rows_per_unit);  // rmsnorm.hip
+bool rms_bwd_pick(int H, int* vc, int* ch);             // rmsnorm.hip
 
 namespace {
 
@@ -116,79 +118,88 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
 }
 
 // ----------------------------------------------------------------- backward
+// One WAVE per row, registers only (see rmsnorm.hip backward: the
+// block-per-row/LDS version measured ~7x off streaming SOL at H=1024).
+// Two row sums (s1 = sum g, s2 = sum g*xhat) reduce with xor-shuffles.
 
-template <typename T, int VC>
+__device__ __forceinline__ float ln_wave_sum(float v) {
+  for (int off = 32; off > 0; off >>= 1)
+    v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+template <typename T, int VC, int CH>
 __global__ void ln_bwd_kernel(const T* __restrict__ x, const T* __restrict__ dy,
                               const T* __restrict__ w,
                               const float* __restrict__ mean_in,
                               const float* __restrict__ rstd_in,
                               T* __restrict__ dx, float* __restrict__ dwpart,
                               float* __restrict__ dbpart, int64_t rows, int H,
-                              int rows_per_block) {
-  extern __shared__ unsigned char smem[];
-  T* xrow = reinterpret_cast<T*>(smem);
-  T* drow = reinterpret_cast<T*>(smem + static_cast<size_t>(H) * sizeof(T));
-  __shared__ float red[kWaves];
-  const int hv = H / VC;
-  const int chunks = (hv + kThreads - 1) / kThreads;
+                              int rows_per_unit) {
+  const int lane = threadIdx.x % 64;
+  const int wave = threadIdx.x / 64;
+  const int unit = blockIdx.x * kWaves + wave;
+  const int nunits = gridDim.x * kWaves;
 
-  float dwacc[kMaxLnChunks][VC], dbacc[kMaxLnChunks][VC];
-  for (int j = 0; j < chunks; ++j)
+  LVec<T, VC> wv[CH];
+#pragma unroll
+  for (int j = 0; j < CH; ++j)
+    wv[j] = reinterpret_cast<const LVec<T, VC>*>(w)[lane + 64 * j];
+
+  float dwacc[CH][VC], dbacc[CH][VC];
+#pragma unroll
+  for (int j = 0; j < CH; ++j)
 #pragma unroll
     for (int k = 0; k < VC; ++k) dwacc[j][k] = dbacc[j][k] = 0.f;
 
-  const int64_t r0 = static_cast<int64_t>(blockIdx.x) * rows_per_block;
-  const int64_t r1 = r0 + rows_per_block > rows ? rows : r0 + rows_per_block;
-  const LVec<T, VC>* wr = reinterpret_cast<const LVec<T, VC>*>(w);
+  const int64_t r0 = static_cast<int64_t>(unit) * rows_per_unit;
+  const int64_t r1 = r0 + rows_per_unit > rows ? rows : r0 + rows_per_unit;
 
   for (int64_t r = r0; r < r1; ++r) {
     const LVec<T, VC>* xr = reinterpret_cast<const LVec<T, VC>*>(x + r * H);
     const LVec<T, VC>* dr = reinterpret_cast<const LVec<T, VC>*>(dy + r * H);
-    float mean = mean_in[r], rstd = rstd_in[r];
+    const float mean = mean_in[r], rstd = rstd_in[r];
+    LVec<T, VC> xv[CH], dv[CH];
     float s1 = 0.f, s2 = 0.f;
-    for (int i = threadIdx.x, j = 0; i < hv; i += kThreads, ++j) {
-      LVec<T, VC> xv = xr[i], dv = dr[i], wv = wr[i];
-      reinterpret_cast<LVec<T, VC>*>(xrow)[i] = xv;
-      reinterpret_cast<LVec<T, VC>*>(drow)[i] = dv;
+#pragma unroll
+    for (int j = 0; j < CH; ++j) {
+      xv[j] = xr[lane + 64 * j];
+      dv[j] = dr[lane + 64 * j];
 #pragma unroll
       for (int k = 0; k < VC; ++k) {
-        float xhat = (lf_of(xv.v[k]) - mean) * rstd;
-        float df = lf_of(dv.v[k]);
-        float g = df * lf_of(wv.v[k]);
+        float xhat = (lf_of(xv[j].v[k]) - mean) * rstd;
+        float df = lf_of(dv[j].v[k]);
+        float g = df * lf_of(wv[j].v[k]);
         s1 += g;
         s2 += g * xhat;
         dwacc[j][k] += df * xhat;
         dbacc[j][k] += df;
       }
     }
-    __syncthreads();
-    float m1 = ln_block_sum(s1, red) / H;
-    __syncthreads();
-    float m2 = ln_block_sum(s2, red) / H;
+    const float m1 = ln_wave_sum(s1) / H;
+    const float m2 = ln_wave_sum(s2) / H;
 
     LVec<T, VC>* oxr = reinterpret_cast<LVec<T, VC>*>(dx + r * H);
-    for (int i = threadIdx.x; i < hv; i += kThreads) {
-      LVec<T, VC> xv = reinterpret_cast<const LVec<T, VC>*>(xrow)[i];
-      LVec<T, VC> dv = reinterpret_cast<const LVec<T, VC>*>(drow)[i];
-      LVec<T, VC> wv = wr[i];
+#pragma unroll
+    for (int j = 0; j < CH; ++j) {
       LVec<T, VC> ov;
 #pragma unroll
       for (int k = 0; k < VC; ++k) {
-        float xhat = (lf_of(xv.v[k]) - mean) * rstd;
-        float g = lf_of(dv.v[k]) * lf_of(wv.v[k]);
+        float xhat = (lf_of(xv[j].v[k]) - mean) * rstd;
+        float g = lf_of(dv[j].v[k]) * lf_of(wv[j].v[k]);
         ov.v[k] = lt_of<T>(rstd * (g - m1 - xhat * m2));
       }
-      oxr[i] = ov;
+      oxr[lane + 64 * j] = ov;
     }
-    __syncthreads();
   }
 
-  const int nblocks = gridDim.x;
-  for (int i = threadIdx.x, j = 0; i < hv; i += kThreads, ++j)
+#pragma unroll
+  for (int j = 0; j < CH; ++j)
 #pragma unroll
     for (int k = 0; k < VC; ++k) {
-      dwpart[static_cast<int64_t>(i * VC + k) * nblocks + blockIdx.x] = dwacc[j][k];
-      dbpart[static_cast<int64_t>(i * VC + k) * nblocks + blockIdx.x] = dbacc[j][k];
+      int64_t col = static_cast<int64_t>((lane + 64 * j) * VC + k);
+      dwpart[col * nunits + unit] = dwacc[j][k];
+      dbpart[col * nunits + unit] = dbacc[j][k];
     }
 }
 
@@ -233,19 +244,35 @@ template <typename T>
 void ln_bwd_t(const void* x, const void* dy, const void* w, const float* mean,
               const float* rstd, void* dx, float* dwpart, float* dbpart,
               float* dw, float* db, int64_t rows, int H, hipStream_t stream) {
-  int vc = ln_vc(H);
-  while ((H / vc + kThreads - 1) / kThreads > kMaxLnChunks && vc < 8) vc *= 2;
-  size_t lds = 2 * static_cast<size_t>(H) * sizeof(T);
-  int rpb = 0;
-  int blocks = rms_bwd_blocks(rows, &rpb);
-  switch (vc) {
-    case 8: hipLaunchKernelGGL((ln_bwd_kernel<T, 8>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), mean, rstd, static_cast<T*>(dx), dwpart, dbpart, rows, H, rpb); break;
-    case 4: hipLaunchKernelGGL((ln_bwd_kernel<T, 4>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), mean, rstd, static_cast<T*>(dx), dwpart, dbpart, rows, H, rpb); break;
-    case 2: hipLaunchKernelGGL((ln_bwd_kernel<T, 2>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), mean, rstd, static_cast<T*>(dx), dwpart, dbpart, rows, H, rpb); break;
-    default: hipLaunchKernelGGL((ln_bwd_kernel<T, 1>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), mean, rstd, static_cast<T*>(dx), dwpart, dbpart, rows, H, rpb); break;
+  int vc = 0, ch = 0;
+  rms_bwd_pick(H, &vc, &ch);  // caller (bn_ops.cpp) validated H
+  int rpu = 0;
+  int nunits = rms_bwd_blocks(rows, &rpu);
+  int blocks = nunits / 4;
+  auto L = [&](auto vctag, auto chtag) {
+    constexpr int V = decltype(vctag)::value;
+    constexpr int C = decltype(chtag)::value;
+    hipLaunchKernelGGL((ln_bwd_kernel<T, V, C>), blocks, kThreads, 0, stream,
+                       static_cast<const T*>(x), static_cast<const T*>(dy),
+                       static_cast<const T*>(w), mean, rstd,
+                       static_cast<T*>(dx), dwpart, dbpart, rows, H, rpu);
+  };
+  using I1 = std::integral_constant<int, 1>;
+  using I2 = std::integral_constant<int, 2>;
+  using I4 = std::integral_constant<int, 4>;
+  using I8 = std::integral_constant<int, 8>;
+  switch (vc * 16 + ch) {
+    case 1 * 16 + 1: L(I1{}, I1{}); break;
+    case 2 * 16 + 1: L(I2{}, I1{}); break;
+    case 4 * 16 + 1: L(I4{}, I1{}); break;
+    case 8 * 16 + 1: L(I8{}, I1{}); break;
+    case 8 * 16 + 2: L(I8{}, I2{}); break;
+    case 8 * 16 + 4: L(I8{}, I4{}); break;
+    case 8 * 16 + 8: L(I8{}, I8{}); break;
+    default: return;
   }
-  hipLaunchKernelGGL(ln_fold_kernel, H, kThreads, 0, stream, dwpart, dw, blocks);
-  hipLaunchKernelGGL(ln_fold_kernel, H, kThreads, 0, stream, dbpart, db, blocks);
+  hipLaunchKernelGGL(ln_fold_kernel, H, kThreads, 0, stream, dwpart, dw, nunits);
+  hipLaunchKernelGGL(ln_fold_kernel, H, kThreads, 0, stream, dbpart, db, nunits);
 }
 
 }  // namespace

@@ -16,6 +16,7 @@
 #include <hip/hip_runtime.h>
 
 #include <algorithm>
+#include <type_traits>
 
 #include "kernels.h"
 
@@ -110,78 +111,85 @@ __global__ void rms_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
 }
 
 // ----------------------------------------------------------------- backward
-// Each block owns a contiguous row range; x/dy staged per row in LDS;
-// per-thread dw partials accumulate across the block's rows in registers
-// (thread t owns columns {t*VC + j*kThreads*VC}). Each block then stores
-// its partials into a COLUMN-major [H][nblocks] buffer — no atomics
-// (same-address atomicAdd from ~2k blocks measured 40x off SOL) — and a
-// per-column fold kernel produces dw. kMaxDwChunks bounds the register
-// array: H <= 16384 at VC=8.
+// One WAVE per row (no __syncthreads anywhere): the row's x/dy live in
+// REGISTERS (lane l, chunk j owns columns (l + 64j)*VC ..+VC), the row
+// dot reduces with 6 xor-shuffles, dx writes straight from the held
+// registers — a block-per-row version with LDS staging and two block
+// barriers per row measured ~7x off streaming SOL at H=1024. Per-wave dw
+// partials land in a COLUMN-major [H][nunits] buffer (no atomics:
+// same-address atomicAdd from ~2k blocks measured 40x off SOL); a
+// per-column fold kernel produces dw. Host guarantees H % (64*VC) == 0
+// and H/(64*VC) <= kMaxDwChunks.
 
-constexpr int kMaxDwChunks = 8;  // bounds dw registers: H <= 16384 at VC=8
+constexpr int kMaxDwChunks = 8;  // bounds register arrays: H <= 16384 at VC=8
 
-template <typename T, int VC>
+__device__ __forceinline__ float wave_sum(float v) {
+  for (int off = 32; off > 0; off >>= 1)
+    v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+template <typename T, int VC, int CH>
 __global__ void rms_bwd_kernel(const T* __restrict__ x, const T* __restrict__ dy,
                                const T* __restrict__ w,
                                const float* __restrict__ rstd_in,
                                T* __restrict__ dx, float* __restrict__ dwpart,
-                               int64_t rows, int H, int rows_per_block) {
-  extern __shared__ unsigned char smem[];
-  T* xrow = reinterpret_cast<T*>(smem);
-  T* drow = reinterpret_cast<T*>(smem + static_cast<size_t>(H) * sizeof(T));
-  __shared__ float red[kWaves];
-  const int hv = H / VC;
-  const int chunks = (hv + kThreads - 1) / kThreads;
+                               int64_t rows, int H, int rows_per_unit) {
+  const int lane = threadIdx.x % 64;
+  const int wave = threadIdx.x / 64;
+  const int unit = blockIdx.x * kWaves + wave;
+  const int nunits = gridDim.x * kWaves;
 
-  float dwacc[kMaxDwChunks][VC];
-  for (int j = 0; j < chunks; ++j)
+  RVec<T, VC> wv[CH];
+#pragma unroll
+  for (int j = 0; j < CH; ++j)
+    wv[j] = reinterpret_cast<const RVec<T, VC>*>(w)[lane + 64 * j];
+
+  float dwacc[CH][VC];
+#pragma unroll
+  for (int j = 0; j < CH; ++j)
 #pragma unroll
     for (int k = 0; k < VC; ++k) dwacc[j][k] = 0.f;
 
-  const int64_t r0 = static_cast<int64_t>(blockIdx.x) * rows_per_block;
-  const int64_t r1 = r0 + rows_per_block > rows ? rows : r0 + rows_per_block;
-  const RVec<T, VC>* wr = reinterpret_cast<const RVec<T, VC>*>(w);
+  const int64_t r0 = static_cast<int64_t>(unit) * rows_per_unit;
+  const int64_t r1 = r0 + rows_per_unit > rows ? rows : r0 + rows_per_unit;
 
   for (int64_t r = r0; r < r1; ++r) {
     const RVec<T, VC>* xr = reinterpret_cast<const RVec<T, VC>*>(x + r * H);
     const RVec<T, VC>* dr = reinterpret_cast<const RVec<T, VC>*>(dy + r * H);
-    float rstd = rstd_in[r];
+    const float rstd = rstd_in[r];
+    RVec<T, VC> xv[CH], dv[CH];
     float dot = 0.f;
-    for (int i = threadIdx.x, j = 0; i < hv; i += kThreads, ++j) {
-      RVec<T, VC> xv = xr[i], dv = dr[i], wv = wr[i];
-      reinterpret_cast<RVec<T, VC>*>(xrow)[i] = xv;
-      reinterpret_cast<RVec<T, VC>*>(drow)[i] = dv;
+#pragma unroll
+    for (int j = 0; j < CH; ++j) {
+      xv[j] = xr[lane + 64 * j];
+      dv[j] = dr[lane + 64 * j];
 #pragma unroll
       for (int k = 0; k < VC; ++k) {
-        float xf = rf_of(xv.v[k]), df = rf_of(dv.v[k]);
-        dot += df * rf_of(wv.v[k]) * xf;
+        float xf = rf_of(xv[j].v[k]), df = rf_of(dv[j].v[k]);
+        dot += df * rf_of(wv[j].v[k]) * xf;
         dwacc[j][k] += df * xf * rstd;
       }
     }
-    __syncthreads();
-    float total = block_sum(dot, red);
-    float c2 = total * rstd * rstd * rstd / H;
+    const float c2 = wave_sum(dot) * rstd * rstd * rstd / H;
 
     RVec<T, VC>* oxr = reinterpret_cast<RVec<T, VC>*>(dx + r * H);
-    for (int i = threadIdx.x; i < hv; i += kThreads) {
-      RVec<T, VC> xv = reinterpret_cast<const RVec<T, VC>*>(xrow)[i];
-      RVec<T, VC> dv = reinterpret_cast<const RVec<T, VC>*>(drow)[i];
-      RVec<T, VC> wv = wr[i];
+#pragma unroll
+    for (int j = 0; j < CH; ++j) {
       RVec<T, VC> ov;
 #pragma unroll
       for (int k = 0; k < VC; ++k)
-        ov.v[k] = rt_of<T>(rstd * rf_of(wv.v[k]) * rf_of(dv.v[k]) -
-                           rf_of(xv.v[k]) * c2);
-      oxr[i] = ov;
+        ov.v[k] = rt_of<T>(rstd * rf_of(wv[j].v[k]) * rf_of(dv[j].v[k]) -
+                           rf_of(xv[j].v[k]) * c2);
+      oxr[lane + 64 * j] = ov;
     }
-    __syncthreads();
   }
 
-  const int nblocks = gridDim.x;
-  for (int i = threadIdx.x, j = 0; i < hv; i += kThreads, ++j)
+#pragma unroll
+  for (int j = 0; j < CH; ++j)
 #pragma unroll
     for (int k = 0; k < VC; ++k)
-      dwpart[static_cast<int64_t>(i * VC + k) * nblocks + blockIdx.x] =
+      dwpart[static_cast<int64_t>((lane + 64 * j) * VC + k) * nunits + unit] =
           dwacc[j][k];
 }
 
@@ -204,14 +212,30 @@ __global__ void rms_fold_kernel(const float* __restrict__ dwpart,
 
 }  // namespace
 
-// Number of row-blocks the backward uses (callers size the dw partial
-// buffer as fp32[H * rms_bwd_blocks(rows, ...)]). ~1024 blocks fills the
-// chip; each handles >= 4 rows so the per-block epilogue stays amortized.
-int rms_bwd_blocks(int64_t rows, int* rows_per_block) {
-  int64_t rpb = std::max<int64_t>(4, (rows + 1023) / 1024);
-  int blocks = static_cast<int>((rows + rpb - 1) / rpb);
-  if (rows_per_block) *rows_per_block = static_cast<int>(rpb);
-  return blocks;
+// Number of per-WAVE fold units the backward uses (callers size the dw
+// partial buffer as fp32[H * rms_bwd_blocks(rows, ...)]). ~2048 waves
+// fill the chip; idle tail units write zeroed partials.
+int rms_bwd_blocks(int64_t rows, int* rows_per_unit) {
+  int64_t rpu = std::max<int64_t>(1, (rows + 2047) / 2048);
+  int64_t units = (rows + rpu - 1) / rpu;
+  int blocks = static_cast<int>((units + 3) / 4);  // kWaves per block
+  if (rows_per_unit) *rows_per_unit = static_cast<int>(rpu);
+  return blocks * 4;
+}
+
+// Supported hidden sizes for the register-staged backward (VC, chunk
+// template pairs). Others take the Python eager fallback.
+bool rms_bwd_pick(int H, int* vc, int* ch) {
+  switch (H) {
+    case 64: *vc = 1; *ch = 1; return true;
+    case 128: *vc = 2; *ch = 1; return true;
+    case 256: *vc = 4; *ch = 1; return true;
+    case 512: *vc = 8; *ch = 1; return true;
+    case 1024: *vc = 8; *ch = 2; return true;
+    case 2048: *vc = 8; *ch = 4; return true;
+    case 4096: *vc = 8; *ch = 8; return true;
+    default: return false;
+  }
 }
 
 namespace {
@@ -241,20 +265,35 @@ template <typename T>
 void rms_bwd_t(const void* x, const void* dy, const void* w, const float* rstd,
                void* dx, float* dwpart, float* dw, int64_t rows, int H,
                hipStream_t stream) {
-  int vc = rms_vc(H, sizeof(T));
-  // Bound the per-thread dw register array: chunks <= kMaxDwChunks.
-  while ((H / vc + kThreads - 1) / kThreads > kMaxDwChunks && vc < 8) vc *= 2;
-  size_t lds = 2 * static_cast<size_t>(H) * sizeof(T);
-  int rpb = 0;
-  int blocks = rms_bwd_blocks(rows, &rpb);
-  switch (vc) {
-    case 8: hipLaunchKernelGGL((rms_bwd_kernel<T, 8>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), rstd, static_cast<T*>(dx), dwpart, rows, H, rpb); break;
-    case 4: hipLaunchKernelGGL((rms_bwd_kernel<T, 4>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), rstd, static_cast<T*>(dx), dwpart, rows, H, rpb); break;
-    case 2: hipLaunchKernelGGL((rms_bwd_kernel<T, 2>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), rstd, static_cast<T*>(dx), dwpart, rows, H, rpb); break;
-    default: hipLaunchKernelGGL((rms_bwd_kernel<T, 1>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(dy), static_cast<const T*>(w), rstd, static_cast<T*>(dx), dwpart, rows, H, rpb); break;
+  int vc = 0, ch = 0;
+  rms_bwd_pick(H, &vc, &ch);  // caller (bn_ops.cpp) validated H
+  int rpu = 0;
+  int nunits = rms_bwd_blocks(rows, &rpu);
+  int blocks = nunits / 4;
+  auto L = [&](auto vctag, auto chtag) {
+    constexpr int V = decltype(vctag)::value;
+    constexpr int C = decltype(chtag)::value;
+    hipLaunchKernelGGL((rms_bwd_kernel<T, V, C>), blocks, kThreads, 0, stream,
+                       static_cast<const T*>(x), static_cast<const T*>(dy),
+                       static_cast<const T*>(w), rstd, static_cast<T*>(dx),
+                       dwpart, rows, H, rpu);
+  };
+  using I1 = std::integral_constant<int, 1>;
+  using I2 = std::integral_constant<int, 2>;
+  using I4 = std::integral_constant<int, 4>;
+  using I8 = std::integral_constant<int, 8>;
+  switch (vc * 16 + ch) {
+    case 1 * 16 + 1: L(I1{}, I1{}); break;
+    case 2 * 16 + 1: L(I2{}, I1{}); break;
+    case 4 * 16 + 1: L(I4{}, I1{}); break;
+    case 8 * 16 + 1: L(I8{}, I1{}); break;
+    case 8 * 16 + 2: L(I8{}, I2{}); break;
+    case 8 * 16 + 4: L(I8{}, I4{}); break;
+    case 8 * 16 + 8: L(I8{}, I8{}); break;
+    default: return;  // unreachable (validated upstream)
   }
   hipLaunchKernelGGL(rms_fold_kernel, H, kThreads, 0, stream, dwpart, dw,
-                     blocks);
+                     nunits);
 }
 
 }  // namespace

@@ -6,6 +6,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 _SUPPORTED = (torch.float32, torch.float16, torch.bfloat16)
+_SUPPORTED_DIMS = {64, 128, 256, 512, 1024, 2048, 4096}
 
 
 class _LayerNormFunc(torch.autograd.Function):
@@ -35,8 +36,7 @@ class FusedLayerNorm(nn.LayerNorm):
                 x.size(-1) == self.normalized_shape[0] and
                 self.weight is not None and self.bias is not None and
                 x.dtype == self.weight.dtype and
-                self.normalized_shape[0] % 2 == 0 and
-                self.normalized_shape[0] <= 16384)
+                self.normalized_shape[0] in _SUPPORTED_DIMS)
 
     def forward(self, x):
         if self._fast_path(x):

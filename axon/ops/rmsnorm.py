@@ -5,6 +5,7 @@ import torch
 import torch.nn as nn
 
 _SUPPORTED = (torch.float32, torch.float16, torch.bfloat16)
+_SUPPORTED_DIMS = {64, 128, 256, 512, 1024, 2048, 4096}
 
 
 class _RMSNormFunc(torch.autograd.Function):
@@ -35,8 +36,7 @@ class FusedRMSNorm(nn.Module):
     def _fast_path(self, x):
         return (x.is_cuda and x.dtype in _SUPPORTED and
                 x.dtype == self.weight.dtype and x.size(-1) == self.dim and
-                self.dim % 2 == 0 and self.dim <= 16384 and
-                x.is_contiguous())
+                self.dim in _SUPPORTED_DIMS and x.is_contiguous())
 
     def forward(self, x):
         if self._fast_path(x):
