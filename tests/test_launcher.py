@@ -89,17 +89,24 @@ def test_timeline_written():
 
 def test_duplicate_inflight_name_rejected():
     """Two async ops with the same name in flight must error loudly, not
-    corrupt the pending table."""
+    corrupt the pending table. Race-free setup: rank 1 delays its first
+    submission, so rank 0's op is GUARANTEED still pending (negotiation
+    cannot finish without rank 1) when rank 0 re-submits the name."""
     script = (
-        "import torch, axon as hvd\n"
+        "import time, torch, axon as hvd\n"
         "hvd.init()\n"
-        "h1 = hvd.allreduce_async_(torch.ones(4), name='dup')\n"
-        "try:\n"
-        "    h2 = hvd.allreduce_async_(torch.ones(4), name='dup')\n"
-        "    hvd.synchronize(h2)\n"
-        "    raise SystemExit('second submit should have failed')\n"
-        "except RuntimeError as e:\n"
-        "    assert 'unique name' in str(e) or 'in flight' in str(e), e\n"
+        "if hvd.rank() == 0:\n"
+        "    h1 = hvd.allreduce_async_(torch.ones(4), name='dup')\n"
+        "    time.sleep(0.2)  # let the first submit drain into pending\n"
+        "    try:\n"
+        "        h2 = hvd.allreduce_async_(torch.ones(4), name='dup')\n"
+        "        hvd.synchronize(h2)\n"
+        "        raise SystemExit('second submit should have failed')\n"
+        "    except RuntimeError as e:\n"
+        "        assert 'unique name' in str(e) or 'in flight' in str(e), e\n"
+        "else:\n"
+        "    time.sleep(0.8)\n"
+        "    h1 = hvd.allreduce_async_(torch.ones(4), name='dup')\n"
         "hvd.synchronize(h1)\n"
         "hvd.shutdown()\n")
     r = _run(2, script)
