@@ -289,8 +289,9 @@ void bg_loop(std::promise<void> ready) {
         cycle = std::chrono::duration<double, std::milli>(g.cycle_ms);
       }
       for (auto& r : responses) {
-        if (g.pm && r.op == OpType::ALLREDUCE && r.device >= 0 &&
-            r.error.empty()) {
+        // Score by allreduced bytes/sec on whichever data plane is in
+        // use (the CPU path also makes the tuner testable without a GPU).
+        if (g.pm && r.op == OpType::ALLREDUCE && r.error.empty()) {
           int64_t bytes = 0;
           size_t esz = dtype_size(id_dtype(r.wire_dtype ? r.wire_dtype : r.dtype));
           for (auto nn : r.numels) bytes += nn * static_cast<int64_t>(esz);

@@ -10,6 +10,7 @@ ParameterManager::ParameterManager(int64_t threshold, double cycle_ms) {
   enabled_ = env_int("AXON_AUTOTUNE", 0) != 0;
   log_ = env_int("AXON_AUTOTUNE_LOG", 0) != 0;
   sample_sec_ = env_double("AXON_AUTOTUNE_SAMPLE_SEC", 1.0);
+  warmup_until_ = env_double("AXON_AUTOTUNE_WARMUP_SEC", 3.0);
   best_ = Sample{threshold, cycle_ms, -1.0};
   if (!enabled_) return;
   // Coarse 2-D neighborhood around the xGMI-tuned defaults. Thresholds in

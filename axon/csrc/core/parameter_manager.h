@@ -40,7 +40,7 @@ class ParameterManager {
   bool done_ = false;
   int64_t bytes_ = 0;
   std::chrono::steady_clock::time_point window_start_;
-  double warmup_until_ = 3.0;  // seconds before sampling starts
+  double warmup_until_ = 3.0;  // seconds before sampling (AXON_AUTOTUNE_WARMUP_SEC)
   double sample_sec_;
   std::chrono::steady_clock::time_point t0_;
   std::vector<Sample> plan_;

@@ -75,7 +75,10 @@ void launch_bn_bwd(int x_dtype, int param_dtype, const void* x, const void* dy,
 void launch_rms_fwd(int dtype, const void* x, const void* w, void* y,
                     float* rstd, int64_t rows, int H, float eps,
                     hipStream_t stream);
-int rms_bwd_blocks(int64_t rows, int* rows_per_block);
+int rms_bwd_blocks(int64_t rows, int* rows_per_unit);
+// Supported (VC, chunk) pairs for the register-staged LN/RMS backward:
+// H in {64,128,256,512,1024,2048,4096}; false => Python eager fallback.
+bool rms_bwd_pick(int H, int* vc, int* ch);
 void launch_rms_bwd(int dtype, const void* x, const void* dy, const void* w,
                     const float* rstd, void* dx, float* dwpart, float* dw,
                     int64_t rows, int H, hipStream_t stream);

@@ -146,3 +146,23 @@ def test_stall_inspector_warns():
     r = _run(2, script, env={"AXON_STALL_CHECK_TIME": "1"})
     assert r.returncode == 0, r.stderr
     assert "stall inspector" in r.stderr
+
+
+def test_autotune_converges():
+    """AXON_AUTOTUNE=1: the coordinator sweeps the (threshold, cycle) grid
+    scored by allreduced bytes/s, ships knob updates in ResponseLists, and
+    settles on a best sample (SURVEY.md §2.1 parameter manager)."""
+    script = (
+        "import torch, axon as hvd\n"
+        "hvd.init()\n"
+        "t = torch.ones(1 << 16)\n"
+        "for i in range(400):\n"
+        "    hvd.allreduce(t, average=True, name=f'a{i % 7}')\n"
+        "hvd.shutdown()\n")
+    r = _run(2, script, env={"AXON_AUTOTUNE": "1", "AXON_AUTOTUNE_LOG": "1",
+                             "AXON_AUTOTUNE_SAMPLE_SEC": "0.05",
+                             "AXON_AUTOTUNE_WARMUP_SEC": "0.1",
+                             "AXON_LOG_LEVEL": "info"}, timeout=150)
+    assert r.returncode == 0, r.stderr
+    assert "autotune sample" in r.stderr
+    assert "autotune done" in r.stderr
