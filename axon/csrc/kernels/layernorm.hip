@@ -1,7 +1,7 @@
 // Fused LayerNorm for CDNA4 (the BERT path; same design as rmsnorm.hip):
-// one kernel and one global read of x per direction, rows staged through
-// LDS, fp32 arithmetic, column-major partial buffers for dgamma/dbeta
-// (no atomics).
+// one kernel and one global read of x per direction, one WAVE per row
+// with the row held in registers (no LDS, no barriers), fp32 arithmetic,
+// column-major partial buffers for dgamma/dbeta (no atomics).
 //
 //   fwd: y = (x - mean) * rstd * w + b ; saves per-row mean/rstd (fp32)
 //   bwd: g = dy*w ; dx = rstd*(g - mean_r(g) - xhat*mean_r(g*xhat))
@@ -50,70 +50,67 @@ struct alignas(sizeof(T) * V) LVec {
   T v[V];
 };
 
-__device__ float ln_block_sum(float v, float* red) {
+__device__ __forceinline__ float ln_wave_sum(float v) {
   for (int off = 32; off > 0; off >>= 1)
     v += __shfl_xor(v, off, 64);
-  const int lane = threadIdx.x % 64, wave = threadIdx.x / 64;
-  if (lane == 0) red[wave] = v;
-  __syncthreads();
-  float total = red[0];
-#pragma unroll
-  for (int w = 1; w < kWaves; ++w) total += red[w];
-  return total;
+  return v;
 }
 
 // ------------------------------------------------------------------ forward
+// One WAVE per row, registers only (see the backward below).
 
-template <typename T, int VC>
+template <typename T, int VC, int CH>
 __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
                               const T* __restrict__ b, T* __restrict__ y,
                               float* __restrict__ mean_out,
                               float* __restrict__ rstd_out, int64_t rows, int H,
-                              float eps) {
-  extern __shared__ unsigned char smem[];
-  T* row = reinterpret_cast<T*>(smem);
-  __shared__ float red[kWaves];
-  const int hv = H / VC;
+                              float eps, int rows_per_unit) {
+  const int lane = threadIdx.x % 64;
+  const int wave = threadIdx.x / 64;
+  const int unit = blockIdx.x * kWaves + wave;
 
-  for (int64_t r = blockIdx.x; r < rows; r += gridDim.x) {
+  LVec<T, VC> wv[CH], bv[CH];
+#pragma unroll
+  for (int j = 0; j < CH; ++j) {
+    wv[j] = reinterpret_cast<const LVec<T, VC>*>(w)[lane + 64 * j];
+    bv[j] = reinterpret_cast<const LVec<T, VC>*>(b)[lane + 64 * j];
+  }
+
+  const int64_t r0 = static_cast<int64_t>(unit) * rows_per_unit;
+  const int64_t r1 = r0 + rows_per_unit > rows ? rows : r0 + rows_per_unit;
+
+  for (int64_t r = r0; r < r1; ++r) {
     const LVec<T, VC>* xr = reinterpret_cast<const LVec<T, VC>*>(x + r * H);
+    LVec<T, VC> xv[CH];
     float s = 0.f, q = 0.f;
-    for (int i = threadIdx.x; i < hv; i += kThreads) {
-      LVec<T, VC> xv = xr[i];
-      reinterpret_cast<LVec<T, VC>*>(row)[i] = xv;
+#pragma unroll
+    for (int j = 0; j < CH; ++j) {
+      xv[j] = xr[lane + 64 * j];
 #pragma unroll
       for (int k = 0; k < VC; ++k) {
-        float f = lf_of(xv.v[k]);
+        float f = lf_of(xv[j].v[k]);
         s += f;
         q += f * f;
       }
     }
-    __syncthreads();
-    float tot_s = ln_block_sum(s, red);
-    __syncthreads();  // red[] reuse between the two reductions
-    float tot_q = ln_block_sum(q, red);
-    float mean = tot_s / H;
-    float var = fmaxf(tot_q / H - mean * mean, 0.f);
-    float rstd = rsqrtf(var + eps);
-    if (threadIdx.x == 0) {
+    const float mean = ln_wave_sum(s) / H;
+    const float var = fmaxf(ln_wave_sum(q) / H - mean * mean, 0.f);
+    const float rstd = rsqrtf(var + eps);
+    if (lane == 0) {
       if (mean_out) mean_out[r] = mean;
       if (rstd_out) rstd_out[r] = rstd;
     }
 
     LVec<T, VC>* yr = reinterpret_cast<LVec<T, VC>*>(y + r * H);
-    const LVec<T, VC>* wr = reinterpret_cast<const LVec<T, VC>*>(w);
-    const LVec<T, VC>* br = reinterpret_cast<const LVec<T, VC>*>(b);
-    for (int i = threadIdx.x; i < hv; i += kThreads) {
-      LVec<T, VC> xv = reinterpret_cast<const LVec<T, VC>*>(row)[i];
-      LVec<T, VC> wv = wr[i], bv = br[i];
+#pragma unroll
+    for (int j = 0; j < CH; ++j) {
       LVec<T, VC> yv;
 #pragma unroll
       for (int k = 0; k < VC; ++k)
-        yv.v[k] = lt_of<T>(fmaf((lf_of(xv.v[k]) - mean) * rstd, lf_of(wv.v[k]),
-                                lf_of(bv.v[k])));
-      yr[i] = yv;
+        yv.v[k] = lt_of<T>(fmaf((lf_of(xv[j].v[k]) - mean) * rstd,
+                                lf_of(wv[j].v[k]), lf_of(bv[j].v[k])));
+      yr[lane + 64 * j] = yv;
     }
-    __syncthreads();
   }
 }
 
@@ -121,12 +118,6 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
 // One WAVE per row, registers only (see rmsnorm.hip backward: the
 // block-per-row/LDS version measured ~7x off streaming SOL at H=1024).
 // Two row sums (s1 = sum g, s2 = sum g*xhat) reduce with xor-shuffles.
-
-__device__ __forceinline__ float ln_wave_sum(float v) {
-  for (int off = 32; off > 0; off >>= 1)
-    v += __shfl_xor(v, off, 64);
-  return v;
-}
 
 template <typename T, int VC, int CH>
 __global__ void ln_bwd_kernel(const T* __restrict__ x, const T* __restrict__ dy,
@@ -229,14 +220,31 @@ template <typename T>
 void ln_fwd_t(const void* x, const void* w, const void* b, void* y,
               float* mean, float* rstd, int64_t rows, int H, float eps,
               hipStream_t stream) {
-  int vc = ln_vc(H);
-  size_t lds = static_cast<size_t>(H) * sizeof(T);
-  int blocks = static_cast<int>(std::min<int64_t>(rows, 16384));
-  switch (vc) {
-    case 8: hipLaunchKernelGGL((ln_fwd_kernel<T, 8>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(w), static_cast<const T*>(b), static_cast<T*>(y), mean, rstd, rows, H, eps); break;
-    case 4: hipLaunchKernelGGL((ln_fwd_kernel<T, 4>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(w), static_cast<const T*>(b), static_cast<T*>(y), mean, rstd, rows, H, eps); break;
-    case 2: hipLaunchKernelGGL((ln_fwd_kernel<T, 2>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(w), static_cast<const T*>(b), static_cast<T*>(y), mean, rstd, rows, H, eps); break;
-    default: hipLaunchKernelGGL((ln_fwd_kernel<T, 1>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(w), static_cast<const T*>(b), static_cast<T*>(y), mean, rstd, rows, H, eps); break;
+  int vc = 0, ch = 0;
+  rms_bwd_pick(H, &vc, &ch);  // caller validated H
+  int rpu = 0;
+  int blocks = rms_bwd_blocks(rows, &rpu) / 4;
+  auto L = [&](auto vctag, auto chtag) {
+    constexpr int V = decltype(vctag)::value;
+    constexpr int C = decltype(chtag)::value;
+    hipLaunchKernelGGL((ln_fwd_kernel<T, V, C>), blocks, kThreads, 0, stream,
+                       static_cast<const T*>(x), static_cast<const T*>(w),
+                       static_cast<const T*>(b), static_cast<T*>(y), mean,
+                       rstd, rows, H, eps, rpu);
+  };
+  using I1 = std::integral_constant<int, 1>;
+  using I2 = std::integral_constant<int, 2>;
+  using I4 = std::integral_constant<int, 4>;
+  using I8 = std::integral_constant<int, 8>;
+  switch (vc * 16 + ch) {
+    case 1 * 16 + 1: L(I1{}, I1{}); break;
+    case 2 * 16 + 1: L(I2{}, I1{}); break;
+    case 4 * 16 + 1: L(I4{}, I1{}); break;
+    case 8 * 16 + 1: L(I8{}, I1{}); break;
+    case 8 * 16 + 2: L(I8{}, I2{}); break;
+    case 8 * 16 + 4: L(I8{}, I4{}); break;
+    case 8 * 16 + 8: L(I8{}, I8{}); break;
+    default: return;
   }
 }
 

@@ -67,46 +67,58 @@ __device__ float block_sum(float v, float* red) {
 }
 
 // ------------------------------------------------------------------ forward
-// One block per row; row staged in LDS (bytes = H * sizeof(T)).
+// One WAVE per row, registers only (same structure as the backward below:
+// the row lives in lane registers, the sum-of-squares reduces with 6
+// xor-shuffles, no LDS and no barriers).
 
-template <typename T, int VC>
+__device__ __forceinline__ float rms_wave_sum_f(float v) {
+  for (int off = 32; off > 0; off >>= 1)
+    v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+template <typename T, int VC, int CH>
 __global__ void rms_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
                                T* __restrict__ y, float* __restrict__ rstd_out,
-                               int64_t rows, int H, float eps) {
-  extern __shared__ unsigned char smem[];
-  T* row = reinterpret_cast<T*>(smem);
-  __shared__ float red[kWaves];
-  const int hv = H / VC;
+                               int64_t rows, int H, float eps,
+                               int rows_per_unit) {
+  const int lane = threadIdx.x % 64;
+  const int wave = threadIdx.x / 64;
+  const int unit = blockIdx.x * kWaves + wave;
 
-  for (int64_t r = blockIdx.x; r < rows; r += gridDim.x) {
+  RVec<T, VC> wv[CH];
+#pragma unroll
+  for (int j = 0; j < CH; ++j)
+    wv[j] = reinterpret_cast<const RVec<T, VC>*>(w)[lane + 64 * j];
+
+  const int64_t r0 = static_cast<int64_t>(unit) * rows_per_unit;
+  const int64_t r1 = r0 + rows_per_unit > rows ? rows : r0 + rows_per_unit;
+
+  for (int64_t r = r0; r < r1; ++r) {
     const RVec<T, VC>* xr = reinterpret_cast<const RVec<T, VC>*>(x + r * H);
+    RVec<T, VC> xv[CH];
     float ss = 0.f;
-    for (int i = threadIdx.x; i < hv; i += kThreads) {
-      RVec<T, VC> xv = xr[i];
-      reinterpret_cast<RVec<T, VC>*>(row)[i] = xv;  // stage for pass 2
+#pragma unroll
+    for (int j = 0; j < CH; ++j) {
+      xv[j] = xr[lane + 64 * j];
 #pragma unroll
       for (int k = 0; k < VC; ++k) {
-        float f = rf_of(xv.v[k]);
+        float f = rf_of(xv[j].v[k]);
         ss += f * f;
       }
     }
-    __syncthreads();  // red[] reuse across rows
-    float total = block_sum(ss, red);
-    float rstd = rsqrtf(total / H + eps);
-    if (threadIdx.x == 0 && rstd_out) rstd_out[r] = rstd;
+    const float rstd = rsqrtf(rms_wave_sum_f(ss) / H + eps);
+    if (lane == 0 && rstd_out) rstd_out[r] = rstd;
 
     RVec<T, VC>* yr = reinterpret_cast<RVec<T, VC>*>(y + r * H);
-    const RVec<T, VC>* wr = reinterpret_cast<const RVec<T, VC>*>(w);
-    for (int i = threadIdx.x; i < hv; i += kThreads) {
-      RVec<T, VC> xv = reinterpret_cast<const RVec<T, VC>*>(row)[i];
-      RVec<T, VC> wv = wr[i];
+#pragma unroll
+    for (int j = 0; j < CH; ++j) {
       RVec<T, VC> yv;
 #pragma unroll
       for (int k = 0; k < VC; ++k)
-        yv.v[k] = rt_of<T>(rf_of(xv.v[k]) * rstd * rf_of(wv.v[k]));
-      yr[i] = yv;
+        yv.v[k] = rt_of<T>(rf_of(xv[j].v[k]) * rstd * rf_of(wv[j].v[k]));
+      yr[lane + 64 * j] = yv;
     }
-    __syncthreads();
   }
 }
 
@@ -250,14 +262,30 @@ int rms_vc(int H, size_t /*tsize*/) {
 template <typename T>
 void rms_fwd_t(const void* x, const void* w, void* y, float* rstd, int64_t rows,
                int H, float eps, hipStream_t stream) {
-  int vc = rms_vc(H, sizeof(T));
-  size_t lds = static_cast<size_t>(H) * sizeof(T);
-  int blocks = static_cast<int>(std::min<int64_t>(rows, 16384));
-  switch (vc) {
-    case 8: hipLaunchKernelGGL((rms_fwd_kernel<T, 8>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(w), static_cast<T*>(y), rstd, rows, H, eps); break;
-    case 4: hipLaunchKernelGGL((rms_fwd_kernel<T, 4>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(w), static_cast<T*>(y), rstd, rows, H, eps); break;
-    case 2: hipLaunchKernelGGL((rms_fwd_kernel<T, 2>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(w), static_cast<T*>(y), rstd, rows, H, eps); break;
-    default: hipLaunchKernelGGL((rms_fwd_kernel<T, 1>), blocks, kThreads, lds, stream, static_cast<const T*>(x), static_cast<const T*>(w), static_cast<T*>(y), rstd, rows, H, eps); break;
+  int vc = 0, ch = 0;
+  rms_bwd_pick(H, &vc, &ch);  // caller validated H
+  int rpu = 0;
+  int blocks = rms_bwd_blocks(rows, &rpu) / 4;
+  auto L = [&](auto vctag, auto chtag) {
+    constexpr int V = decltype(vctag)::value;
+    constexpr int C = decltype(chtag)::value;
+    hipLaunchKernelGGL((rms_fwd_kernel<T, V, C>), blocks, kThreads, 0, stream,
+                       static_cast<const T*>(x), static_cast<const T*>(w),
+                       static_cast<T*>(y), rstd, rows, H, eps, rpu);
+  };
+  using I1 = std::integral_constant<int, 1>;
+  using I2 = std::integral_constant<int, 2>;
+  using I4 = std::integral_constant<int, 4>;
+  using I8 = std::integral_constant<int, 8>;
+  switch (vc * 16 + ch) {
+    case 1 * 16 + 1: L(I1{}, I1{}); break;
+    case 2 * 16 + 1: L(I2{}, I1{}); break;
+    case 4 * 16 + 1: L(I4{}, I1{}); break;
+    case 8 * 16 + 1: L(I8{}, I1{}); break;
+    case 8 * 16 + 2: L(I8{}, I2{}); break;
+    case 8 * 16 + 4: L(I8{}, I4{}); break;
+    case 8 * 16 + 8: L(I8{}, I8{}); break;
+    default: return;
   }
 }
 
