@@ -243,9 +243,37 @@ std::vector<at::Tensor> swiglu_bwd(const at::Tensor& dz_in, const at::Tensor& a,
   return {da, dg};
 }
 
+// EXPERIMENTAL (docs/FA_PLAN.md): flash-attention forward, v0.
+std::vector<at::Tensor> fa_fwd(const at::Tensor& q, const at::Tensor& k,
+                               const at::Tensor& v, bool causal) {
+  TORCH_CHECK(q.is_cuda() && q.dim() == 4 && q.is_contiguous() &&
+                  k.is_contiguous() && v.is_contiguous(),
+              "fa_fwd: contiguous [B,H,S,D] GPU tensors required");
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16 &&
+                  k.scalar_type() == at::kBFloat16 &&
+                  v.scalar_type() == at::kBFloat16,
+              "fa_fwd: bf16 only (v0)");
+  const int B = q.size(0), HQ = q.size(1), S = q.size(2), D = q.size(3);
+  const int HK = k.size(1);
+  TORCH_CHECK(D == 128, "fa_fwd: head dim 128 only (v0)");
+  TORCH_CHECK(k.size(2) == S && v.size(2) == S && HQ % HK == 0,
+              "fa_fwd: shape mismatch");
+  at::Tensor o = at::empty_like(q);
+  at::Tensor lse = at::empty({B, HQ, S},
+                             at::TensorOptions().dtype(at::kFloat)
+                                 .device(q.device()));
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(q.get_device());
+  launch_fa_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                lse.data_ptr<float>(), B, HQ, HK, S, D, causal,
+                stream.stream());
+  return {o, lse};
+}
+
 }  // namespace
 
 void register_bn(pybind11::module_& m) {
+  m.def("fa_fwd", &fa_fwd, py::arg("q"), py::arg("k"), py::arg("v"),
+        py::arg("causal") = true);
   m.def("ln_fwd", &ln_fwd, py::arg("x"), py::arg("w"), py::arg("b"),
         py::arg("eps"));
   m.def("ln_bwd", &ln_bwd, py::arg("x"), py::arg("dy"), py::arg("w"),

@@ -106,4 +106,11 @@ void launch_swiglu_fwd(int dtype, const void* a, const void* g, void* z,
 void launch_swiglu_bwd(int dtype, const void* dz, const void* a, const void* g,
                        void* da, void* dg, int64_t numel, hipStream_t stream);
 
+// EXPERIMENTAL flash-attention forward (flash_attn.hip, docs/FA_PLAN.md):
+// q/k/v/o are [B, H(Q|K), S, 128] contiguous bf16; lse fp32 [B, HQ, S]
+// (nullable). v0 correctness build, reachable only via tools/fa_probe.py.
+void launch_fa_fwd(const void* q, const void* k, const void* v, void* o,
+                   float* lse, int B, int HQ, int HK, int S, int D,
+                   bool causal, hipStream_t stream);
+
 }  // namespace axon

@@ -41,9 +41,11 @@ def run_impl(impl, q, k, v, causal=True):
         return F.scaled_dot_product_attention(q, k, v, is_causal=causal,
                                               enable_gqa=True)
     if impl == "axon":
-        from axon.ops import flash_attention  # round-2 entry point
+        from axon import _core  # experimental fwd-only kernel (FA_PLAN v0)
 
-        return flash_attention(q, k, v, causal=causal)
+        with torch.no_grad():
+            o, _lse = _core.fa_fwd(q, k, v, causal)
+        return o
     raise ValueError(impl)
 
 
@@ -86,24 +88,30 @@ def bench(args):
     device, dtype = "cuda", torch.bfloat16
     q, k, v = make_case("random", args.batch, args.heads, args.kv_heads,
                         args.seq, args.head_dim, device, dtype)
-    q = q.requires_grad_(True)
+    fwd_only = args.impl == "axon"  # v0 kernel has no backward yet
+    if not fwd_only:
+        q = q.requires_grad_(True)
     for _ in range(3):  # warmup
         out = run_impl(args.impl, q, k, v)
-        out.sum().backward()
-        q.grad = None
+        if not fwd_only:
+            out.sum().backward()
+            q.grad = None
     torch.cuda.synchronize()
     import time
 
     t0 = time.perf_counter()
     for _ in range(args.steps):
         out = run_impl(args.impl, q, k, v)
-        out.sum().backward()
-        q.grad = None
+        if not fwd_only:
+            out.sum().backward()
+            q.grad = None
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / args.steps
-    flops = 2 * 2 * args.batch * args.heads * args.seq ** 2 * args.head_dim
-    flops = flops / 2 * 3.5  # causal halves; fwd+bwd ~ 3.5x fwd
-    print(f"{args.impl}: {dt * 1e3:.2f} ms/iter fwd+bwd, "
+    flops = 2 * 2 * args.batch * args.heads * args.seq ** 2 * args.head_dim / 2
+    if not fwd_only:
+        flops *= 3.5  # fwd+bwd ~ 3.5x fwd (causal already halved)
+    tag = "fwd" if fwd_only else "fwd+bwd"
+    print(f"{args.impl}: {dt * 1e3:.2f} ms/iter {tag}, "
           f"~{flops / dt / 1e12:.0f} TF effective")
 
 
