@@ -533,3 +533,27 @@ def test_bert_tiny_step_gpu(hvd):
         losses.append(loss.item())
     assert all(torch.isfinite(torch.tensor(losses)))
     assert losses[-1] < losses[0]
+
+
+def test_fa_fwd_experimental_matches_sdpa(hvd):
+    """Experimental CDNA4 flash-attention forward (docs/FA_PLAN.md v0,
+    probe-only — not wired into models): matches SDPA on causal GQA,
+    including the spiked-K rescale case and a tile-ragged length."""
+    from axon import _core
+
+    for kind, s in (("random", 192), ("random", 512), ("spiked", 256)):
+        torch.manual_seed(s)
+        q = torch.randn(2, 8, s, 128).clamp(-2, 2)
+        k = torch.randn(2, 2, s, 128).clamp(-2, 2)
+        v = torch.randn(2, 2, s, 128).clamp(-2, 2)
+        if kind == "spiked":
+            k[:, :, s // 2, :] = 6.0
+            q[:, :, 3 * s // 4, :] = 6.0
+        q, k, v = (t.to(DEV, torch.bfloat16) for t in (q, k, v))
+        with torch.no_grad():
+            out, lse = _core.fa_fwd(q, k, v, True)
+            ref = torch.nn.functional.scaled_dot_product_attention(
+                q, k, v, is_causal=True, enable_gqa=True)
+        assert torch.allclose(out.float(), ref.float(), rtol=3e-2, atol=3e-2), \
+            (kind, s, (out.float() - ref.float()).abs().max())
+        assert torch.isfinite(lse).all()
