@@ -56,3 +56,28 @@ def test_elastic_respawn_on_failure(tmp_path):
         assert res[w]["size"] == 3
         assert res[w]["step"] == 12
     assert abs(res[0]["wsum"] - res[2]["wsum"]) < 1e-9
+
+
+def test_elastic_double_failure_shrinks(tmp_path):
+    """Two workers die at the same step: one generation change handles the
+    multi-death sweep, the world shrinks 4 -> 2, survivors finish."""
+    out = str(tmp_path / f"el_{uuid.uuid4().hex[:8]}")
+    cmd = [sys.executable, "-m", "axon.runner.run", "-np", "4",
+           "--elastic", "--no-pin", "-v", "--min-np", "2",
+           "--max-restarts", "0",
+           sys.executable, os.path.join(REPO, "tests", "elastic_script2.py"),
+           out]
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["OMP_NUM_THREADS"] = "1"
+    r = subprocess.run(cmd, env=env, capture_output=True, text=True,
+                       timeout=180)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert os.path.exists(out + ".died2") and os.path.exists(out + ".died3")
+    res = {}
+    for w in (0, 1):
+        with open(f"{out}.w{w}.json") as f:
+            res[w] = json.load(f)
+    assert res[0]["size"] == 2 and res[1]["size"] == 2
+    assert res[0]["step"] == 10 and res[1]["step"] == 10
+    assert abs(res[0]["wsum"] - res[1]["wsum"]) < 1e-9
