@@ -269,10 +269,41 @@ std::vector<at::Tensor> fa_fwd(const at::Tensor& q, const at::Tensor& k,
   return {o, lse};
 }
 
+// EXPERIMENTAL: FA backward v0 (probe-only; docs/FA_PLAN.md).
+std::vector<at::Tensor> fa_bwd(const at::Tensor& q, const at::Tensor& k,
+                               const at::Tensor& v, const at::Tensor& o,
+                               const at::Tensor& dout, const at::Tensor& lse,
+                               bool causal) {
+  TORCH_CHECK(q.is_cuda() && q.dim() == 4 && q.is_contiguous() &&
+                  k.is_contiguous() && v.is_contiguous() &&
+                  o.is_contiguous() && lse.is_contiguous(),
+              "fa_bwd: contiguous GPU tensors required");
+  at::Tensor dy = dout.is_contiguous() ? dout : dout.contiguous();
+  const int B = q.size(0), HQ = q.size(1), S = q.size(2), D = q.size(3);
+  const int HK = k.size(1);
+  TORCH_CHECK(D == 128 && q.scalar_type() == at::kBFloat16,
+              "fa_bwd: bf16 head dim 128 only (v0)");
+  at::Tensor dq = at::empty_like(q);
+  at::Tensor dk = at::empty_like(k);
+  at::Tensor dv = at::empty_like(v);
+  at::Tensor delta = at::empty({static_cast<int64_t>(B) * HQ * S},
+                               at::TensorOptions().dtype(at::kFloat)
+                                   .device(q.device()));
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(q.get_device());
+  launch_fa_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), dy.data_ptr(),
+                o.data_ptr(), lse.data_ptr<float>(), delta.data_ptr<float>(),
+                dq.data_ptr(), dk.data_ptr(), dv.data_ptr(), B, HQ, HK, S, D,
+                causal, stream.stream());
+  return {dq, dk, dv};
+}
+
 }  // namespace
 
 void register_bn(pybind11::module_& m) {
   m.def("fa_fwd", &fa_fwd, py::arg("q"), py::arg("k"), py::arg("v"),
+        py::arg("causal") = true);
+  m.def("fa_bwd", &fa_bwd, py::arg("q"), py::arg("k"), py::arg("v"),
+        py::arg("o"), py::arg("dout"), py::arg("lse"),
         py::arg("causal") = true);
   m.def("ln_fwd", &ln_fwd, py::arg("x"), py::arg("w"), py::arg("b"),
         py::arg("eps"));

@@ -221,7 +221,333 @@ __global__ void fa_fwd_kernel(const bf16* __restrict__ q,
   }
 }
 
+// ---------------------------------------------------------------- backward
+// EXPERIMENTAL v0 (probe-only, never run in any model/test path until
+// round 2 validates it through tools/fa_probe.py --check-bwd). FA-2 style:
+// P recomputed from saved logsumexp; delta = rowsum(dO*O) precomputed.
+//
+//   dkdv kernel: one wave per 16-row K/V tile; inner loop over the
+//     (causal-reachable) 32-row q-tiles and, inside, over the q-heads
+//     sharing this kv head (GQA accumulates into the same dK/dV, so no
+//     atomics and no workspace).
+//       ST(16k x 32q) = K Q^T          P^T = exp(ST*scale - lse[col])
+//       dV += P^T dO                   (P^T restaged via LDS into A-frag)
+//       S(32q x 16k half-wise), dP = dO V^T, dS = P (dP - delta[row])
+//       dK += dS^T Q                   (dS^T restaged via LDS)
+//   dq kernel: one wave per 16-row q-tile; loop over k tiles:
+//       S, P, dP = dO V^T, dS = P (dP - delta); dQ += dS K.
+
+__global__ void fa_delta_kernel(const bf16* __restrict__ o,
+                                const bf16* __restrict__ dout,
+                                float* __restrict__ delta, int64_t rows) {
+  // delta[r] = sum_d o[r][d] * dout[r][d]; one wave per row, D=128.
+  const int lane = threadIdx.x % 64;
+  const int wave = threadIdx.x / 64;
+  const int64_t r = static_cast<int64_t>(blockIdx.x) * kWavesFA + wave;
+  if (r >= rows) return;
+  float acc = 0.f;
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    int d = lane + 64 * j;
+    acc += __bfloat162float(o[r * kD + d]) * __bfloat162float(dout[r * kD + d]);
+  }
+  for (int off = 32; off > 0; off >>= 1)
+    acc += __shfl_xor(acc, off, 64);
+  if (lane == 0) delta[r] = acc;
+}
+
+// Shared helper: S(16q x 16k) C-frag from Q rows [q0+ ] and K rows [kk0+ ].
+__device__ __forceinline__ frag_cd fa_score_tile(
+    const bf16* __restrict__ q, const bf16* __restrict__ k, int64_t qbase,
+    int64_t kbase, int q0, int kk0, int S, int lane) {
+  const int row = lane & 15, grp = lane >> 4;
+  frag_cd acc = frag_cd{0.f, 0.f, 0.f, 0.f};
+  const int qr = q0 + row < S ? q0 + row : S - 1;
+  const int kr = kk0 + row < S ? kk0 + row : S - 1;
+#pragma unroll
+  for (int c = 0; c < kD / 32; ++c) {
+    frag_ab qf = *reinterpret_cast<const frag_ab*>(
+        q + (qbase + qr) * kD + c * 32 + grp * 8);
+    frag_ab kf = *reinterpret_cast<const frag_ab*>(
+        k + (kbase + kr) * kD + c * 32 + grp * 8);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf, kf, acc, 0, 0, 0);
+  }
+  return acc;
+}
+
+__global__ void fa_bwd_dkdv_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ k,
+    const bf16* __restrict__ v, const bf16* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    bf16* __restrict__ dk, bf16* __restrict__ dv, int B, int HQ, int HK,
+    int S, float scale, int causal) {
+  const int lane = threadIdx.x % 64;
+  const int wave = threadIdx.x / 64;
+  const int kt = blockIdx.x * kWavesFA + wave;  // 16-row k tile
+  const int hkv = blockIdx.y;
+  const int b = blockIdx.z;
+  const int kt0 = kt * kQTile;
+  if (kt0 >= S) return;
+  const int row = lane & 15, grp = lane >> 4;
+  const int gqa = HQ / HK;
+
+  __shared__ bf16 stage_lds[kWavesFA][kQTile][kKTile];  // P^T / dS^T restage
+
+  const int64_t kbase = (static_cast<int64_t>(b) * HK + hkv) * S;
+
+  frag_cd dkacc[kD / 16], dvacc[kD / 16];
+#pragma unroll
+  for (int cc = 0; cc < kD / 16; ++cc) {
+    dkacc[cc] = frag_cd{0.f, 0.f, 0.f, 0.f};
+    dvacc[cc] = frag_cd{0.f, 0.f, 0.f, 0.f};
+  }
+
+  for (int g = 0; g < gqa; ++g) {
+    const int h = hkv * gqa + g;
+    const int64_t qbase = (static_cast<int64_t>(b) * HQ + h) * S;
+    const float* lse_h = lse + (static_cast<int64_t>(b) * HQ + h) * S;
+    const float* dlt_h = delta + (static_cast<int64_t>(b) * HQ + h) * S;
+
+    const int q_start = causal ? (kt0 / kKTile) * kKTile : 0;
+    for (int q0 = q_start; q0 < S; q0 += kKTile) {
+      // ---- P^T (16k x 32q) for dV: ST = K Q^T, exp(ST*scale - lse[q]).
+      float pT[2][4];
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        frag_cd st = fa_score_tile(k, q, kbase, qbase, kt0, q0 + half * 16,
+                                   S, lane);
+        // ST C-frag: row = k index (grp*4+r), col = q index (lane&15)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int ki = kt0 + grp * 4 + r;
+          int qi = q0 + half * 16 + row;
+          float sval = st[r] * scale;
+          bool dead = ki >= S || qi >= S || (causal && ki > qi);
+          pT[half][r] = dead ? 0.f : __expf(sval - lse_h[qi < S ? qi : 0]);
+        }
+      }
+      // restage P^T into A-frag layout (rows = k, kdim = q, 32 wide)
+#pragma unroll
+      for (int half = 0; half < 2; ++half)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          stage_lds[wave][grp * 4 + r][half * 16 + row] =
+              __float2bfloat16(pT[half][r]);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      frag_ab ptf;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        reinterpret_cast<bf16*>(&ptf)[j] = stage_lds[wave][row][grp * 8 + j];
+
+      // dV += P^T dO : B-frag of dO chunk cc -> strided global reads
+      //   dO[q0 + grp*8 + j][cc*16 + row]
+#pragma unroll
+      for (int cc = 0; cc < kD / 16; ++cc) {
+        frag_ab dof;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int qi = q0 + grp * 8 + j;
+          reinterpret_cast<bf16*>(&dof)[j] =
+              qi < S ? dout[(qbase + qi) * kD + cc * 16 + row]
+                     : __float2bfloat16(0.f);
+        }
+        dvacc[cc] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(ptf, dof, dvacc[cc], 0, 0, 0);
+      }
+
+      // ---- dS (32q x 16k), computed half-wise in (q,k) orientation.
+      float dsv[2][4];
+#pragma unroll
+      for (int qh = 0; qh < 2; ++qh) {
+        // S half: rows q, cols k
+        frag_cd sq = fa_score_tile(q, k, qbase, kbase, q0 + qh * 16, kt0,
+                                   S, lane);
+        // dP half: dO(16q x 128) V^T -> A = dO chunks, B = V rows
+        frag_cd dp = frag_cd{0.f, 0.f, 0.f, 0.f};
+        const int qr = q0 + qh * 16 + row < S ? q0 + qh * 16 + row : S - 1;
+        const int vr = kt0 + row < S ? kt0 + row : S - 1;
+#pragma unroll
+        for (int c = 0; c < kD / 32; ++c) {
+          frag_ab dof = *reinterpret_cast<const frag_ab*>(
+              dout + (qbase + qr) * kD + c * 32 + grp * 8);
+          frag_ab vf = *reinterpret_cast<const frag_ab*>(
+              v + (kbase + vr) * kD + c * 32 + grp * 8);
+          dp = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dof, vf, dp, 0, 0, 0);
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int qi = q0 + qh * 16 + grp * 4 + r;
+          int ki = kt0 + row;
+          bool dead = ki >= S || qi >= S || (causal && ki > qi);
+          float pv = dead ? 0.f
+                          : __expf(sq[r] * scale - lse_h[qi < S ? qi : 0]);
+          dsv[qh][r] = pv * (dp[r] - dlt_h[qi < S ? qi : 0]) * scale;
+        }
+      }
+      // restage dS^T: A-frag rows = k (lane&15), kdim = q
+#pragma unroll
+      for (int qh = 0; qh < 2; ++qh)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          stage_lds[wave][row][qh * 16 + grp * 4 + r] =
+              __float2bfloat16(dsv[qh][r]);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      frag_ab dstf;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        reinterpret_cast<bf16*>(&dstf)[j] = stage_lds[wave][row][grp * 8 + j];
+
+      // dK += dS^T Q : B-frag = Q[q0 + grp*8 + j][cc*16 + row]
+#pragma unroll
+      for (int cc = 0; cc < kD / 16; ++cc) {
+        frag_ab qf2;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int qi = q0 + grp * 8 + j;
+          reinterpret_cast<bf16*>(&qf2)[j] =
+              qi < S ? q[(qbase + qi) * kD + cc * 16 + row]
+                     : __float2bfloat16(0.f);
+        }
+        dkacc[cc] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(dstf, qf2, dkacc[cc], 0, 0, 0);
+      }
+    }
+  }
+
+  // store dK/dV (C layout: col = lane&15, row = grp*4 + r)
+#pragma unroll
+  for (int cc = 0; cc < kD / 16; ++cc)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int ki = kt0 + grp * 4 + r;
+      if (ki < S) {
+        dk[(kbase + ki) * kD + cc * 16 + row] = __float2bfloat16(dkacc[cc][r]);
+        dv[(kbase + ki) * kD + cc * 16 + row] = __float2bfloat16(dvacc[cc][r]);
+      }
+    }
+}
+
+__global__ void fa_bwd_dq_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ k,
+    const bf16* __restrict__ v, const bf16* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    bf16* __restrict__ dq, int B, int HQ, int HK, int S, float scale,
+    int causal) {
+  const int lane = threadIdx.x % 64;
+  const int wave = threadIdx.x / 64;
+  const int qt = blockIdx.x * kWavesFA + wave;
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int hk = h / (HQ / HK);
+  const int q0 = qt * kQTile;
+  if (q0 >= S) return;
+  const int row = lane & 15, grp = lane >> 4;
+
+  __shared__ bf16 ds_lds[kWavesFA][kQTile][kKTile];
+
+  const int64_t qbase = (static_cast<int64_t>(b) * HQ + h) * S;
+  const int64_t kbase = (static_cast<int64_t>(b) * HK + hk) * S;
+  const float* lse_h = lse + (static_cast<int64_t>(b) * HQ + h) * S;
+  const float* dlt_h = delta + (static_cast<int64_t>(b) * HQ + h) * S;
+
+  frag_cd dqacc[kD / 16];
+#pragma unroll
+  for (int cc = 0; cc < kD / 16; ++cc) dqacc[cc] = frag_cd{0.f, 0.f, 0.f, 0.f};
+
+  const int kend = causal ? (q0 + kQTile < S ? q0 + kQTile : S) : S;
+  for (int k0 = 0; k0 < kend; k0 += kKTile) {
+    // dS (16q x 32k) half-wise in (q,k) orientation.
+    float dsv[2][4];
+#pragma unroll
+    for (int kh = 0; kh < 2; ++kh) {
+      frag_cd sq = fa_score_tile(q, k, qbase, kbase, q0, k0 + kh * 16, S, lane);
+      frag_cd dp = frag_cd{0.f, 0.f, 0.f, 0.f};
+      const int qr = q0 + row < S ? q0 + row : S - 1;
+      const int vr = k0 + kh * 16 + row < S ? k0 + kh * 16 + row : S - 1;
+#pragma unroll
+      for (int c = 0; c < kD / 32; ++c) {
+        frag_ab dof = *reinterpret_cast<const frag_ab*>(
+            dout + (qbase + qr) * kD + c * 32 + grp * 8);
+        frag_ab vf = *reinterpret_cast<const frag_ab*>(
+            v + (kbase + vr) * kD + c * 32 + grp * 8);
+        dp = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dof, vf, dp, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int qi = q0 + grp * 4 + r;
+        int ki = k0 + kh * 16 + row;
+        bool dead = ki >= S || qi >= S || (causal && ki > qi);
+        float pv = dead ? 0.f
+                        : __expf(sq[r] * scale - lse_h[qi < S ? qi : 0]);
+        dsv[kh][r] = pv * (dp[r] - dlt_h[qi < S ? qi : 0]) * scale;
+      }
+    }
+    // restage dS into A-frag layout (rows = q, kdim = k)
+#pragma unroll
+    for (int kh = 0; kh < 2; ++kh)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        ds_lds[wave][grp * 4 + r][kh * 16 + row] = __float2bfloat16(dsv[kh][r]);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    frag_ab dsf;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      reinterpret_cast<bf16*>(&dsf)[j] = ds_lds[wave][row][grp * 8 + j];
+
+    // dQ += dS K : B-frag = K[k0 + grp*8 + j][cc*16 + row]
+#pragma unroll
+    for (int cc = 0; cc < kD / 16; ++cc) {
+      frag_ab kf2;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int ki = k0 + grp * 8 + j;
+        reinterpret_cast<bf16*>(&kf2)[j] =
+            ki < S ? k[(kbase + ki) * kD + cc * 16 + row]
+                   : __float2bfloat16(0.f);
+      }
+      dqacc[cc] =
+          __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, kf2, dqacc[cc], 0, 0, 0);
+    }
+  }
+
+#pragma unroll
+  for (int cc = 0; cc < kD / 16; ++cc)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int qi = q0 + grp * 4 + r;
+      if (qi < S)
+        dq[(qbase + qi) * kD + cc * 16 + row] = __float2bfloat16(dqacc[cc][r]);
+    }
+}
+
 }  // namespace
+
+void launch_fa_bwd(const void* q, const void* k, const void* v,
+                   const void* dout, const void* o, const float* lse,
+                   float* delta, void* dq, void* dk, void* dv, int B, int HQ,
+                   int HK, int S, int D, bool causal, hipStream_t stream) {
+  if (D != kD) return;  // v0: head dim 128 only (wrapper validates)
+  float scale = 1.f / sqrtf(static_cast<float>(D));
+  int64_t rows = static_cast<int64_t>(B) * HQ * S;
+  hipLaunchKernelGGL(fa_delta_kernel,
+                     static_cast<unsigned>((rows + kWavesFA - 1) / kWavesFA),
+                     256, 0, stream, static_cast<const bf16*>(o),
+                     static_cast<const bf16*>(dout), delta, rows);
+  dim3 gkv((S + kQTile * kWavesFA - 1) / (kQTile * kWavesFA), HK, B);
+  hipLaunchKernelGGL(fa_bwd_dkdv_kernel, gkv, 256, 0, stream,
+                     static_cast<const bf16*>(q), static_cast<const bf16*>(k),
+                     static_cast<const bf16*>(v),
+                     static_cast<const bf16*>(dout), lse, delta,
+                     static_cast<bf16*>(dk), static_cast<bf16*>(dv), B, HQ,
+                     HK, S, scale, causal ? 1 : 0);
+  dim3 gq((S + kQTile * kWavesFA - 1) / (kQTile * kWavesFA), HQ, B);
+  hipLaunchKernelGGL(fa_bwd_dq_kernel, gq, 256, 0, stream,
+                     static_cast<const bf16*>(q), static_cast<const bf16*>(k),
+                     static_cast<const bf16*>(v),
+                     static_cast<const bf16*>(dout), lse, delta,
+                     static_cast<bf16*>(dq), B, HQ, HK, S, scale,
+                     causal ? 1 : 0);
+}
 
 void launch_fa_fwd(const void* q, const void* k, const void* v, void* o,
                    float* lse, int B, int HQ, int HK, int S, int D,

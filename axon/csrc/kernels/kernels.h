@@ -112,5 +112,10 @@ void launch_swiglu_bwd(int dtype, const void* dz, const void* a, const void* g,
 void launch_fa_fwd(const void* q, const void* k, const void* v, void* o,
                    float* lse, int B, int HQ, int HK, int S, int D,
                    bool causal, hipStream_t stream);
+// EXPERIMENTAL FA backward v0 (same layouts; delta fp32[B*HQ*S] scratch).
+void launch_fa_bwd(const void* q, const void* k, const void* v,
+                   const void* dout, const void* o, const float* lse,
+                   float* delta, void* dq, void* dk, void* dv, int B, int HQ,
+                   int HK, int S, int D, bool causal, hipStream_t stream);
 
 }  // namespace axon

@@ -83,6 +83,39 @@ def correctness(args):
     return 0
 
 
+def check_bwd(args):
+    """Round-2: validate _core.fa_bwd against autograd through the fp64
+    reference (needs a GPU)."""
+    assert torch.cuda.is_available(), "--check-bwd needs a GPU"
+    from axon import _core
+
+    device, dtype = "cuda", torch.bfloat16
+    tol = 5e-2
+    ok = True
+    for kind in ("random", "spiked"):
+        for s in (64, 192, args.seq):
+            q, k, v = make_case(kind, 2, args.heads, args.kv_heads, s,
+                                args.head_dim, device, dtype)
+            g = torch.randn(2, args.heads, s, args.head_dim,
+                            device=device).to(dtype)
+            q64 = q.double().requires_grad_(True)
+            k64 = k.double().requires_grad_(True)
+            v64 = v.double().requires_grad_(True)
+            ref_attention_fp64(q64, k64, v64).backward(g.double())
+            with torch.no_grad():
+                o, lse = _core.fa_fwd(q, k, v, True)
+                dq, dk, dv = _core.fa_bwd(q, k, v, o, g, lse, True)
+            for name, got, want in (("dq", dq, q64.grad), ("dk", dk, k64.grad),
+                                    ("dv", dv, v64.grad)):
+                err = (got.double() - want).abs().max().item()
+                status = "OK " if err < tol else "FAIL"
+                ok &= err < tol
+                print(f"  [{status}] {kind:7s} seq={s:5d} {name} "
+                      f"max|err|={err:.2e}")
+    print("bwd check", "passed" if ok else "FAILED")
+    return 0 if ok else 1
+
+
 def bench(args):
     assert torch.cuda.is_available(), "--bench needs a GPU"
     device, dtype = "cuda", torch.bfloat16
@@ -125,7 +158,10 @@ def main():
     p.add_argument("--head-dim", type=int, default=128)
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--bench", action="store_true")
+    p.add_argument("--check-bwd", action="store_true")
     args = p.parse_args()
+    if args.check_bwd:
+        sys.exit(check_bwd(args))
     sys.exit(bench(args) if args.bench else correctness(args))
 
 
