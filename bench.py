@@ -103,7 +103,9 @@ def build_bert(args, device, dtype):
 
     _load_tunableop("bert")
     torch.manual_seed(1234)
-    model = BertForPreTraining(bert_large_config()).to(device=device, dtype=dtype)
+    with torch.device(device):
+        model = BertForPreTraining(bert_large_config())
+    model = model.to(dtype=dtype)
     batch = args.batch or 64
     seq = args.seq_len or 128
     if args.no_fused_optimizer or device.type == "cpu":
@@ -145,7 +147,12 @@ def build_llama(args, device, dtype, which="llama3-8b"):
             print(f"[bench] AXON_FA_LIB={fa} unavailable: {e}", flush=True)
     torch.manual_seed(1234)
     cfgm = llama3_8b_config() if which == "llama3-8b" else llama3_1b_config()
-    model = LlamaForCausalLM(cfgm).to(device=device, dtype=dtype)
+    # Construct directly on the device: a CPU-first build of 8B params
+    # costs 32 GB host RAM and minutes of single-threaded init PER RANK
+    # (x8 on the scaling run); on-device init is seconds.
+    with torch.device(device):
+        model = LlamaForCausalLM(cfgm)
+    model = model.to(dtype=dtype)
     batch = args.batch or (8 if which == "llama3-8b" else 8)
     seq = args.seq_len or 4096
     if args.no_fused_optimizer or device.type == "cpu":
