@@ -30,7 +30,9 @@ def main():
     device = "cuda" if torch.cuda.is_available() else "cpu"
     dtype = torch.bfloat16 if device == "cuda" else torch.float32
     cfg = llama3_1b_config() if args.small else llama3_8b_config()
-    model = LlamaForCausalLM(cfg).to(device, dtype)
+    with torch.device(device):  # on-device init (8B fp32 on host is slow)
+        model = LlamaForCausalLM(cfg)
+    model = model.to(dtype=dtype)
     opt = FusedAdamW(model.parameters(), lr=3e-4)
     opt = hvd.DistributedOptimizer(opt, named_parameters=model.named_parameters())
     hvd.broadcast_parameters(model.state_dict(), root_rank=0)
