@@ -196,8 +196,8 @@ std::vector<Response> Controller::run_cycle(const std::vector<Request>& own,
 
   // A process set registered after its members' requests arrived: re-check
   // readiness of everything deferred on an unknown set.
-  if (ps_count() != last_ps_count_) {
-    last_ps_count_ = ps_count();
+  if (ps_version() != last_ps_version_) {
+    last_ps_version_ = ps_version();
     rescan_ready();
   }
   std::vector<Response> rs = assemble();
@@ -255,7 +255,15 @@ void Controller::ingest(const Request& q) {
 void Controller::maybe_ready(const std::string& name, PendingTensor& pt) {
   if (pt.queued) return;
   ProcessSetInfo psi = ps_get(pt.pset);
-  if (!psi.valid) return;  // set not registered here yet; re-checked per cycle
+  if (!psi.valid) {
+    // id beyond the registry: may be registered later this cycle — defer.
+    if (pt.pset >= ps_count()) return;
+    // Explicitly removed set: emit make_response's error now instead of
+    // hanging until the stall warning.
+    pt.queued = true;
+    ready_.push_back(name);
+    return;
+  }
   uint32_t need = psi.mask;
   if ((pt.have | (joined_mask_ & need)) == need) {
     pt.queued = true;

@@ -114,7 +114,7 @@ class Controller {
   uint32_t joined_mask_ = 0;
   int last_joined_ = -1;
   int order_counter_ = 0;
-  int last_ps_count_ = 1;  // re-scan readiness when new sets register
+  int last_ps_version_ = 1;  // re-scan readiness on set register/remove
   int64_t fusion_threshold_;
   double stall_warn_sec_;
   bool pending_shutdown_ = false;

@@ -262,7 +262,10 @@ Status CpuBackend::allgather(const Response& r, std::vector<TensorEntry>& entrie
         if (peer != 0) send_tensor(peer, full);
     } else {
       auto& e = entries[ti];
-      send_tensor(0, flat(e.in));
+      // Symmetric with the hub's fd[i]==0 skip: a zero-row contribution
+      // sends nothing (the hub never posts a recv for it).
+      if (psi.my_index < 0 || fd.at(psi.my_index) != 0)
+        send_tensor(0, flat(e.in));
       auto in_sizes = e.in.sizes().vec();
       int64_t row = 1;
       for (size_t d = 1; d < in_sizes.size(); ++d) row *= in_sizes[d];

@@ -53,6 +53,7 @@ struct Global {
   // registration), read by the background thread each cycle.
   std::mutex ps_mu;
   std::vector<ProcessSetInfo> psets;
+  int ps_version = 1;  // bumped on register AND remove (controller rescans)
 };
 
 Global g;
@@ -117,6 +118,18 @@ void execute_response(const Response& r) {
 
   ProcessSetInfo psi = ps_get(r.process_set);
   if (!psi.valid) {
+    if (!r.error.empty()) {
+      // Error response for a removed set: deliver it to whichever of the
+      // named tensors this rank actually submitted; others never will be.
+      for (const auto& name : r.names) {
+        auto it = g.pending.find(name);
+        if (it != g.pending.end()) {
+          complete(it->second, Status::Error(r.error));
+          g.pending.erase(it);
+        }
+      }
+      return;
+    }
     AXON_LOG(LogLevel::ERROR, "axon: response for unknown process set " +
                                   std::to_string(r.process_set));
     return;
@@ -317,6 +330,10 @@ void bg_loop(std::promise<void> ready) {
 void initialize(int rank_, int size_, int local_rank_, const std::string& path) {
   if (g.initialized.load())
     throw std::runtime_error("axon already initialized");
+  if (size_ < 1 || size_ > 32)
+    throw std::runtime_error(
+        "axon: world size " + std::to_string(size_) +
+        " unsupported — rank masks are 32-bit (single-node target; max 32)");
   g.rank = rank_;
   g.size = size_;
   g.local_rank = local_rank_;
@@ -390,6 +407,7 @@ int ps_register(const std::vector<int>& ranks) {
   std::lock_guard<std::mutex> lk(g.ps_mu);
   psi.id = static_cast<int>(g.psets.size());
   g.psets.push_back(psi);
+  ++g.ps_version;
   return psi.id;
 }
 
@@ -414,11 +432,17 @@ void ps_remove(int id) {
   if (id <= 0 || id >= static_cast<int>(g.psets.size()))
     throw std::runtime_error("axon: cannot remove process set " + std::to_string(id));
   g.psets[id].valid = false;
+  ++g.ps_version;
 }
 
 int ps_count() {
   std::lock_guard<std::mutex> lk(g.ps_mu);
   return static_cast<int>(g.psets.size());
+}
+
+int ps_version() {
+  std::lock_guard<std::mutex> lk(g.ps_mu);
+  return g.ps_version;
 }
 
 void enqueue(TensorEntry entry) {

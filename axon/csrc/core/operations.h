@@ -40,6 +40,7 @@ ProcessSetInfo ps_get(int id);  // copy; unknown id -> invalid info (valid=false
 bool ps_known(int id);
 void ps_remove(int id);  // marks invalid; ops on it then error
 int ps_count();
+int ps_version();  // bumped on register and remove; controller rescan key
 
 // GPU backend factory seam (implemented in rccl_backend.cpp).
 class ControlPlane;

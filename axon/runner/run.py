@@ -201,7 +201,11 @@ def run_elastic(args):
                         respawn.append(w)
                     else:
                         evict.append(w)
-                members = [w for w in members if w not in evict]
+                # Drop evicted workers AND workers that already finished
+                # cleanly: a new generation must not list a rank that will
+                # never re-join (survivors would hang waiting for it).
+                members = [w for w in members
+                           if w not in evict and w not in done_ok]
                 if len(members) < args.min_np:
                     print(f"[axonrun] world would shrink below --min-np "
                           f"{args.min_np} — aborting", file=sys.stderr)

@@ -208,9 +208,31 @@ def _allgather_variable(rank, size):
         off += r + 1
 
 
+def _allgather_zero_rows(rank, size):
+    import axon as hvd
+
+    # Rank 1 contributes zero rows — legal; must not leave a stale frame
+    # queued on its channel (regression: ADVICE.md r1 medium finding).
+    rows = 0 if rank == 1 else rank + 1
+    t = torch.full((rows, 2), float(rank))
+    out = hvd.allgather(t, name="zr.ag")
+    expect_rows = sum(0 if r == 1 else r + 1 for r in range(size))
+    assert out.shape == (expect_rows, 2)
+    off = 0
+    for r in range(size):
+        n = 0 if r == 1 else r + 1
+        assert torch.allclose(out[off:off + n], torch.full((n, 2), float(r)))
+        off += n
+    # A following collective must see a clean channel.
+    s = hvd.allreduce(torch.full((4,), float(rank)), op=hvd.Sum, name="zr.after")
+    assert torch.allclose(s, torch.full((4,), float(sum(range(size)))))
+
+
 def test_allgather():
     run_distributed(_allgather_equal, 2)
     run_distributed(_allgather_variable, 3)
+    run_distributed(_allgather_zero_rows, 2)
+    run_distributed(_allgather_zero_rows, 3)
 
 
 # -------------------------------------------------------------- alltoall

@@ -151,3 +151,25 @@ def _subset_errors(rank, size):
 
 def test_process_set_errors():
     run_distributed(_subset_errors, 2)
+
+
+def _removed_set_op_errors(rank, size):
+    import axon as hvd
+
+    ps = hvd.add_process_set(hvd.ProcessSet(list(range(size))))
+    pid = ps.process_set_id
+    hvd.remove_process_set(ps)
+    # Ops submitted with the raw (now removed) id must error promptly, not
+    # deadlock until the stall warning (regression: ADVICE.md r1 finding).
+    try:
+        hvd.allreduce(torch.ones(3), name="rm.ar", process_set=pid)
+        raise AssertionError("op on removed set must raise")
+    except AssertionError:
+        raise
+    except Exception as e:  # noqa: BLE001 - surface is a core RuntimeError
+        assert "removed" in str(e), str(e)
+    hvd.barrier(name="rm.done")
+
+
+def test_removed_set_op_errors():
+    run_distributed(_removed_set_op_errors, 2)
