@@ -166,8 +166,6 @@ Status CpuBackend::adasum(const Response& r, std::vector<TensorEntry>& entries) 
   ProcessSetInfo psi = ps_get(r.process_set);
   const bool im_member = psi.member(rank_);
   const int nmem = psi.size();
-  if ((nmem & (nmem - 1)) != 0)
-    return Status::Error("adasum requires a power-of-two process-set size");
   at::ScalarType dtype = id_dtype(r.dtype);
   const size_t ntensors = im_member ? entries.size() : r.names.size();
   for (size_t ti = 0; ti < ntensors; ++ti) {
@@ -185,8 +183,16 @@ Status CpuBackend::adasum(const Response& r, std::vector<TensorEntry>& entries) 
         v[i] = peer == 0 ? t.to(at::kDouble)
                          : recv_tensor_any(peer, dtype).to(at::kDouble);
       }
-      for (int stride = 1; stride < nmem; stride *= 2)
-        for (int i = 0; i + stride < nmem; i += 2 * stride)
+      // Non-power-of-two: fold extras (i >= p2) into partner i - p2 with a
+      // full-vector pair combine FIRST, then run the power-of-two tree.
+      // This is the semantic the distributed GPU VHDD implements, so the
+      // oracle must use the same association order.
+      int p2 = 1;
+      while (p2 * 2 <= nmem) p2 *= 2;
+      for (int i = 0; i + p2 < nmem; ++i)
+        v[i] = adasum_combine_cpu(v[i], v[i + p2]);
+      for (int stride = 1; stride < p2; stride *= 2)
+        for (int i = 0; i + stride < p2; i += 2 * stride)
           v[i] = adasum_combine_cpu(v[i], v[i + stride]);
       result = v[0].to(dtype);
       for (int peer : psi.ranks)

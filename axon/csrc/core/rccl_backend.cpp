@@ -153,6 +153,9 @@ class RcclBackend : public GpuBackend {
     hip_check(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking),
               "hipStreamCreate(comm)");
     fuse_bytes_ = static_cast<size_t>(env_int("AXON_FUSION_THRESHOLD", 128 << 20));
+    flat_allreduce_ = env_int("AXON_FLAT_ALLREDUCE", 0) != 0;
+    flat_min_bytes_ = static_cast<size_t>(
+        env_int("AXON_FLAT_MIN_BYTES", 4 << 20));
     hip_check(hipMalloc(&fusebuf_, fuse_bytes_ + kAlign), "hipMalloc(fusion buffer)");
     hip_check(hipMalloc(&dots_mine_, 3 * sizeof(double)), "hipMalloc(dots)");
     hip_check(hipMalloc(&dots_peer_, 3 * sizeof(double)), "hipMalloc(dots)");
@@ -287,6 +290,39 @@ class RcclBackend : public GpuBackend {
     }
   }
 
+  // One all-reduced buffer. Default: a single ncclAllReduce (RCCL stripes
+  // it across xGMI links/channels itself). AXON_FLAT_ALLREDUCE=1 switches
+  // large buffers to the explicit flat reduce-scatter + all-gather
+  // composition: on a fully-connected 8-GPU xGMI node
+  // each phase moves S*(N-1)/N bytes as direct per-peer chunks over the 7
+  // p2p links — the fallback if RCCL's channel striping under-saturates.
+  void allreduce_buf(void* buf, int64_t elems, ncclDataType_t dt, size_t esize,
+                     ncclRedOp_t op, SetComm& sc) {
+    if (flat_allreduce_ && sc.size > 1 &&
+        static_cast<size_t>(elems) * esize >= flat_min_bytes_) {
+      char* b = static_cast<char*>(buf);
+      int64_t chunk = elems / sc.size;
+      if (chunk > 0) {
+        // Documented NCCL in-place forms: RS recv = send + rank*chunk,
+        // AG send = recv + rank*chunk.
+        nccl_check(ncclReduceScatter(b, b + sc.rank * chunk * esize, chunk, dt,
+                                     op, sc.comm, stream_),
+                   "ncclReduceScatter(flat)");
+        nccl_check(ncclAllGather(b + sc.rank * chunk * esize, b, chunk, dt,
+                                 sc.comm, stream_),
+                   "ncclAllGather(flat)");
+      }
+      int64_t tail_off = chunk * sc.size;
+      if (tail_off < elems)
+        nccl_check(ncclAllReduce(b + tail_off * esize, b + tail_off * esize,
+                                 elems - tail_off, dt, op, sc.comm, stream_),
+                   "ncclAllReduce(flat tail)");
+      return;
+    }
+    nccl_check(ncclAllReduce(buf, buf, elems, dt, op, sc.comm, stream_),
+               "ncclAllReduce");
+  }
+
   void allreduce(const Response& r, std::vector<TensorEntry>& entries, SetComm& sc) {
     at::ScalarType dtype = id_dtype(r.dtype);
     at::ScalarType wire = r.wire_dtype ? id_dtype(r.wire_dtype) : dtype;
@@ -296,12 +332,22 @@ class RcclBackend : public GpuBackend {
     ncclRedOp_t op = nccl_red(r.red);
 
     if (entries.size() == 1 && wire == dtype && !any_scale) {
-      // Zero-copy path: single tensor, no transform — RCCL stripes the
-      // buffer across xGMI links itself.
+      // Zero-copy path: single tensor, no transform.
       auto& e = entries[0];
-      nccl_check(ncclAllReduce(e.in.data_ptr(), e.out.data_ptr(), e.in.numel(),
-                               nccl_dtype(dtype), op, sc.comm, stream_),
-                 "ncclAllReduce");
+      const size_t ebytes =
+          static_cast<size_t>(e.in.numel()) * e.in.element_size();
+      if (flat_allreduce_ && sc.size > 1 && ebytes >= flat_min_bytes_) {
+        if (e.out.data_ptr() != e.in.data_ptr())
+          hip_check(hipMemcpyAsync(e.out.data_ptr(), e.in.data_ptr(), ebytes,
+                                   hipMemcpyDeviceToDevice, stream_),
+                    "hipMemcpyAsync(flat in->out)");
+        allreduce_buf(e.out.data_ptr(), e.in.numel(), nccl_dtype(dtype),
+                      e.in.element_size(), op, sc);
+      } else {
+        nccl_check(ncclAllReduce(e.in.data_ptr(), e.out.data_ptr(), e.in.numel(),
+                                 nccl_dtype(dtype), op, sc.comm, stream_),
+                   "ncclAllReduce");
+      }
       return;
     }
 
@@ -346,9 +392,7 @@ class RcclBackend : public GpuBackend {
     }
 
     int64_t wire_elems = static_cast<int64_t>(total / wsize);
-    nccl_check(ncclAllReduce(buf, buf, wire_elems, nccl_dtype(wire), op, sc.comm,
-                             stream_),
-               "ncclAllReduce(fused)");
+    allreduce_buf(buf, wire_elems, nccl_dtype(wire), wsize, op, sc);
 
     std::map<double, std::vector<ChunkDesc>> unpacks;
     for (size_t i = 0; i < entries.size(); ++i) {
@@ -508,18 +552,23 @@ class RcclBackend : public GpuBackend {
                                      base * row, dt, op, sc.comm, stream_),
                    "ncclReduceScatter");
       } else {
-        // Uneven split: allreduce into scratch, slice locally. Correct but
-        // 2x traffic — documented limitation; even splits take the fast path.
+        // Uneven split: one rooted ncclReduce per member inside a single
+        // group — every slice is reduced straight to its owner, so total
+        // traffic matches a true reduce-scatter (the old allreduce+slice
+        // fallback moved 2x).
         size_t esize = e.in.element_size();
-        char* buf = static_cast<char*>(ensure_scratch(e.in.numel() * esize));
-        nccl_check(ncclAllReduce(e.in.data_ptr(), buf, e.in.numel(), dt, op,
-                                 sc.comm, stream_),
-                   "ncclAllReduce(rs-uneven)");
-        int64_t off = sc.rank * base + std::min<int64_t>(sc.rank, rem);
-        hip_check(hipMemcpyAsync(e.out.data_ptr(), buf + off * row * esize,
-                                 my_rows * row * esize, hipMemcpyDeviceToDevice,
-                                 stream_),
-                  "hipMemcpyAsync(rs slice)");
+        const char* ib = static_cast<const char*>(e.in.data_ptr());
+        nccl_check(ncclGroupStart(), "ncclGroupStart");
+        for (int p = 0; p < sc.size; ++p) {
+          int64_t p_rows = base + (p < rem ? 1 : 0);
+          if (p_rows == 0) continue;
+          int64_t off = p * base + std::min<int64_t>(p, rem);
+          nccl_check(ncclReduce(ib + off * row * esize,
+                                p == sc.rank ? e.out.data_ptr() : nullptr,
+                                p_rows * row, dt, op, p, sc.comm, stream_),
+                     "ncclReduce(rs-uneven)");
+        }
+        nccl_check(ncclGroupEnd(), "ncclGroupEnd");
       }
     }
   }
@@ -529,8 +578,13 @@ class RcclBackend : public GpuBackend {
   // butterfly of 24-byte p2p exchanges (SURVEY.md §2.1 "Adasum ops"; math
   // parity with cpu_backend.cpp's fp64 oracle).
   void adasum(std::vector<TensorEntry>& entries, SetComm& sc) {
-    if ((sc.size & (sc.size - 1)) != 0)
-      throw std::runtime_error("adasum requires a power-of-two process-set size");
+    // Non-power-of-two sets: extras (rank >= p2) fold their full vector
+    // into partner rank - p2 (one pair combine), the p2-rank VHDD runs on
+    // the low ranks, and the result fans back out. Association order
+    // matches the CPU fp64 oracle (cpu_backend.cpp adasum).
+    int p2 = 1;
+    while (p2 * 2 <= sc.size) p2 *= 2;
+    const int extras = sc.size - p2;
     for (auto& e : entries) {
       int dtype = static_cast<int>(e.in.scalar_type());
       size_t esize = e.in.element_size();
@@ -547,12 +601,45 @@ class RcclBackend : public GpuBackend {
           launch_xform(d, nn, dtype, dtype, static_cast<float>(e.prescale), stream_);
         });
       }
-      char* scratch =
-          static_cast<char*>(ensure_scratch(((n + 1) / 2 + kAlign) * esize));
       auto dt = nccl_dtype(e.in.scalar_type());
+      if (sc.rank >= p2) {
+        // Folded extra: hand the (prescaled) vector to the partner, get
+        // the final combined result back after the VHDD finishes.
+        const int partner = sc.rank - p2;
+        if (n > 0) {
+          nccl_check(ncclSend(b, n, dt, partner, sc.comm, stream_),
+                     "ncclSend(adasum fold)");
+          nccl_check(ncclRecv(b, n, dt, partner, sc.comm, stream_),
+                     "ncclRecv(adasum fold result)");
+        }
+        if (e.postscale != 1.0) {
+          std::vector<ChunkDesc> c;
+          add_chunks(c, b, b, n, esize, kChunkElems);
+          run_chunked(c, [&](const ChunkDesc* d, int nn) {
+            launch_xform(d, nn, dtype, dtype, static_cast<float>(e.postscale),
+                         stream_);
+          });
+        }
+        continue;
+      }
+      const size_t half_scr = (static_cast<size_t>((n + 1) / 2) + kAlign) * esize;
+      const size_t fold_scr =
+          sc.rank < extras ? (static_cast<size_t>(n) + kAlign) * esize : 0;
+      char* scratch =
+          static_cast<char*>(ensure_scratch(std::max(half_scr, fold_scr)));
+      if (sc.rank < extras && n > 0) {
+        // Pair combine with my folded extra: full-vector dots, no
+        // butterfly (only this rank keeps the result).
+        nccl_check(ncclRecv(scratch, n, dt, p2 + sc.rank, sc.comm, stream_),
+                   "ncclRecv(adasum fold)");
+        hip_check(hipMemsetAsync(dots_mine_, 0, 3 * sizeof(double), stream_),
+                  "hipMemset(dots)");
+        launch_adasum_dots(b, scratch, n, dtype, dots_mine_, true, stream_);
+        launch_adasum_apply(b, scratch, n, dtype, dots_mine_, true, stream_);
+      }
 
       int64_t off = 0, len = n;
-      for (int stride = 1; stride < sc.size; stride <<= 1) {
+      for (int stride = 1; stride < p2; stride <<= 1) {
         int peer = sc.rank ^ stride;
         bool a_side = (sc.rank & stride) == 0;
         int64_t len1 = (len + 1) / 2, len2 = len - len1;
@@ -599,7 +686,7 @@ class RcclBackend : public GpuBackend {
       // rank's offsets) via direct p2p.
       auto segment_of = [&](int rk) {
         int64_t o = 0, l = n;
-        for (int stride = 1; stride < sc.size; stride <<= 1) {
+        for (int stride = 1; stride < p2; stride <<= 1) {
           int64_t l1 = (l + 1) / 2;
           if ((rk & stride) == 0) l = l1;
           else { o += l1; l -= l1; }
@@ -607,7 +694,7 @@ class RcclBackend : public GpuBackend {
         return std::make_pair(o, l);
       };
       nccl_check(ncclGroupStart(), "ncclGroupStart");
-      for (int p = 0; p < sc.size; ++p) {
+      for (int p = 0; p < p2; ++p) {
         if (p == sc.rank) continue;
         if (len > 0)
           nccl_check(ncclSend(b + off * esize, len, dt, p, sc.comm, stream_),
@@ -618,6 +705,10 @@ class RcclBackend : public GpuBackend {
                      "ncclRecv(adasum gather)");
       }
       nccl_check(ncclGroupEnd(), "ncclGroupEnd");
+      // Fan the final vector back out to my folded extra, if any.
+      if (sc.rank < extras && n > 0)
+        nccl_check(ncclSend(b, n, dt, p2 + sc.rank, sc.comm, stream_),
+                   "ncclSend(adasum fold result)");
       if (e.postscale != 1.0) {
         std::vector<ChunkDesc> c;
         add_chunks(c, b, b, n, esize, kChunkElems);
@@ -675,6 +766,8 @@ class RcclBackend : public GpuBackend {
   hipStream_t stream_ = nullptr;
   void* fusebuf_ = nullptr;
   size_t fuse_bytes_ = 0;
+  bool flat_allreduce_ = false;
+  size_t flat_min_bytes_ = 4 << 20;
   void* scratch_ = nullptr;
   size_t scratch_bytes_ = 0;
   std::vector<void*> retire_;
