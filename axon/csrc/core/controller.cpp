@@ -113,7 +113,19 @@ std::string Controller::fingerprint(const Request& q) {
     fp += ':';
     fp += std::to_string(d);
   }
+  for (auto v : q.splits) {
+    fp += ',';
+    fp += std::to_string(v);
+  }
   return fp;
+}
+
+bool Controller::cacheable(OpType op) {
+  // Every data op negotiates through the cache (SURVEY.md §2.1 "Response
+  // cache"); JOIN/BARRIER are control-shaped and stay full-frame.
+  return op == OpType::ALLREDUCE || op == OpType::BROADCAST ||
+         op == OpType::ALLGATHER || op == OpType::REDUCESCATTER ||
+         op == OpType::ALLTOALL;
 }
 
 std::vector<Response> Controller::run_cycle(const std::vector<Request>& own,
@@ -128,7 +140,7 @@ std::vector<Response> Controller::run_cycle(const std::vector<Request>& own,
       std::vector<const Request*> full;
       std::vector<uint32_t> ids;
       for (auto& q : own) {
-        if (q.op == OpType::ALLREDUCE) {
+        if (cacheable(q.op)) {
           auto it = cache_.find(q.name);
           if (it != cache_.end() && cache_fp_[q.name] == fingerprint(q)) {
             ids.push_back(it->second);
@@ -180,13 +192,13 @@ std::vector<Response> Controller::run_cycle(const std::vector<Request>& own,
         for (uint32_t i = 0; i < nc; ++i) {
           uint32_t id = r.u32();
           auto it = templates_.find(id);
-          if (it == templates_.end()) {
+          if (it == templates_.end() || !it->second.count(peer)) {
             AXON_LOG(LogLevel::ERROR, "axon: unknown cache id " +
                                           std::to_string(id) + " from rank " +
                                           std::to_string(peer));
             continue;
           }
-          Request q = it->second;
+          Request q = it->second.at(peer);
           q.rank = peer;
           ingest(q);
         }
@@ -243,11 +255,11 @@ void Controller::ingest(const Request& q) {
   }
   pt.reqs[q.rank] = q;
   pt.have |= (1u << q.rank);
-  // Metadata change on a cached tensor: refresh the coordinator template
-  // so later bare-id submissions reconstruct the new request.
-  if (q.op == OpType::ALLREDUCE) {
+  // Metadata change on a cached tensor: refresh this rank's coordinator
+  // template so later bare-id submissions reconstruct the new request.
+  if (cacheable(q.op)) {
     auto it = cache_.find(q.name);
-    if (it != cache_.end()) templates_[it->second] = q;
+    if (it != cache_.end()) templates_[it->second][q.rank] = q;
   }
   maybe_ready(q.name, pt);
 }
@@ -376,8 +388,8 @@ Response Controller::make_response(PendingTensor& pt) {
   }
 
   r.numels.push_back(numel0);
-  // Assign (or reuse) a response-cache id for allreduce tensors.
-  if (t0->op == OpType::ALLREDUCE) {
+  // Assign (or reuse) a response-cache id for cacheable ops.
+  if (cacheable(t0->op)) {
     int32_t id = -1;
     auto it = cache_.find(t0->name);
     if (it != cache_.end()) {
@@ -385,7 +397,9 @@ Response Controller::make_response(PendingTensor& pt) {
     } else if (static_cast<int64_t>(cache_.size()) < cache_capacity_) {
       id = static_cast<int32_t>(next_cache_id_++);
       cache_[t0->name] = static_cast<uint32_t>(id);
-      templates_[static_cast<uint32_t>(id)] = *t0;
+      auto& per_rank = templates_[static_cast<uint32_t>(id)];
+      for (auto& q : pt.reqs)
+        if (!q.name.empty()) per_rank[q.rank] = q;
     }
     r.cache_ids.push_back(id);
   } else {

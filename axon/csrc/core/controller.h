@@ -102,10 +102,13 @@ class Controller {
   // rank learns the id from the response; subsequent submissions with an
   // unchanged fingerprint go on the wire as the id alone.
   static std::string fingerprint(const Request& q);
+  static bool cacheable(OpType op);
   std::unordered_map<std::string, uint32_t> cache_;       // name -> id
   std::unordered_map<std::string, std::string> cache_fp_; // name -> fp
   std::unordered_map<std::string, std::string> sent_fp_;  // awaiting id
-  std::unordered_map<uint32_t, Request> templates_;       // coordinator
+  // Coordinator: per-rank request templates by cache id. Per-rank because
+  // allgather first-dims and alltoall splits legally differ across ranks.
+  std::unordered_map<uint32_t, std::unordered_map<int, Request>> templates_;
   uint32_t next_cache_id_ = 0;
   int64_t cache_capacity_;
   // --- coordinator state ---

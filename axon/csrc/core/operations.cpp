@@ -54,6 +54,10 @@ struct Global {
   std::mutex ps_mu;
   std::vector<ProcessSetInfo> psets;
   int ps_version = 1;  // bumped on register AND remove (controller rescans)
+
+  // End-to-end autotune signal: optimizer steps completed on this rank
+  // (written by the user thread via pm_step_mark, drained by the bg loop).
+  std::atomic<int64_t> step_marks{0};
 };
 
 Global g;
@@ -287,6 +291,7 @@ void bg_loop(std::promise<void> ready) {
       // Autotuner (rank 0): score by GPU-allreduced bytes/sec, ship knob
       // updates with the next broadcast (SURVEY.md §2.1 parameter manager).
       if (g.pm && g.pm->enabled()) {
+        g.pm->record_steps(g.step_marks.exchange(0, std::memory_order_relaxed));
         int64_t thr = g.ctl->fusion_threshold();
         double cyc = g.cycle_ms;
         if (g.pm->maybe_update(&thr, &cyc)) {
@@ -438,6 +443,10 @@ void ps_remove(int id) {
 int ps_count() {
   std::lock_guard<std::mutex> lk(g.ps_mu);
   return static_cast<int>(g.psets.size());
+}
+
+void pm_step_mark() {
+  g.step_marks.fetch_add(1, std::memory_order_relaxed);
 }
 
 int ps_version() {

@@ -42,6 +42,10 @@ void ps_remove(int id);  // marks invalid; ops on it then error
 int ps_count();
 int ps_version();  // bumped on register and remove; controller rescan key
 
+// End-to-end autotune signal: called once per optimizer step (rank-local);
+// rank 0's marks feed the ParameterManager's steps/sec score.
+void pm_step_mark();
+
 // GPU backend factory seam (implemented in rccl_backend.cpp).
 class ControlPlane;
 class Timeline;

@@ -1,7 +1,10 @@
 // Online autotuner for the two engine knobs: fusion threshold (bytes) and
 // cycle time (ms) — capability parity with the reference's ParameterManager
 // (SURVEY.md §2.1 "Parameter manager (autotuner)"), re-designed as a
-// coordinator-driven grid hill-climb scored by allreduced bytes/sec:
+// coordinator-driven grid hill-climb scored end-to-end by optimizer
+// steps/sec (reported by DistributedOptimizer.step via autotune_step_mark;
+// falls back to allreduced bytes/sec for op-only workloads), followed by
+// refinement rounds around the best point:
 // xGMI-era defaults are already large (128 MiB), so the tuner explores a
 // coarse 2-D neighborhood instead of running full Bayesian optimization.
 //
@@ -23,6 +26,9 @@ class ParameterManager {
   bool enabled() const { return enabled_; }
   // Account bytes moved by GPU allreduce responses this cycle.
   void record_bytes(int64_t bytes);
+  // Account optimizer steps completed since the last cycle (end-to-end
+  // signal: steps/sec outranks allreduced-bytes/sec when present).
+  void record_steps(int64_t steps);
   // Called once per cycle on the coordinator. Returns true when the knobs
   // changed; outputs the new values (to apply locally + broadcast).
   bool maybe_update(int64_t* threshold, double* cycle_ms);
@@ -35,10 +41,16 @@ class ParameterManager {
   };
   void advance();
 
+  void refine();  // build a finer plan around the current best
+
   bool enabled_;
   bool log_;
   bool done_ = false;
   int64_t bytes_ = 0;
+  int64_t steps_ = 0;
+  bool using_steps_ = false;
+  int refine_rounds_ = 1;
+  int round_ = 0;
   std::chrono::steady_clock::time_point window_start_;
   double warmup_until_ = 3.0;  // seconds before sampling (AXON_AUTOTUNE_WARMUP_SEC)
   double sample_sec_;

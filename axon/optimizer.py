@@ -108,7 +108,12 @@ class _DistributedOptimizer:
     def step(self, closure=None):
         if self._require_sync and self._ps_size > 1:
             self.synchronize()
-        return super().step(closure)
+        ret = super().step(closure)
+        # End-to-end autotune signal: steps/sec on rank 0 scores the
+        # fusion-threshold/cycle-time sweep (no-op unless AXON_AUTOTUNE=1).
+        from axon import _core
+        _core.autotune_step_mark()
+        return ret
 
     def zero_grad(self, set_to_none=True):
         if self._handles:

@@ -162,7 +162,38 @@ def test_autotune_converges():
     r = _run(2, script, env={"AXON_AUTOTUNE": "1", "AXON_AUTOTUNE_LOG": "1",
                              "AXON_AUTOTUNE_SAMPLE_SEC": "0.05",
                              "AXON_AUTOTUNE_WARMUP_SEC": "0.1",
+                             "AXON_AUTOTUNE_REFINE": "0",
                              "AXON_LOG_LEVEL": "info"}, timeout=150)
     assert r.returncode == 0, r.stderr
     assert "autotune sample" in r.stderr
     assert "autotune done" in r.stderr
+
+
+def test_autotune_end_to_end_signal():
+    """With a DistributedOptimizer stepping, the tuner scores steps/sec
+    (end-to-end throughput) instead of allreduced bytes/sec, and runs a
+    refinement round around the grid's best point."""
+    script = (
+        "import torch, axon as hvd\n"
+        "hvd.init()\n"
+        "m = torch.nn.Linear(64, 64)\n"
+        "opt = hvd.DistributedOptimizer(torch.optim.SGD(m.parameters(),"
+        " lr=0.01), named_parameters=m.named_parameters())\n"
+        "hvd.broadcast_parameters(m.state_dict(), root_rank=0)\n"
+        "x = torch.randn(8, 64)\n"
+        "for i in range(900):\n"
+        "    opt.zero_grad()\n"
+        "    m(x).sum().backward()\n"
+        "    opt.step()\n"
+        "hvd.shutdown()\n")
+    r = _run(2, script, env={"AXON_AUTOTUNE": "1", "AXON_AUTOTUNE_LOG": "1",
+                             "AXON_AUTOTUNE_SAMPLE_SEC": "0.04",
+                             "AXON_AUTOTUNE_WARMUP_SEC": "0.1",
+                             "AXON_AUTOTUNE_REFINE": "1",
+                             "AXON_LOG_LEVEL": "info"}, timeout=150)
+    assert r.returncode == 0, r.stderr
+    assert "steps/s" in r.stderr, r.stderr[-2000:]
+    assert "autotune done" in r.stderr
+    # refinement happened: more samples than the 12-point base grid
+    assert r.stderr.count("autotune sample") > 12, \
+        r.stderr.count("autotune sample")

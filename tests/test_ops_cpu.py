@@ -475,6 +475,60 @@ def test_steady_state_response_cache():
     run_distributed(_steady_state_cache, 4)
 
 
+def _steady_state_cache_all_ops(rank, size):
+    # Round-2 extension: broadcast / allgather /
+    # reducescatter / alltoall negotiate through the response cache too.
+    # Correctness must hold across repeats (bare-id path), including an
+    # allgather whose per-rank first dims differ (per-rank templates) and
+    # a shape change mid-stream (fingerprint invalidation).
+    import axon as hvd
+
+    for it in range(5):
+        b = hvd.broadcast(torch.full((32,), float(rank * 10 + it)),
+                          root_rank=1, name="sc.bc")
+        assert torch.allclose(b, torch.full((32,), float(10 + it)))
+
+        g = hvd.allgather(torch.full((rank + 1, 2), float(rank + it)),
+                          name="sc.ag")
+        off = 0
+        for r in range(size):
+            assert torch.allclose(g[off:off + r + 1],
+                                  torch.full((r + 1, 2), float(r + it)))
+            off += r + 1
+
+        rows = size * 2
+        rs = hvd.reducescatter(torch.full((rows, 3), float(rank + 1 + it)),
+                               op=hvd.Sum, name="sc.rs")
+        expect = float(sum(r + 1 + it for r in range(size)))
+        assert rs.shape == (2, 3)
+        assert torch.allclose(rs, torch.full((2, 3), expect))
+
+        splits = [d + 1 for d in range(size)]
+        t = torch.cat([torch.full((d + 1, 2), float(rank * 10 + d + it))
+                       for d in range(size)])
+        out, _ = hvd.alltoall(t, splits=torch.tensor(splits), name="sc.a2a")
+        off = 0
+        for r in range(size):
+            assert torch.allclose(out[off:off + rank + 1],
+                                  torch.full((rank + 1, 2),
+                                             float(r * 10 + rank + it)))
+            off += rank + 1
+
+    # Shape change on a cached name: fingerprint mismatch must force a full
+    # re-negotiation and still produce the right answer.
+    g2 = hvd.allgather(torch.full((2 * rank + 1, 2), float(rank)), name="sc.ag")
+    off = 0
+    for r in range(size):
+        assert torch.allclose(g2[off:off + 2 * r + 1],
+                              torch.full((2 * r + 1, 2), float(r)))
+        off += 2 * r + 1
+
+
+def test_steady_state_cache_all_ops():
+    run_distributed(_steady_state_cache_all_ops, 2)
+    run_distributed(_steady_state_cache_all_ops, 3)
+
+
 def _ws8_smoke(rank, size):
     import axon as hvd
 
