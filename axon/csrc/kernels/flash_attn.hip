@@ -221,6 +221,218 @@ __global__ void fa_fwd_kernel(const bf16* __restrict__ q,
   }
 }
 
+// ------------------------------------------------------------- forward v2
+// Guide-structure forward (cdna_hip_programming.md T12/T13 + §5 anatomy):
+// 8 waves per block over 256 q rows; per wave a 32q x 32k tile via
+// mfma_f32_32x32x16_bf16 with SWAPPED QK^T (A=K, B=Q) so each lane owns
+// one q row's scores — softmax is per-lane + one permlane32_swap
+// cross-half merge, and P is repacked to the PV A-fragment entirely in
+// registers (cvt_pk pairs + 2 permlane32_swap per 16-key slice); K is
+// staged row-major in LDS, V transposed ([d][key]) so the PV B-fragment
+// is one ds_read_b128 of 8 consecutive keys; defer-max (T13, textbook
+// order: the rescale decision covers every P of its own tile) skips the
+// O-wide rescale unless a row max grows by > THR.
+//
+// Fragment maps (mfma_f32_32x32x16_bf16, guide §3):
+//   A: row = lane&31, k = (lane>>5)*8 + j          (8 bf16 / lane)
+//   B: col = lane&31, k = (lane>>5)*8 + j
+//   C/D: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+
+using frag16 = __attribute__((ext_vector_type(16))) float;  // 32x32 C/D
+
+constexpr int kBq2 = 256;   // q rows per block
+constexpr int kW2 = 8;      // waves per block
+constexpr int kBr2 = 32;    // q rows per wave
+constexpr int kBc2 = 32;    // keys per tile
+constexpr int kKPad = 8;    // bf16 pad per K-LDS row
+constexpr int kVPad = 8;    // bf16 pad per V-LDS row
+constexpr float kFaThr = 8.f;  // defer-max threshold (T13)
+
+__device__ __forceinline__ uint32_t pack_bf16(float lo, float hi) {
+  // compiler emits v_cvt_pk_bf16_f32 for this pattern (guide T12 note)
+  union { __hip_bfloat16 b; unsigned short u; } a{__float2bfloat16(lo)},
+      c{__float2bfloat16(hi)};
+  return static_cast<uint32_t>(a.u) | (static_cast<uint32_t>(c.u) << 16);
+}
+
+__device__ __forceinline__ float xhalf_swap_f(float x) {
+  // value held by lane l^32 (cross-half partner)
+  auto r = __builtin_amdgcn_permlane32_swap(__float_as_int(x),
+                                            __float_as_int(x), false, false);
+  return __int_as_float(threadIdx.x % 64 < 32 ? r[1] : r[0]);
+}
+
+__global__ __launch_bounds__(512, 1) void fa_fwd_v2_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ k,
+    const bf16* __restrict__ v, bf16* __restrict__ o, float* __restrict__ lse,
+    int B, int HQ, int HK, int S, float scale, int causal) {
+  const int lane = threadIdx.x % 64;
+  const int wave = threadIdx.x / 64;
+  const int idx = threadIdx.x;           // 0..511
+  const int col = lane & 31;             // q row (S^T C col) / d col (O)
+  const int half = lane >> 5;
+  const int qb = blockIdx.x * kBq2;
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int hk = h / (HQ / HK);
+  const int q0w = qb + wave * kBr2;
+
+  __shared__ bf16 k_lds[kBc2][kD + kKPad];        // row-major K tile
+  __shared__ bf16 v_lds[kD][kBc2 + kVPad];        // transposed V tile
+  __shared__ float red_lds[kW2][kBr2];            // per-row alpha/l exchange
+
+  const int64_t qbase = (static_cast<int64_t>(b) * HQ + h) * S;
+  const int64_t kbase = (static_cast<int64_t>(b) * HK + hk) * S;
+
+  // Q fragments, loaded once: B-frag col = q row (col), k-dim = d.
+  const int qi = q0w + col;
+  const int qrow = qi < S ? qi : S - 1;
+  frag_ab qf[kD / 16];
+#pragma unroll
+  for (int c = 0; c < kD / 16; ++c)
+    qf[c] = *reinterpret_cast<const frag_ab*>(
+        q + (qbase + qrow) * kD + c * 16 + half * 8);
+
+  frag16 oacc[kD / 32];
+#pragma unroll
+  for (int cc = 0; cc < kD / 32; ++cc)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) oacc[cc][r] = 0.f;
+  float m_run = -1e30f, l_run = 0.f;  // row stats for q row `qi` (raw scores)
+
+  // V staging map: vr = idx&31 (key), dc = (idx>>5)*8 (d chunk) — keeps the
+  // transposing ds_write_b16s spread across banks (see round-2 notes).
+  const int vr = idx & 31, vdc = (idx >> 5) * 8;
+  // K staging map: kr = idx>>4 (row), dc = (idx&15)*8 — coalesced + b128 write.
+  const int kr = idx >> 4, kdc = (idx & 15) * 8;
+
+  const int kend = causal ? (qb + kBq2 < S ? qb + kBq2 : S) : S;
+  const int ntiles = (kend + kBc2 - 1) / kBc2;
+
+  for (int t = 0; t < ntiles; ++t) {
+    const int k0 = t * kBc2;
+    // ---- stage tile t (all 8 waves cooperate) ----
+    {
+      float4 kv4 = float4{0.f, 0.f, 0.f, 0.f};
+      float4 vv4 = float4{0.f, 0.f, 0.f, 0.f};
+      if (k0 + kr < S)
+        kv4 = *reinterpret_cast<const float4*>(k + (kbase + k0 + kr) * kD + kdc);
+      if (k0 + vr < S)
+        vv4 = *reinterpret_cast<const float4*>(v + (kbase + k0 + vr) * kD + vdc);
+      __syncthreads();  // everyone done READING the previous tile
+      *reinterpret_cast<float4*>(&k_lds[kr][kdc]) = kv4;
+      const bf16* vb = reinterpret_cast<const bf16*>(&vv4);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v_lds[vdc + j][vr] = vb[j];
+      __syncthreads();
+    }
+
+    // ---- S^T = K Q^T (A = K rows, B = Q cols) ----
+    frag16 st;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) st[r] = 0.f;
+#pragma unroll
+    for (int c = 0; c < kD / 16; ++c) {
+      frag_ab kf = *reinterpret_cast<const frag_ab*>(
+          &k_lds[col][c * 16 + half * 8]);
+      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[c], st, 0, 0, 0);
+    }
+
+    // ---- masking + tile max (raw-score domain) ----
+    const bool full_tile = causal ? (k0 + kBc2 - 1 <= q0w) : (k0 + kBc2 <= S);
+    if (!full_tile) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int key = k0 + (r & 3) + 8 * (r >> 2) + 4 * half;
+        if (key >= S || (causal && key > qi)) st[r] = -1e30f;
+      }
+    }
+    float tmax = st[0];
+#pragma unroll
+    for (int r = 1; r < 16; ++r) tmax = fmaxf(tmax, st[r]);
+    tmax = fmaxf(tmax, xhalf_swap_f(tmax));
+
+    // ---- defer-max online softmax (textbook order) ----
+    float alpha = 1.f;
+    bool rescale =
+        !__all(scale * (tmax - m_run) <= kFaThr) || t == 0;
+    if (rescale) {
+      float m_new = fmaxf(m_run, tmax);
+      alpha = __expf(scale * (m_run - m_new));
+      m_run = m_new;
+      // O *= alpha: alpha is per-q-row; O holds 16 q rows per lane —
+      // redistribute via this wave's red_lds row (no block barrier:
+      // same-wave LDS handoff fenced by lgkmcnt).
+      red_lds[wave][col] = alpha;
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      float a16[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r)
+        a16[r] = red_lds[wave][(r & 3) + 8 * (r >> 2) + 4 * half];
+#pragma unroll
+      for (int cc = 0; cc < kD / 32; ++cc)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) oacc[cc][r] *= a16[r];
+    }
+    float p[16], rsum = 0.f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      p[r] = __expf(scale * (st[r] - m_run));
+      rsum += p[r];
+    }
+    l_run = l_run * alpha + rsum + xhalf_swap_f(rsum);
+
+    // ---- pack P to PV A-fragments (in registers, T12) ----
+    uint32_t pk[8];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) pk[i] = pack_bf16(p[2 * i], p[2 * i + 1]);
+    frag_ab pa1, pa2;
+    {
+      auto r0 = __builtin_amdgcn_permlane32_swap(pk[0], pk[2], false, false);
+      auto r1 = __builtin_amdgcn_permlane32_swap(pk[1], pk[3], false, false);
+      auto r2 = __builtin_amdgcn_permlane32_swap(pk[4], pk[6], false, false);
+      auto r3 = __builtin_amdgcn_permlane32_swap(pk[5], pk[7], false, false);
+      uint32_t* a1 = reinterpret_cast<uint32_t*>(&pa1);
+      uint32_t* a2 = reinterpret_cast<uint32_t*>(&pa2);
+      a1[0] = r0[0]; a1[1] = r1[0]; a1[2] = r0[1]; a1[3] = r1[1];
+      a2[0] = r2[0]; a2[1] = r3[0]; a2[2] = r2[1]; a2[3] = r3[1];
+    }
+
+    // ---- O += P V (B-frag: 8 consecutive keys from transposed V) ----
+#pragma unroll
+    for (int cc = 0; cc < kD / 32; ++cc) {
+      frag_ab v1 = *reinterpret_cast<const frag_ab*>(
+          &v_lds[cc * 32 + col][half * 8]);
+      frag_ab v2 = *reinterpret_cast<const frag_ab*>(
+          &v_lds[cc * 32 + col][16 + half * 8]);
+      oacc[cc] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa1, v1, oacc[cc],
+                                                         0, 0, 0);
+      oacc[cc] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa2, v2, oacc[cc],
+                                                         0, 0, 0);
+    }
+  }
+
+  // ---- epilogue: O /= l (per-row, via red_lds), store O + LSE ----
+  red_lds[wave][col] = l_run > 0.f ? 1.f / l_run : 0.f;
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  float li16[16];
+#pragma unroll
+  for (int r = 0; r < 16; ++r)
+    li16[r] = red_lds[wave][(r & 3) + 8 * (r >> 2) + 4 * half];
+#pragma unroll
+  for (int cc = 0; cc < kD / 32; ++cc)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int qrow_r = q0w + (r & 3) + 8 * (r >> 2) + 4 * half;
+      if (qrow_r < S)
+        o[(qbase + qrow_r) * kD + cc * 32 + col] =
+            __float2bfloat16(oacc[cc][r] * li16[r]);
+    }
+  if (lse != nullptr && half == 0 && qi < S)
+    lse[(static_cast<int64_t>(b) * HQ + h) * S + qi] =
+        scale * m_run + __logf(l_run > 0.f ? l_run : 1.f);
+}
+
 // ---------------------------------------------------------------- backward
 // EXPERIMENTAL v0 (probe-only, never run in any model/test path until
 // round 2 validates it through tools/fa_probe.py --check-bwd). FA-2 style:
@@ -552,10 +764,19 @@ void launch_fa_bwd(const void* q, const void* k, const void* v,
 void launch_fa_fwd(const void* q, const void* k, const void* v, void* o,
                    float* lse, int B, int HQ, int HK, int S, int D,
                    bool causal, hipStream_t stream) {
-  if (D != kD) return;  // v0: Llama head dim only (wrapper validates)
+  if (D != kD) return;  // Llama head dim only (wrapper validates)
   float scale = 1.f / sqrtf(static_cast<float>(D));
-  dim3 grid((S + kQTile * kWavesFA - 1) / (kQTile * kWavesFA), HQ, B);
-  hipLaunchKernelGGL(fa_fwd_kernel, grid, 256, 0, stream,
+  const char* fv = getenv("AXON_FA_V");
+  if (fv && fv[0] == '0') {  // debug fallback: round-1 v0 kernel
+    dim3 grid((S + kQTile * kWavesFA - 1) / (kQTile * kWavesFA), HQ, B);
+    hipLaunchKernelGGL(fa_fwd_kernel, grid, 256, 0, stream,
+                       static_cast<const bf16*>(q), static_cast<const bf16*>(k),
+                       static_cast<const bf16*>(v), static_cast<bf16*>(o), lse,
+                       B, HQ, HK, S, scale, causal ? 1 : 0);
+    return;
+  }
+  dim3 grid((S + kBq2 - 1) / kBq2, HQ, B);
+  hipLaunchKernelGGL(fa_fwd_v2_kernel, grid, 512, 0, stream,
                      static_cast<const bf16*>(q), static_cast<const bf16*>(k),
                      static_cast<const bf16*>(v), static_cast<bf16*>(o), lse,
                      B, HQ, HK, S, scale, causal ? 1 : 0);

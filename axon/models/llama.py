@@ -77,9 +77,11 @@ class LlamaAttention(nn.Module):
         q = rope(self.wq(x).view(b, s, self.heads, self.head_dim), cos, sin)
         k = rope(self.wk(x).view(b, s, self.kv_heads, self.head_dim), cos, sin)
         v = self.wv(x).view(b, s, self.kv_heads, self.head_dim)
-        o = F.scaled_dot_product_attention(
-            q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2),
-            is_causal=True, enable_gqa=True)
+        # AXON_FA=1 routes through the CDNA4 flash-attention kernels
+        # (axon/ops/flash_attention.py); default is torch SDPA.
+        from axon.ops.flash_attention import flash_attention
+        o = flash_attention(q.transpose(1, 2), k.transpose(1, 2),
+                            v.transpose(1, 2), causal=True)
         return self.wo(o.transpose(1, 2).reshape(b, s, -1))
 
 
