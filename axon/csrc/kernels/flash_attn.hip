@@ -262,6 +262,25 @@ __device__ __forceinline__ float xhalf_swap_f(float x) {
   return __int_as_float(threadIdx.x % 64 < 32 ? r[1] : r[0]);
 }
 
+// Repack a 32x32 C-fragment column (16 f32, C rows (r&3)+8*(r>>2)+4*half)
+// into the two A-fragments of the NEXT mfma whose M-dim equals this C's
+// col-dim: pack adjacent rows to bf16 pairs, then 2 permlane32_swap per
+// 16-slice moves each half's missing quads across (guide T12).
+__device__ __forceinline__ void p16_to_afrags(const float p[16], frag_ab* a1,
+                                              frag_ab* a2) {
+  uint32_t pk[8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i) pk[i] = pack_bf16(p[2 * i], p[2 * i + 1]);
+  auto r0 = __builtin_amdgcn_permlane32_swap(pk[0], pk[2], false, false);
+  auto r1 = __builtin_amdgcn_permlane32_swap(pk[1], pk[3], false, false);
+  auto r2 = __builtin_amdgcn_permlane32_swap(pk[4], pk[6], false, false);
+  auto r3 = __builtin_amdgcn_permlane32_swap(pk[5], pk[7], false, false);
+  uint32_t* u1 = reinterpret_cast<uint32_t*>(a1);
+  uint32_t* u2 = reinterpret_cast<uint32_t*>(a2);
+  u1[0] = r0[0]; u1[1] = r1[0]; u1[2] = r0[1]; u1[3] = r1[1];
+  u2[0] = r2[0]; u2[1] = r3[0]; u2[2] = r2[1]; u2[3] = r3[1];
+}
+
 __global__ __launch_bounds__(512, 1) void fa_fwd_v2_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, bf16* __restrict__ o, float* __restrict__ lse,
@@ -383,20 +402,8 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_v2_kernel(
     l_run = l_run * alpha + rsum + xhalf_swap_f(rsum);
 
     // ---- pack P to PV A-fragments (in registers, T12) ----
-    uint32_t pk[8];
-#pragma unroll
-    for (int i = 0; i < 8; ++i) pk[i] = pack_bf16(p[2 * i], p[2 * i + 1]);
     frag_ab pa1, pa2;
-    {
-      auto r0 = __builtin_amdgcn_permlane32_swap(pk[0], pk[2], false, false);
-      auto r1 = __builtin_amdgcn_permlane32_swap(pk[1], pk[3], false, false);
-      auto r2 = __builtin_amdgcn_permlane32_swap(pk[4], pk[6], false, false);
-      auto r3 = __builtin_amdgcn_permlane32_swap(pk[5], pk[7], false, false);
-      uint32_t* a1 = reinterpret_cast<uint32_t*>(&pa1);
-      uint32_t* a2 = reinterpret_cast<uint32_t*>(&pa2);
-      a1[0] = r0[0]; a1[1] = r1[0]; a1[2] = r0[1]; a1[3] = r1[1];
-      a2[0] = r2[0]; a2[1] = r3[0]; a2[2] = r2[1]; a2[3] = r3[1];
-    }
+    p16_to_afrags(p, &pa1, &pa2);
 
     // ---- O += P V (B-frag: 8 consecutive keys from transposed V) ----
 #pragma unroll
@@ -732,28 +739,363 @@ __global__ void fa_bwd_dq_kernel(
     }
 }
 
+// ------------------------------------------------------------ backward v2
+// Same structural idioms as the v2 forward: 32x32x16 MFMA tiles, operands
+// staged once per tile in LDS (row-major for A/B fragments whose M/N dim
+// is the lane index; transposed [d][row] images for the B fragments of
+// the accumulating GEMMs), P/dS repacked to A-fragments in registers via
+// p16_to_afrags. FA-2 recompute from the forward's logsumexp; GQA
+// accumulated inside the k-tile loop (no atomics, no workspace).
+//
+// dkdv: 4 waves over 128 k rows of one kv head. Per (q-head, 32-q tile):
+//   S(32q x 32k)  = Q K^T      A = Q (LDS row),   B = K (LDS row)
+//   dP(32q x 32k) = dO V^T     A = dO (LDS row),  B = V (LDS row)
+//   P = exp(scale*S - lse); dS = P (dP - delta) scale   [in-register]
+//   dV += P^T dO   A = repack(P),  B = dO ([d][q] image)
+//   dK += dS^T Q   A = repack(dS), B = Q  ([d][q] image)
+// dq: 4 waves over 128 q rows of one q head. Per 32-k tile:
+//   S^T = K Q^T, dP^T = V dO^T (Q/dO persistent registers), dS^T
+//   in-register (lse/delta are per-lane scalars), dQ += dS K with K in a
+//   transposed [d][k] image.
+
+constexpr int kBK = 128;  // k rows per dkdv block / q rows per dq block
+constexpr int kW4 = 4;
+
+__global__ __launch_bounds__(256, 2) void fa_bwd_dkdv_v2_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ k,
+    const bf16* __restrict__ v, const bf16* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    bf16* __restrict__ dk, bf16* __restrict__ dv, int B, int HQ, int HK,
+    int S, float scale, int causal) {
+  const int lane = threadIdx.x % 64;
+  const int wave = threadIdx.x / 64;
+  const int idx = threadIdx.x;  // 0..255
+  const int col = lane & 31;
+  const int half = lane >> 5;
+  const int kb0 = blockIdx.x * kBK;
+  const int hkv = blockIdx.y;
+  const int b = blockIdx.z;
+  const int gqa = HQ / HK;
+  const int kw0 = kb0 + wave * 32;  // this wave's k-tile base
+
+  __shared__ bf16 q_row[32][kD + kKPad];
+  __shared__ bf16 q_tr[kD][32 + kVPad];
+  __shared__ bf16 do_row[32][kD + kKPad];
+  __shared__ bf16 do_tr[kD][32 + kVPad];
+  __shared__ float lse_lds[32], dlt_lds[32];
+
+  const int64_t kbase = (static_cast<int64_t>(b) * HK + hkv) * S;
+
+  // This wave's K/V rows live in registers for the whole kernel: the lane
+  // layout of a B fragment (col = lane&31, k-dim = (lane>>5)*8+j) matches a
+  // per-lane row load, so no LDS image is needed (keeps LDS under 40 KB ->
+  // 2 blocks/CU instead of the 108 KB 1-block version).
+  const int kc = kw0 + col;
+  const int krow_g = kc < S ? kc : S - 1;
+  frag_ab kreg[kD / 16], vreg[kD / 16];
+#pragma unroll
+  for (int c = 0; c < kD / 16; ++c) {
+    kreg[c] = *reinterpret_cast<const frag_ab*>(
+        k + (kbase + krow_g) * kD + c * 16 + half * 8);
+    vreg[c] = *reinterpret_cast<const frag_ab*>(
+        v + (kbase + krow_g) * kD + c * 16 + half * 8);
+  }
+
+  frag16 dvacc[kD / 32], dkacc[kD / 32];
+#pragma unroll
+  for (int cc = 0; cc < kD / 32; ++cc)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      dvacc[cc][r] = 0.f;
+      dkacc[cc][r] = 0.f;
+    }
+
+  const int qsr = idx & 31, qsd = (idx >> 5) * 16;  // staging map
+  const int q_lo = causal ? kb0 : 0;
+
+  for (int g = 0; g < gqa; ++g) {
+    const int h = hkv * gqa + g;
+    const int64_t qbase = (static_cast<int64_t>(b) * HQ + h) * S;
+    const float* lse_h = lse + (static_cast<int64_t>(b) * HQ + h) * S;
+    const float* dlt_h = delta + (static_cast<int64_t>(b) * HQ + h) * S;
+
+    for (int q0 = q_lo; q0 < S; q0 += 32) {
+      // ---- stage Q/dO tile (row + transposed images) ----
+      {
+        int qr = q0 + qsr < S ? q0 + qsr : S - 1;
+        float4 qa = *reinterpret_cast<const float4*>(
+            q + (qbase + qr) * kD + qsd);
+        float4 qb2 = *reinterpret_cast<const float4*>(
+            q + (qbase + qr) * kD + qsd + 8);
+        float4 da = *reinterpret_cast<const float4*>(
+            dout + (qbase + qr) * kD + qsd);
+        float4 db = *reinterpret_cast<const float4*>(
+            dout + (qbase + qr) * kD + qsd + 8);
+        float lv = 0.f, dvl = 0.f;
+        if (idx < 32) {
+          int qi2 = q0 + idx < S ? q0 + idx : S - 1;
+          lv = lse_h[qi2];
+          dvl = dlt_h[qi2];
+        }
+        __syncthreads();  // previous tile fully consumed
+        *reinterpret_cast<float4*>(&q_row[qsr][qsd]) = qa;
+        *reinterpret_cast<float4*>(&q_row[qsr][qsd + 8]) = qb2;
+        *reinterpret_cast<float4*>(&do_row[qsr][qsd]) = da;
+        *reinterpret_cast<float4*>(&do_row[qsr][qsd + 8]) = db;
+        const bf16* qp = reinterpret_cast<const bf16*>(&qa);
+        const bf16* qp2 = reinterpret_cast<const bf16*>(&qb2);
+        const bf16* dp_ = reinterpret_cast<const bf16*>(&da);
+        const bf16* dp2 = reinterpret_cast<const bf16*>(&db);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          q_tr[qsd + j][qsr] = qp[j];
+          q_tr[qsd + 8 + j][qsr] = qp2[j];
+          do_tr[qsd + j][qsr] = dp_[j];
+          do_tr[qsd + 8 + j][qsr] = dp2[j];
+        }
+        if (idx < 32) {
+          lse_lds[idx] = lv;
+          dlt_lds[idx] = dvl;
+        }
+        __syncthreads();
+      }
+
+      // ---- S = Q K^T and dP = dO V^T (C: col = k, rows = 16 q) ----
+      frag16 st, dp;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        st[r] = 0.f;
+        dp[r] = 0.f;
+      }
+#pragma unroll
+      for (int c = 0; c < kD / 16; ++c) {
+        frag_ab qa = *reinterpret_cast<const frag_ab*>(
+            &q_row[col][c * 16 + half * 8]);
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kreg[c], st, 0, 0, 0);
+        frag_ab da = *reinterpret_cast<const frag_ab*>(
+            &do_row[col][c * 16 + half * 8]);
+        dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vreg[c], dp, 0, 0, 0);
+      }
+
+      // ---- P and dS (in-register; lse/delta via LDS broadcast) ----
+      float pv[16], ds[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int qroff = (r & 3) + 8 * (r >> 2) + 4 * half;
+        int qi2 = q0 + qroff;
+        bool dead = kc >= S || qi2 >= S || (causal && kc > qi2);
+        pv[r] = dead ? 0.f
+                     : __expf(scale * st[r] - lse_lds[qroff]);
+        ds[r] = dead ? 0.f : pv[r] * (dp[r] - dlt_lds[qroff]) * scale;
+      }
+
+      // ---- dV += P^T dO, dK += dS^T Q ----
+      frag_ab pa1, pa2, sa1, sa2;
+      p16_to_afrags(pv, &pa1, &pa2);
+      p16_to_afrags(ds, &sa1, &sa2);
+#pragma unroll
+      for (int cc = 0; cc < kD / 32; ++cc) {
+        frag_ab d1 = *reinterpret_cast<const frag_ab*>(
+            &do_tr[cc * 32 + col][half * 8]);
+        frag_ab d2 = *reinterpret_cast<const frag_ab*>(
+            &do_tr[cc * 32 + col][16 + half * 8]);
+        dvacc[cc] =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa1, d1, dvacc[cc], 0, 0, 0);
+        dvacc[cc] =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa2, d2, dvacc[cc], 0, 0, 0);
+        frag_ab q1 = *reinterpret_cast<const frag_ab*>(
+            &q_tr[cc * 32 + col][half * 8]);
+        frag_ab q2 = *reinterpret_cast<const frag_ab*>(
+            &q_tr[cc * 32 + col][16 + half * 8]);
+        dkacc[cc] =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(sa1, q1, dkacc[cc], 0, 0, 0);
+        dkacc[cc] =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(sa2, q2, dkacc[cc], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- store dK/dV (rows = k, col = d within cc) ----
+#pragma unroll
+  for (int cc = 0; cc < kD / 32; ++cc)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int ki = kw0 + (r & 3) + 8 * (r >> 2) + 4 * half;
+      if (ki < S) {
+        dk[(kbase + ki) * kD + cc * 32 + col] = __float2bfloat16(dkacc[cc][r]);
+        dv[(kbase + ki) * kD + cc * 32 + col] = __float2bfloat16(dvacc[cc][r]);
+      }
+    }
+}
+
+__global__ __launch_bounds__(256, 2) void fa_bwd_dq_v2_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ k,
+    const bf16* __restrict__ v, const bf16* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    bf16* __restrict__ dq, int B, int HQ, int HK, int S, float scale,
+    int causal) {
+  const int lane = threadIdx.x % 64;
+  const int wave = threadIdx.x / 64;
+  const int idx = threadIdx.x;
+  const int col = lane & 31;
+  const int half = lane >> 5;
+  const int qb = blockIdx.x * kBK;
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int hk = h / (HQ / HK);
+  const int q0w = qb + wave * 32;
+
+  __shared__ bf16 k_row2[32][kD + kKPad];
+  __shared__ bf16 v_row2[32][kD + kKPad];
+  __shared__ bf16 k_tr[kD][32 + kVPad];
+
+  const int64_t qbase = (static_cast<int64_t>(b) * HQ + h) * S;
+  const int64_t kbase = (static_cast<int64_t>(b) * HK + hk) * S;
+  const float* lse_h = lse + (static_cast<int64_t>(b) * HQ + h) * S;
+  const float* dlt_h = delta + (static_cast<int64_t>(b) * HQ + h) * S;
+
+  const int qi = q0w + col;
+  const int qrow = qi < S ? qi : S - 1;
+  frag_ab qreg[kD / 16], doreg[kD / 16];
+#pragma unroll
+  for (int c = 0; c < kD / 16; ++c) {
+    qreg[c] = *reinterpret_cast<const frag_ab*>(
+        q + (qbase + qrow) * kD + c * 16 + half * 8);
+    doreg[c] = *reinterpret_cast<const frag_ab*>(
+        dout + (qbase + qrow) * kD + c * 16 + half * 8);
+  }
+  const float lse_q = qi < S ? lse_h[qi] : 0.f;
+  const float dlt_q = qi < S ? dlt_h[qi] : 0.f;
+
+  frag16 dqacc[kD / 32];
+#pragma unroll
+  for (int cc = 0; cc < kD / 32; ++cc)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dqacc[cc][r] = 0.f;
+
+  const int ksr = idx & 31, ksd = (idx >> 5) * 16;
+  const int kend = causal ? (qb + kBK < S ? qb + kBK : S) : S;
+  const int ntiles = (kend + 31) / 32;
+
+  for (int t = 0; t < ntiles; ++t) {
+    const int k0 = t * 32;
+    {
+      int kr = k0 + ksr < S ? k0 + ksr : S - 1;
+      float4 ka = *reinterpret_cast<const float4*>(k + (kbase + kr) * kD + ksd);
+      float4 kb2 =
+          *reinterpret_cast<const float4*>(k + (kbase + kr) * kD + ksd + 8);
+      float4 va = *reinterpret_cast<const float4*>(v + (kbase + kr) * kD + ksd);
+      float4 vb =
+          *reinterpret_cast<const float4*>(v + (kbase + kr) * kD + ksd + 8);
+      __syncthreads();
+      *reinterpret_cast<float4*>(&k_row2[ksr][ksd]) = ka;
+      *reinterpret_cast<float4*>(&k_row2[ksr][ksd + 8]) = kb2;
+      *reinterpret_cast<float4*>(&v_row2[ksr][ksd]) = va;
+      *reinterpret_cast<float4*>(&v_row2[ksr][ksd + 8]) = vb;
+      const bf16* kp = reinterpret_cast<const bf16*>(&ka);
+      const bf16* kp2 = reinterpret_cast<const bf16*>(&kb2);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        k_tr[ksd + j][ksr] = kp[j];
+        k_tr[ksd + 8 + j][ksr] = kp2[j];
+      }
+      __syncthreads();
+    }
+
+    // ---- S^T = K Q^T, dP^T = V dO^T (C: col = q, rows = 16 k) ----
+    frag16 st, dpt;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      st[r] = 0.f;
+      dpt[r] = 0.f;
+    }
+#pragma unroll
+    for (int c = 0; c < kD / 16; ++c) {
+      frag_ab ka = *reinterpret_cast<const frag_ab*>(
+          &k_row2[col][c * 16 + half * 8]);
+      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qreg[c], st, 0, 0, 0);
+      frag_ab va = *reinterpret_cast<const frag_ab*>(
+          &v_row2[col][c * 16 + half * 8]);
+      dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, doreg[c], dpt, 0, 0, 0);
+    }
+
+    // ---- dS^T in-register (per-lane lse/delta scalars) ----
+    float ds[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int ki = k0 + (r & 3) + 8 * (r >> 2) + 4 * half;
+      bool dead = ki >= S || qi >= S || (causal && ki > qi);
+      float pT = dead ? 0.f : __expf(scale * st[r] - lse_q);
+      ds[r] = dead ? 0.f : pT * (dpt[r] - dlt_q) * scale;
+    }
+
+    // ---- dQ += dS K (B from transposed K image) ----
+    frag_ab a1, a2;
+    p16_to_afrags(ds, &a1, &a2);
+#pragma unroll
+    for (int cc = 0; cc < kD / 32; ++cc) {
+      frag_ab k1 = *reinterpret_cast<const frag_ab*>(
+          &k_tr[cc * 32 + col][half * 8]);
+      frag_ab k2 = *reinterpret_cast<const frag_ab*>(
+          &k_tr[cc * 32 + col][16 + half * 8]);
+      dqacc[cc] =
+          __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, k1, dqacc[cc], 0, 0, 0);
+      dqacc[cc] =
+          __builtin_amdgcn_mfma_f32_32x32x16_bf16(a2, k2, dqacc[cc], 0, 0, 0);
+    }
+  }
+
+#pragma unroll
+  for (int cc = 0; cc < kD / 32; ++cc)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int qr2 = q0w + (r & 3) + 8 * (r >> 2) + 4 * half;
+      if (qr2 < S)
+        dq[(qbase + qr2) * kD + cc * 32 + col] =
+            __float2bfloat16(dqacc[cc][r]);
+    }
+}
+
 }  // namespace
 
 void launch_fa_bwd(const void* q, const void* k, const void* v,
                    const void* dout, const void* o, const float* lse,
                    float* delta, void* dq, void* dk, void* dv, int B, int HQ,
                    int HK, int S, int D, bool causal, hipStream_t stream) {
-  if (D != kD) return;  // v0: head dim 128 only (wrapper validates)
+  if (D != kD) return;  // head dim 128 only (wrapper validates)
   float scale = 1.f / sqrtf(static_cast<float>(D));
   int64_t rows = static_cast<int64_t>(B) * HQ * S;
   hipLaunchKernelGGL(fa_delta_kernel,
                      static_cast<unsigned>((rows + kWavesFA - 1) / kWavesFA),
                      256, 0, stream, static_cast<const bf16*>(o),
                      static_cast<const bf16*>(dout), delta, rows);
-  dim3 gkv((S + kQTile * kWavesFA - 1) / (kQTile * kWavesFA), HK, B);
-  hipLaunchKernelGGL(fa_bwd_dkdv_kernel, gkv, 256, 0, stream,
+  const char* fv = getenv("AXON_FA_V");
+  if (fv && fv[0] == '0') {  // debug fallback: round-1 v0 kernels
+    dim3 gkv((S + kQTile * kWavesFA - 1) / (kQTile * kWavesFA), HK, B);
+    hipLaunchKernelGGL(fa_bwd_dkdv_kernel, gkv, 256, 0, stream,
+                       static_cast<const bf16*>(q), static_cast<const bf16*>(k),
+                       static_cast<const bf16*>(v),
+                       static_cast<const bf16*>(dout), lse, delta,
+                       static_cast<bf16*>(dk), static_cast<bf16*>(dv), B, HQ,
+                       HK, S, scale, causal ? 1 : 0);
+    dim3 gq((S + kQTile * kWavesFA - 1) / (kQTile * kWavesFA), HQ, B);
+    hipLaunchKernelGGL(fa_bwd_dq_kernel, gq, 256, 0, stream,
+                       static_cast<const bf16*>(q), static_cast<const bf16*>(k),
+                       static_cast<const bf16*>(v),
+                       static_cast<const bf16*>(dout), lse, delta,
+                       static_cast<bf16*>(dq), B, HQ, HK, S, scale,
+                       causal ? 1 : 0);
+    return;
+  }
+  dim3 gkv((S + kBK - 1) / kBK, HK, B);
+  hipLaunchKernelGGL(fa_bwd_dkdv_v2_kernel, gkv, 256, 0, stream,
                      static_cast<const bf16*>(q), static_cast<const bf16*>(k),
                      static_cast<const bf16*>(v),
                      static_cast<const bf16*>(dout), lse, delta,
                      static_cast<bf16*>(dk), static_cast<bf16*>(dv), B, HQ,
                      HK, S, scale, causal ? 1 : 0);
-  dim3 gq((S + kQTile * kWavesFA - 1) / (kQTile * kWavesFA), HQ, B);
-  hipLaunchKernelGGL(fa_bwd_dq_kernel, gq, 256, 0, stream,
+  dim3 gq((S + kBK - 1) / kBK, HQ, B);
+  hipLaunchKernelGGL(fa_bwd_dq_v2_kernel, gq, 256, 0, stream,
                      static_cast<const bf16*>(q), static_cast<const bf16*>(k),
                      static_cast<const bf16*>(v),
                      static_cast<const bf16*>(dout), lse, delta,
