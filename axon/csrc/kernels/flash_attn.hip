@@ -296,8 +296,8 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_v2_kernel(
   const int hk = h / (HQ / HK);
   const int q0w = qb + wave * kBr2;
 
-  __shared__ bf16 k_lds[kBc2][kD + kKPad];        // row-major K tile
-  __shared__ bf16 v_lds[kD][kBc2 + kVPad];        // transposed V tile
+  __shared__ bf16 k_lds[2][kBc2][kD + kKPad];     // row-major K tiles (dbuf)
+  __shared__ bf16 v_lds[2][kD][kBc2 + kVPad];     // transposed V tiles (dbuf)
   __shared__ float red_lds[kW2][kBr2];            // per-row alpha/l exchange
 
   const int64_t qbase = (static_cast<int64_t>(b) * HQ + h) * S;
@@ -328,23 +328,33 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_v2_kernel(
   const int kend = causal ? (qb + kBq2 < S ? qb + kBq2 : S) : S;
   const int ntiles = (kend + kBc2 - 1) / kBc2;
 
+  // T14 issue-early/write-late with double-buffered LDS: tile t+1's global
+  // loads are issued before tile t's compute, the LDS write lands after it,
+  // and ONE barrier per tile publishes the next buffer.
+  auto load_tile = [&](int k0, float4* kv4, float4* vv4) {
+    *kv4 = float4{0.f, 0.f, 0.f, 0.f};
+    *vv4 = float4{0.f, 0.f, 0.f, 0.f};
+    if (k0 + kr < S)
+      *kv4 = *reinterpret_cast<const float4*>(k + (kbase + k0 + kr) * kD + kdc);
+    if (k0 + vr < S)
+      *vv4 = *reinterpret_cast<const float4*>(v + (kbase + k0 + vr) * kD + vdc);
+  };
+  auto write_tile = [&](int buf, const float4& kv4, const float4& vv4) {
+    *reinterpret_cast<float4*>(&k_lds[buf][kr][kdc]) = kv4;
+    const bf16* vb = reinterpret_cast<const bf16*>(&vv4);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) v_lds[buf][vdc + j][vr] = vb[j];
+  };
+
+  float4 kv4, vv4;
+  load_tile(0, &kv4, &vv4);
+  write_tile(0, kv4, vv4);
+  __syncthreads();
+
   for (int t = 0; t < ntiles; ++t) {
     const int k0 = t * kBc2;
-    // ---- stage tile t (all 8 waves cooperate) ----
-    {
-      float4 kv4 = float4{0.f, 0.f, 0.f, 0.f};
-      float4 vv4 = float4{0.f, 0.f, 0.f, 0.f};
-      if (k0 + kr < S)
-        kv4 = *reinterpret_cast<const float4*>(k + (kbase + k0 + kr) * kD + kdc);
-      if (k0 + vr < S)
-        vv4 = *reinterpret_cast<const float4*>(v + (kbase + k0 + vr) * kD + vdc);
-      __syncthreads();  // everyone done READING the previous tile
-      *reinterpret_cast<float4*>(&k_lds[kr][kdc]) = kv4;
-      const bf16* vb = reinterpret_cast<const bf16*>(&vv4);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) v_lds[vdc + j][vr] = vb[j];
-      __syncthreads();
-    }
+    const int buf = t & 1;
+    if (t + 1 < ntiles) load_tile(k0 + kBc2, &kv4, &vv4);
 
     // ---- S^T = K Q^T (A = K rows, B = Q cols) ----
     frag16 st;
@@ -353,7 +363,7 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_v2_kernel(
 #pragma unroll
     for (int c = 0; c < kD / 16; ++c) {
       frag_ab kf = *reinterpret_cast<const frag_ab*>(
-          &k_lds[col][c * 16 + half * 8]);
+          &k_lds[buf][col][c * 16 + half * 8]);
       st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[c], st, 0, 0, 0);
     }
 
@@ -409,14 +419,17 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_v2_kernel(
 #pragma unroll
     for (int cc = 0; cc < kD / 32; ++cc) {
       frag_ab v1 = *reinterpret_cast<const frag_ab*>(
-          &v_lds[cc * 32 + col][half * 8]);
+          &v_lds[buf][cc * 32 + col][half * 8]);
       frag_ab v2 = *reinterpret_cast<const frag_ab*>(
-          &v_lds[cc * 32 + col][16 + half * 8]);
+          &v_lds[buf][cc * 32 + col][16 + half * 8]);
       oacc[cc] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa1, v1, oacc[cc],
                                                          0, 0, 0);
       oacc[cc] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa2, v2, oacc[cc],
                                                          0, 0, 0);
     }
+
+    if (t + 1 < ntiles) write_tile(buf ^ 1, kv4, vv4);
+    __syncthreads();  // publish buf^1; also fences buf re-use next round
   }
 
   // ---- epilogue: O /= l (per-row, via red_lds), store O + LSE ----
