@@ -266,6 +266,18 @@ __device__ __forceinline__ float xhalf_swap_f(float x) {
 // into the two A-fragments of the NEXT mfma whose M-dim equals this C's
 // col-dim: pack adjacent rows to bf16 pairs, then 2 permlane32_swap per
 // 16-slice moves each half's missing quads across (guide T12).
+__device__ __forceinline__ void packs_to_afrags(const uint32_t pk[8],
+                                                frag_ab* a1, frag_ab* a2) {
+  auto r0 = __builtin_amdgcn_permlane32_swap(pk[0], pk[2], false, false);
+  auto r1 = __builtin_amdgcn_permlane32_swap(pk[1], pk[3], false, false);
+  auto r2 = __builtin_amdgcn_permlane32_swap(pk[4], pk[6], false, false);
+  auto r3 = __builtin_amdgcn_permlane32_swap(pk[5], pk[7], false, false);
+  uint32_t* u1 = reinterpret_cast<uint32_t*>(a1);
+  uint32_t* u2 = reinterpret_cast<uint32_t*>(a2);
+  u1[0] = r0[0]; u1[1] = r1[0]; u1[2] = r0[1]; u1[3] = r1[1];
+  u2[0] = r2[0]; u2[1] = r3[0]; u2[2] = r2[1]; u2[3] = r3[1];
+}
+
 __device__ __forceinline__ void p16_to_afrags(const float p[16], frag_ab* a1,
                                               frag_ab* a2) {
   uint32_t pk[8];
@@ -357,15 +369,27 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_v2_kernel(
     if (t + 1 < ntiles) load_tile(k0 + kBc2, &kv4, &vv4);
 
     // ---- S^T = K Q^T (A = K rows, B = Q cols) ----
-    frag16 st;
+    // Two independent accumulator chains (even/odd d-chunks) halve the
+    // dependent-MFMA stall of one 8-deep chain; merged with 16 VALU adds.
+    frag16 st, st1;
 #pragma unroll
-    for (int r = 0; r < 16; ++r) st[r] = 0.f;
+    for (int r = 0; r < 16; ++r) {
+      st[r] = 0.f;
+      st1[r] = 0.f;
+    }
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int c = 0; c < kD / 16; ++c) {
+    for (int c = 0; c < kD / 16; c += 2) {
       frag_ab kf = *reinterpret_cast<const frag_ab*>(
           &k_lds[buf][col][c * 16 + half * 8]);
       st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[c], st, 0, 0, 0);
+      frag_ab kf1 = *reinterpret_cast<const frag_ab*>(
+          &k_lds[buf][col][(c + 1) * 16 + half * 8]);
+      st1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf1, qf[c + 1], st1, 0, 0, 0);
     }
+    __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+    for (int r = 0; r < 16; ++r) st[r] += st1[r];
 
     // ---- masking + tile max (raw-score domain) ----
     const bool full_tile = causal ? (k0 + kBc2 - 1 <= q0w) : (k0 + kBc2 <= S);
@@ -416,6 +440,7 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_v2_kernel(
     p16_to_afrags(p, &pa1, &pa2);
 
     // ---- O += P V (B-frag: 8 consecutive keys from transposed V) ----
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int cc = 0; cc < kD / 32; ++cc) {
       frag_ab v1 = *reinterpret_cast<const frag_ab*>(
@@ -427,6 +452,7 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_v2_kernel(
       oacc[cc] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa2, v2, oacc[cc],
                                                          0, 0, 0);
     }
+    __builtin_amdgcn_s_setprio(0);
 
     if (t + 1 < ntiles) write_tile(buf ^ 1, kv4, vv4);
     __syncthreads();  // publish buf^1; also fences buf re-use next round
@@ -791,11 +817,11 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dkdv_v2_kernel(
   const int gqa = HQ / HK;
   const int kw0 = kb0 + wave * 32;  // this wave's k-tile base
 
-  __shared__ bf16 q_row[32][kD + kKPad];
-  __shared__ bf16 q_tr[kD][32 + kVPad];
-  __shared__ bf16 do_row[32][kD + kKPad];
-  __shared__ bf16 do_tr[kD][32 + kVPad];
-  __shared__ float lse_lds[32], dlt_lds[32];
+  __shared__ bf16 q_row[2][32][kD + kKPad];
+  __shared__ bf16 q_tr[2][kD][32 + kVPad];
+  __shared__ bf16 do_row[2][32][kD + kKPad];
+  __shared__ bf16 do_tr[2][kD][32 + kVPad];
+  __shared__ float lse_lds[2][32], dlt_lds[2][32];
 
   const int64_t kbase = (static_cast<int64_t>(b) * HK + hkv) * S;
 
@@ -825,53 +851,62 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dkdv_v2_kernel(
 
   const int qsr = idx & 31, qsd = (idx >> 5) * 16;  // staging map
   const int q_lo = causal ? kb0 : 0;
+  const int nq = (S - q_lo + 31) / 32;
+  const int niter = gqa * nq;
 
-  for (int g = 0; g < gqa; ++g) {
-    const int h = hkv * gqa + g;
-    const int64_t qbase = (static_cast<int64_t>(b) * HQ + h) * S;
-    const float* lse_h = lse + (static_cast<int64_t>(b) * HQ + h) * S;
-    const float* dlt_h = delta + (static_cast<int64_t>(b) * HQ + h) * S;
-
-    for (int q0 = q_lo; q0 < S; q0 += 32) {
-      // ---- stage Q/dO tile (row + transposed images) ----
-      {
-        int qr = q0 + qsr < S ? q0 + qsr : S - 1;
-        float4 qa = *reinterpret_cast<const float4*>(
-            q + (qbase + qr) * kD + qsd);
-        float4 qb2 = *reinterpret_cast<const float4*>(
-            q + (qbase + qr) * kD + qsd + 8);
-        float4 da = *reinterpret_cast<const float4*>(
-            dout + (qbase + qr) * kD + qsd);
-        float4 db = *reinterpret_cast<const float4*>(
-            dout + (qbase + qr) * kD + qsd + 8);
-        float lv = 0.f, dvl = 0.f;
-        if (idx < 32) {
-          int qi2 = q0 + idx < S ? q0 + idx : S - 1;
-          lv = lse_h[qi2];
-          dvl = dlt_h[qi2];
-        }
-        __syncthreads();  // previous tile fully consumed
-        *reinterpret_cast<float4*>(&q_row[qsr][qsd]) = qa;
-        *reinterpret_cast<float4*>(&q_row[qsr][qsd + 8]) = qb2;
-        *reinterpret_cast<float4*>(&do_row[qsr][qsd]) = da;
-        *reinterpret_cast<float4*>(&do_row[qsr][qsd + 8]) = db;
-        const bf16* qp = reinterpret_cast<const bf16*>(&qa);
-        const bf16* qp2 = reinterpret_cast<const bf16*>(&qb2);
-        const bf16* dp_ = reinterpret_cast<const bf16*>(&da);
-        const bf16* dp2 = reinterpret_cast<const bf16*>(&db);
+  // Flattened (q-head, q-tile) loop with T14 double-buffered staging.
+  // Prefetch holds the FIRST 16B of each row (T14); the adjacent second
+  // 16B lands on the same cache line and is loaded at write time.
+  float4 qa, da;
+  int64_t pre_row = 0, pre_stat = 0;
+  auto load_qdo = [&](int it) {
+    int g = it / nq, q0 = q_lo + (it % nq) * 32;
+    int64_t qb_ = (static_cast<int64_t>(b) * HQ + hkv * gqa + g) * S;
+    int qr = q0 + qsr < S ? q0 + qsr : S - 1;
+    pre_row = qb_ + qr;
+    qa = *reinterpret_cast<const float4*>(q + pre_row * kD + qsd);
+    da = *reinterpret_cast<const float4*>(dout + pre_row * kD + qsd);
+    pre_stat = qb_ + (q0 + idx < S ? q0 + idx : S - 1);
+  };
+  auto write_qdo = [&](int bu) {
+    float4 qb2 = *reinterpret_cast<const float4*>(q + pre_row * kD + qsd + 8);
+    float4 db = *reinterpret_cast<const float4*>(dout + pre_row * kD + qsd + 8);
+    float lv = 0.f, dvl = 0.f;
+    if (idx < 32) {
+      lv = lse[pre_stat];
+      dvl = delta[pre_stat];
+    }
+    *reinterpret_cast<float4*>(&q_row[bu][qsr][qsd]) = qa;
+    *reinterpret_cast<float4*>(&q_row[bu][qsr][qsd + 8]) = qb2;
+    *reinterpret_cast<float4*>(&do_row[bu][qsr][qsd]) = da;
+    *reinterpret_cast<float4*>(&do_row[bu][qsr][qsd + 8]) = db;
+    const bf16* qp = reinterpret_cast<const bf16*>(&qa);
+    const bf16* qp2 = reinterpret_cast<const bf16*>(&qb2);
+    const bf16* dp_ = reinterpret_cast<const bf16*>(&da);
+    const bf16* dp2 = reinterpret_cast<const bf16*>(&db);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          q_tr[qsd + j][qsr] = qp[j];
-          q_tr[qsd + 8 + j][qsr] = qp2[j];
-          do_tr[qsd + j][qsr] = dp_[j];
-          do_tr[qsd + 8 + j][qsr] = dp2[j];
-        }
-        if (idx < 32) {
-          lse_lds[idx] = lv;
-          dlt_lds[idx] = dvl;
-        }
-        __syncthreads();
-      }
+    for (int j = 0; j < 8; ++j) {
+      q_tr[bu][qsd + j][qsr] = qp[j];
+      q_tr[bu][qsd + 8 + j][qsr] = qp2[j];
+      do_tr[bu][qsd + j][qsr] = dp_[j];
+      do_tr[bu][qsd + 8 + j][qsr] = dp2[j];
+    }
+    if (idx < 32) {
+      lse_lds[bu][idx] = lv;
+      dlt_lds[bu][idx] = dvl;
+    }
+  };
+  if (niter > 0) {
+    load_qdo(0);
+    write_qdo(0);
+  }
+  __syncthreads();
+
+  for (int it = 0; it < niter; ++it) {
+    {
+      const int q0 = q_lo + (it % nq) * 32;
+      const int bu = it & 1;
+      if (it + 1 < niter) load_qdo(it + 1);
 
       // ---- S = Q K^T and dP = dO V^T (C: col = k, rows = 16 q) ----
       frag16 st, dp;
@@ -880,15 +915,17 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dkdv_v2_kernel(
         st[r] = 0.f;
         dp[r] = 0.f;
       }
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int c = 0; c < kD / 16; ++c) {
-        frag_ab qa = *reinterpret_cast<const frag_ab*>(
-            &q_row[col][c * 16 + half * 8]);
-        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kreg[c], st, 0, 0, 0);
-        frag_ab da = *reinterpret_cast<const frag_ab*>(
-            &do_row[col][c * 16 + half * 8]);
-        dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vreg[c], dp, 0, 0, 0);
+        frag_ab qaf = *reinterpret_cast<const frag_ab*>(
+            &q_row[bu][col][c * 16 + half * 8]);
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qaf, kreg[c], st, 0, 0, 0);
+        frag_ab daf = *reinterpret_cast<const frag_ab*>(
+            &do_row[bu][col][c * 16 + half * 8]);
+        dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(daf, vreg[c], dp, 0, 0, 0);
       }
+      __builtin_amdgcn_s_setprio(0);
 
       // ---- P and dS (in-register; lse/delta via LDS broadcast) ----
       float pv[16], ds[16];
@@ -898,33 +935,38 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dkdv_v2_kernel(
         int qi2 = q0 + qroff;
         bool dead = kc >= S || qi2 >= S || (causal && kc > qi2);
         pv[r] = dead ? 0.f
-                     : __expf(scale * st[r] - lse_lds[qroff]);
-        ds[r] = dead ? 0.f : pv[r] * (dp[r] - dlt_lds[qroff]) * scale;
+                     : __expf(scale * st[r] - lse_lds[bu][qroff]);
+        ds[r] = dead ? 0.f : pv[r] * (dp[r] - dlt_lds[bu][qroff]) * scale;
       }
 
       // ---- dV += P^T dO, dK += dS^T Q ----
       frag_ab pa1, pa2, sa1, sa2;
       p16_to_afrags(pv, &pa1, &pa2);
       p16_to_afrags(ds, &sa1, &sa2);
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int cc = 0; cc < kD / 32; ++cc) {
         frag_ab d1 = *reinterpret_cast<const frag_ab*>(
-            &do_tr[cc * 32 + col][half * 8]);
+            &do_tr[bu][cc * 32 + col][half * 8]);
         frag_ab d2 = *reinterpret_cast<const frag_ab*>(
-            &do_tr[cc * 32 + col][16 + half * 8]);
+            &do_tr[bu][cc * 32 + col][16 + half * 8]);
         dvacc[cc] =
             __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa1, d1, dvacc[cc], 0, 0, 0);
         dvacc[cc] =
             __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa2, d2, dvacc[cc], 0, 0, 0);
         frag_ab q1 = *reinterpret_cast<const frag_ab*>(
-            &q_tr[cc * 32 + col][half * 8]);
+            &q_tr[bu][cc * 32 + col][half * 8]);
         frag_ab q2 = *reinterpret_cast<const frag_ab*>(
-            &q_tr[cc * 32 + col][16 + half * 8]);
+            &q_tr[bu][cc * 32 + col][16 + half * 8]);
         dkacc[cc] =
             __builtin_amdgcn_mfma_f32_32x32x16_bf16(sa1, q1, dkacc[cc], 0, 0, 0);
         dkacc[cc] =
             __builtin_amdgcn_mfma_f32_32x32x16_bf16(sa2, q2, dkacc[cc], 0, 0, 0);
       }
+      __builtin_amdgcn_s_setprio(0);
+
+      if (it + 1 < niter) write_qdo(bu ^ 1);
+      __syncthreads();
     }
   }
 
@@ -958,9 +1000,9 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dq_v2_kernel(
   const int hk = h / (HQ / HK);
   const int q0w = qb + wave * 32;
 
-  __shared__ bf16 k_row2[32][kD + kKPad];
-  __shared__ bf16 v_row2[32][kD + kKPad];
-  __shared__ bf16 k_tr[kD][32 + kVPad];
+  __shared__ bf16 k_row2[2][32][kD + kKPad];
+  __shared__ bf16 v_row2[2][32][kD + kKPad];
+  __shared__ bf16 k_tr[2][kD][32 + kVPad];
 
   const int64_t qbase = (static_cast<int64_t>(b) * HQ + h) * S;
   const int64_t kbase = (static_cast<int64_t>(b) * HK + hk) * S;
@@ -990,30 +1032,37 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dq_v2_kernel(
   const int kend = causal ? (qb + kBK < S ? qb + kBK : S) : S;
   const int ntiles = (kend + 31) / 32;
 
+  // T14: prefetch tile t+1's K/V during tile t's compute; one barrier per
+  // tile publishes the other LDS buffer.
+  float4 ka, kb2, va, vb;
+  auto load_kv = [&](int k0) {
+    int kr = k0 + ksr < S ? k0 + ksr : S - 1;
+    ka = *reinterpret_cast<const float4*>(k + (kbase + kr) * kD + ksd);
+    kb2 = *reinterpret_cast<const float4*>(k + (kbase + kr) * kD + ksd + 8);
+    va = *reinterpret_cast<const float4*>(v + (kbase + kr) * kD + ksd);
+    vb = *reinterpret_cast<const float4*>(v + (kbase + kr) * kD + ksd + 8);
+  };
+  auto write_kv = [&](int buf) {
+    *reinterpret_cast<float4*>(&k_row2[buf][ksr][ksd]) = ka;
+    *reinterpret_cast<float4*>(&k_row2[buf][ksr][ksd + 8]) = kb2;
+    *reinterpret_cast<float4*>(&v_row2[buf][ksr][ksd]) = va;
+    *reinterpret_cast<float4*>(&v_row2[buf][ksr][ksd + 8]) = vb;
+    const bf16* kp = reinterpret_cast<const bf16*>(&ka);
+    const bf16* kp2 = reinterpret_cast<const bf16*>(&kb2);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      k_tr[buf][ksd + j][ksr] = kp[j];
+      k_tr[buf][ksd + 8 + j][ksr] = kp2[j];
+    }
+  };
+  load_kv(0);
+  write_kv(0);
+  __syncthreads();
+
   for (int t = 0; t < ntiles; ++t) {
     const int k0 = t * 32;
-    {
-      int kr = k0 + ksr < S ? k0 + ksr : S - 1;
-      float4 ka = *reinterpret_cast<const float4*>(k + (kbase + kr) * kD + ksd);
-      float4 kb2 =
-          *reinterpret_cast<const float4*>(k + (kbase + kr) * kD + ksd + 8);
-      float4 va = *reinterpret_cast<const float4*>(v + (kbase + kr) * kD + ksd);
-      float4 vb =
-          *reinterpret_cast<const float4*>(v + (kbase + kr) * kD + ksd + 8);
-      __syncthreads();
-      *reinterpret_cast<float4*>(&k_row2[ksr][ksd]) = ka;
-      *reinterpret_cast<float4*>(&k_row2[ksr][ksd + 8]) = kb2;
-      *reinterpret_cast<float4*>(&v_row2[ksr][ksd]) = va;
-      *reinterpret_cast<float4*>(&v_row2[ksr][ksd + 8]) = vb;
-      const bf16* kp = reinterpret_cast<const bf16*>(&ka);
-      const bf16* kp2 = reinterpret_cast<const bf16*>(&kb2);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        k_tr[ksd + j][ksr] = kp[j];
-        k_tr[ksd + 8 + j][ksr] = kp2[j];
-      }
-      __syncthreads();
-    }
+    const int buf = t & 1;
+    if (t + 1 < ntiles) load_kv(k0 + 32);
 
     // ---- S^T = K Q^T, dP^T = V dO^T (C: col = q, rows = 16 k) ----
     frag16 st, dpt;
@@ -1024,12 +1073,12 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dq_v2_kernel(
     }
 #pragma unroll
     for (int c = 0; c < kD / 16; ++c) {
-      frag_ab ka = *reinterpret_cast<const frag_ab*>(
-          &k_row2[col][c * 16 + half * 8]);
-      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qreg[c], st, 0, 0, 0);
-      frag_ab va = *reinterpret_cast<const frag_ab*>(
-          &v_row2[col][c * 16 + half * 8]);
-      dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, doreg[c], dpt, 0, 0, 0);
+      frag_ab kaf = *reinterpret_cast<const frag_ab*>(
+          &k_row2[buf][col][c * 16 + half * 8]);
+      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kaf, qreg[c], st, 0, 0, 0);
+      frag_ab vaf = *reinterpret_cast<const frag_ab*>(
+          &v_row2[buf][col][c * 16 + half * 8]);
+      dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vaf, doreg[c], dpt, 0, 0, 0);
     }
 
     // ---- dS^T in-register (per-lane lse/delta scalars) ----
@@ -1048,14 +1097,17 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dq_v2_kernel(
 #pragma unroll
     for (int cc = 0; cc < kD / 32; ++cc) {
       frag_ab k1 = *reinterpret_cast<const frag_ab*>(
-          &k_tr[cc * 32 + col][half * 8]);
+          &k_tr[buf][cc * 32 + col][half * 8]);
       frag_ab k2 = *reinterpret_cast<const frag_ab*>(
-          &k_tr[cc * 32 + col][16 + half * 8]);
+          &k_tr[buf][cc * 32 + col][16 + half * 8]);
       dqacc[cc] =
           __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, k1, dqacc[cc], 0, 0, 0);
       dqacc[cc] =
           __builtin_amdgcn_mfma_f32_32x32x16_bf16(a2, k2, dqacc[cc], 0, 0, 0);
     }
+
+    if (t + 1 < ntiles) write_kv(buf ^ 1);
+    __syncthreads();
   }
 
 #pragma unroll
