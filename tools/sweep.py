@@ -21,7 +21,7 @@ def run_one(args, thr_mb, cyc_ms):
     if args.gpus > 1:
         cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
                f"--nproc-per-node={args.gpus}", "--master-addr", "127.0.0.1",
-               "--master-port", "29517", os.path.join(REPO, "bench.py")]
+               "--master-port", str(29500 + os.getpid() % 500), os.path.join(REPO, "bench.py")]
     else:
         cmd = [sys.executable, os.path.join(REPO, "bench.py")]
     cmd += ["--gpus", str(args.gpus), "--steps", str(args.steps),
@@ -54,7 +54,36 @@ def main():
                    help="fusion thresholds to try (MiB)")
     p.add_argument("--cycles", default="0.5,1,2", help="cycle times (ms)")
     p.add_argument("--out", default=None, help="write results JSON here")
+    p.add_argument("--curve", action="store_true",
+                   help="scaling mode: run N=1,2,4,8 at fixed knobs and "
+                        "print the whole-job curve + efficiency (VERDICT r1 "
+                        "item 1b); knobs come from the first values of "
+                        "--thresholds/--cycles")
     args = p.parse_args()
+
+    if args.curve:
+        thr = int(args.thresholds.split(",")[0])
+        cyc = float(args.cycles.split(",")[0])
+        import torch
+        ndev = torch.cuda.device_count()
+        curve = []
+        for n in (1, 2, 4, 8):
+            if n > ndev:
+                print(f"  n={n}: skipped ({ndev} GPUs visible)")
+                continue
+            args.gpus = n
+            r = run_one(args, thr, cyc)
+            if r:
+                curve.append({"n": n, "value": r["value"],
+                              "ms_per_step": r["ms_per_step"]})
+                base = curve[0]["value"]
+                eff = r["value"] / (base * n) if base else 0.0
+                print(f"  n={n}: {r['value']:.1f} {r['unit']} "
+                      f"(eff {eff * 100:.1f}%)")
+        if args.out:
+            with open(args.out, "w") as f:
+                json.dump(curve, f, indent=1)
+        return
 
     results = []
     for thr in [int(x) for x in args.thresholds.split(",")]:
