@@ -369,27 +369,19 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_v2_kernel(
     if (t + 1 < ntiles) load_tile(k0 + kBc2, &kv4, &vv4);
 
     // ---- S^T = K Q^T (A = K rows, B = Q cols) ----
-    // Two independent accumulator chains (even/odd d-chunks) halve the
-    // dependent-MFMA stall of one 8-deep chain; merged with 16 VALU adds.
-    frag16 st, st1;
+    // (One accumulator chain: a split even/odd two-chain variant measured
+    // 356 vs 411 TF — the +16 VGPRs dropped occupancy 3 -> 2.)
+    frag16 st;
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      st[r] = 0.f;
-      st1[r] = 0.f;
-    }
+    for (int r = 0; r < 16; ++r) st[r] = 0.f;
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int c = 0; c < kD / 16; c += 2) {
+    for (int c = 0; c < kD / 16; ++c) {
       frag_ab kf = *reinterpret_cast<const frag_ab*>(
           &k_lds[buf][col][c * 16 + half * 8]);
       st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[c], st, 0, 0, 0);
-      frag_ab kf1 = *reinterpret_cast<const frag_ab*>(
-          &k_lds[buf][col][(c + 1) * 16 + half * 8]);
-      st1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf1, qf[c + 1], st1, 0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
-#pragma unroll
-    for (int r = 0; r < 16; ++r) st[r] += st1[r];
 
     // ---- masking + tile max (raw-score domain) ----
     const bool full_tile = causal ? (k0 + kBc2 - 1 <= q0w) : (k0 + kBc2 <= S);
@@ -439,6 +431,10 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_v2_kernel(
     frag_ab pa1, pa2;
     p16_to_afrags(p, &pa1, &pa2);
 
+    // Write tile t+1 into the other buffer NOW so the LDS stores dual-issue
+    // under the PV MFMAs (safe: nobody reads buf^1 until after the barrier).
+    if (t + 1 < ntiles) write_tile(buf ^ 1, kv4, vv4);
+
     // ---- O += P V (B-frag: 8 consecutive keys from transposed V) ----
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -453,8 +449,6 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_v2_kernel(
                                                          0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
-
-    if (t + 1 < ntiles) write_tile(buf ^ 1, kv4, vv4);
     __syncthreads();  // publish buf^1; also fences buf re-use next round
   }
 
@@ -964,7 +958,6 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dkdv_v2_kernel(
             __builtin_amdgcn_mfma_f32_32x32x16_bf16(sa2, q2, dkacc[cc], 0, 0, 0);
       }
       __builtin_amdgcn_s_setprio(0);
-
       if (it + 1 < niter) write_qdo(bu ^ 1);
       __syncthreads();
     }
@@ -1091,6 +1084,8 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dq_v2_kernel(
       ds[r] = dead ? 0.f : pT * (dpt[r] - dlt_q) * scale;
     }
 
+    if (t + 1 < ntiles) write_kv(buf ^ 1);
+
     // ---- dQ += dS K (B from transposed K image) ----
     frag_ab a1, a2;
     p16_to_afrags(ds, &a1, &a2);
@@ -1105,8 +1100,6 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dq_v2_kernel(
       dqacc[cc] =
           __builtin_amdgcn_mfma_f32_32x32x16_bf16(a2, k2, dqacc[cc], 0, 0, 0);
     }
-
-    if (t + 1 < ntiles) write_kv(buf ^ 1);
     __syncthreads();
   }
 
