@@ -240,8 +240,8 @@ __global__ void fa_fwd_kernel(const bf16* __restrict__ q,
 
 using frag16 = __attribute__((ext_vector_type(16))) float;  // 32x32 C/D
 
-constexpr int kBq2 = 256;   // q rows per block
-constexpr int kW2 = 8;      // waves per block
+constexpr int kBq2 = 256;   // q rows per block (8-wave variant)
+constexpr int kW2 = 8;      // max waves per block (red_lds sizing)
 constexpr int kBr2 = 32;    // q rows per wave
 constexpr int kBc2 = 32;    // keys per tile
 constexpr int kKPad = 8;    // bf16 pad per K-LDS row
@@ -293,16 +293,19 @@ __device__ __forceinline__ void p16_to_afrags(const float p[16], frag_ab* a1,
   u2[0] = r2[0]; u2[1] = r3[0]; u2[2] = r2[1]; u2[3] = r3[1];
 }
 
-__global__ __launch_bounds__(512, 1) void fa_fwd_v2_kernel(
+template <int NW>  // waves per block (8 -> 256 q rows, 4 -> 128)
+__global__ __launch_bounds__(NW * 64, 1) void fa_fwd_v2_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, bf16* __restrict__ o, float* __restrict__ lse,
     int B, int HQ, int HK, int S, float scale, int causal) {
+  constexpr int kThreads = NW * 64;
+  constexpr int kRep = (kBc2 * kD / 8) / kThreads;  // staged chunks/thread
   const int lane = threadIdx.x % 64;
   const int wave = threadIdx.x / 64;
-  const int idx = threadIdx.x;           // 0..511
+  const int idx = threadIdx.x;
   const int col = lane & 31;             // q row (S^T C col) / d col (O)
   const int half = lane >> 5;
-  const int qb = blockIdx.x * kBq2;
+  const int qb = blockIdx.x * (NW * kBr2);
   const int h = blockIdx.y;
   const int b = blockIdx.z;
   const int hk = h / (HQ / HK);
@@ -331,42 +334,54 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_v2_kernel(
     for (int r = 0; r < 16; ++r) oacc[cc][r] = 0.f;
   float m_run = -1e30f, l_run = 0.f;  // row stats for q row `qi` (raw scores)
 
-  // V staging map: vr = idx&31 (key), dc = (idx>>5)*8 (d chunk) — keeps the
-  // transposing ds_write_b16s spread across banks (see round-2 notes).
-  const int vr = idx & 31, vdc = (idx >> 5) * 8;
-  // K staging map: kr = idx>>4 (row), dc = (idx&15)*8 — coalesced + b128 write.
-  const int kr = idx >> 4, kdc = (idx & 15) * 8;
-
-  const int kend = causal ? (qb + kBq2 < S ? qb + kBq2 : S) : S;
+  // Staging maps (kRep chunks per thread; chunk id cid = idx + r*kThreads):
+  //   K: row cid>>4, d (cid&15)*8 — coalesced loads + b128 LDS writes.
+  //   V: key cid&31, d (cid>>5)*8 — keeps the transposing ds_write_b16s
+  //      spread across banks (see round-2 notes).
+  const int kend = causal ? (qb + NW * kBr2 < S ? qb + NW * kBr2 : S) : S;
   const int ntiles = (kend + kBc2 - 1) / kBc2;
 
   // T14 issue-early/write-late with double-buffered LDS: tile t+1's global
   // loads are issued before tile t's compute, the LDS write lands after it,
   // and ONE barrier per tile publishes the next buffer.
   auto load_tile = [&](int k0, float4* kv4, float4* vv4) {
-    *kv4 = float4{0.f, 0.f, 0.f, 0.f};
-    *vv4 = float4{0.f, 0.f, 0.f, 0.f};
-    if (k0 + kr < S)
-      *kv4 = *reinterpret_cast<const float4*>(k + (kbase + k0 + kr) * kD + kdc);
-    if (k0 + vr < S)
-      *vv4 = *reinterpret_cast<const float4*>(v + (kbase + k0 + vr) * kD + vdc);
-  };
-  auto write_tile = [&](int buf, const float4& kv4, const float4& vv4) {
-    *reinterpret_cast<float4*>(&k_lds[buf][kr][kdc]) = kv4;
-    const bf16* vb = reinterpret_cast<const bf16*>(&vv4);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) v_lds[buf][vdc + j][vr] = vb[j];
+    for (int r = 0; r < kRep; ++r) {
+      int cid = idx + r * kThreads;
+      int kr = cid >> 4, kdc = (cid & 15) * 8;
+      int vr = cid & 31, vdc = (cid >> 5) * 8;
+      kv4[r] = float4{0.f, 0.f, 0.f, 0.f};
+      vv4[r] = float4{0.f, 0.f, 0.f, 0.f};
+      if (k0 + kr < S)
+        kv4[r] =
+            *reinterpret_cast<const float4*>(k + (kbase + k0 + kr) * kD + kdc);
+      if (k0 + vr < S)
+        vv4[r] =
+            *reinterpret_cast<const float4*>(v + (kbase + k0 + vr) * kD + vdc);
+    }
+  };
+  auto write_tile = [&](int buf, const float4* kv4, const float4* vv4) {
+#pragma unroll
+    for (int r = 0; r < kRep; ++r) {
+      int cid = idx + r * kThreads;
+      int kr = cid >> 4, kdc = (cid & 15) * 8;
+      int vr = cid & 31, vdc = (cid >> 5) * 8;
+      *reinterpret_cast<float4*>(&k_lds[buf][kr][kdc]) = kv4[r];
+      const bf16* vb = reinterpret_cast<const bf16*>(&vv4[r]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v_lds[buf][vdc + j][vr] = vb[j];
+    }
   };
 
-  float4 kv4, vv4;
-  load_tile(0, &kv4, &vv4);
+  float4 kv4[kRep], vv4[kRep];
+  load_tile(0, kv4, vv4);
   write_tile(0, kv4, vv4);
   __syncthreads();
 
   for (int t = 0; t < ntiles; ++t) {
     const int k0 = t * kBc2;
     const int buf = t & 1;
-    if (t + 1 < ntiles) load_tile(k0 + kBc2, &kv4, &vv4);
+    if (t + 1 < ntiles) load_tile(k0 + kBc2, kv4, vv4);
 
     // ---- S^T = K Q^T (A = K rows, B = Q cols) ----
     // (One accumulator chain: a split even/odd two-chain variant measured
@@ -431,11 +446,10 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_v2_kernel(
     frag_ab pa1, pa2;
     p16_to_afrags(p, &pa1, &pa2);
 
-    // Write tile t+1 into the other buffer NOW so the LDS stores dual-issue
-    // under the PV MFMAs (safe: nobody reads buf^1 until after the barrier).
-    if (t + 1 < ntiles) write_tile(buf ^ 1, kv4, vv4);
-
     // ---- O += P V (B-frag: 8 consecutive keys from transposed V) ----
+    // (write_tile placed AFTER PV: hoisting it before the cluster measured
+    // 384 vs 411 TF — hipcc's vmcnt(0) at the ds_write drains the t+1
+    // loads mid-compute, the failure mode the guide's T14 notes warn of.)
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int cc = 0; cc < kD / 32; ++cc) {
@@ -449,6 +463,8 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_v2_kernel(
                                                          0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
+
+    if (t + 1 < ntiles) write_tile(buf ^ 1, kv4, vv4);
     __syncthreads();  // publish buf^1; also fences buf re-use next round
   }
 
@@ -849,20 +865,18 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dkdv_v2_kernel(
   const int niter = gqa * nq;
 
   // Flattened (q-head, q-tile) loop with T14 double-buffered staging.
-  // Prefetch holds the FIRST 16B of each row (T14); the adjacent second
-  // 16B lands on the same cache line and is loaded at write time.
-  float4 qa, da;
-  int64_t pre_row = 0, pre_stat = 0;
-  auto load_qdo = [&](int it) {
+  // No register-held prefetch: at the 256-VGPR cap the held float4s spill
+  // into the inner loop (24-56 B/lane measured), which costs more than the
+  // staging latency it hides. Loads happen in write_qdo, right before the
+  // ds_writes; the double-buffered images still keep one barrier per tile.
+  auto write_qdo = [&](int it, int bu) {
     int g = it / nq, q0 = q_lo + (it % nq) * 32;
     int64_t qb_ = (static_cast<int64_t>(b) * HQ + hkv * gqa + g) * S;
     int qr = q0 + qsr < S ? q0 + qsr : S - 1;
-    pre_row = qb_ + qr;
-    qa = *reinterpret_cast<const float4*>(q + pre_row * kD + qsd);
-    da = *reinterpret_cast<const float4*>(dout + pre_row * kD + qsd);
-    pre_stat = qb_ + (q0 + idx < S ? q0 + idx : S - 1);
-  };
-  auto write_qdo = [&](int bu) {
+    int64_t pre_row = qb_ + qr;
+    int64_t pre_stat = qb_ + (q0 + idx < S ? q0 + idx : S - 1);
+    float4 qa = *reinterpret_cast<const float4*>(q + pre_row * kD + qsd);
+    float4 da = *reinterpret_cast<const float4*>(dout + pre_row * kD + qsd);
     float4 qb2 = *reinterpret_cast<const float4*>(q + pre_row * kD + qsd + 8);
     float4 db = *reinterpret_cast<const float4*>(dout + pre_row * kD + qsd + 8);
     float lv = 0.f, dvl = 0.f;
@@ -890,17 +904,13 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dkdv_v2_kernel(
       dlt_lds[bu][idx] = dvl;
     }
   };
-  if (niter > 0) {
-    load_qdo(0);
-    write_qdo(0);
-  }
+  if (niter > 0) write_qdo(0, 0);
   __syncthreads();
 
   for (int it = 0; it < niter; ++it) {
     {
       const int q0 = q_lo + (it % nq) * 32;
       const int bu = it & 1;
-      if (it + 1 < niter) load_qdo(it + 1);
 
       // ---- S = Q K^T and dP = dO V^T (C: col = k, rows = 16 q) ----
       frag16 st, dp;
@@ -958,7 +968,7 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dkdv_v2_kernel(
             __builtin_amdgcn_mfma_f32_32x32x16_bf16(sa2, q2, dkacc[cc], 0, 0, 0);
       }
       __builtin_amdgcn_s_setprio(0);
-      if (it + 1 < niter) write_qdo(bu ^ 1);
+      if (it + 1 < niter) write_qdo(it + 1, bu ^ 1);
       __syncthreads();
     }
   }
@@ -1175,8 +1185,19 @@ void launch_fa_fwd(const void* q, const void* k, const void* v, void* o,
                        B, HQ, HK, S, scale, causal ? 1 : 0);
     return;
   }
-  dim3 grid((S + kBq2 - 1) / kBq2, HQ, B);
-  hipLaunchKernelGGL(fa_fwd_v2_kernel, grid, 512, 0, stream,
+  const char* fw = getenv("AXON_FA_W");
+  if (fw && fw[0] == '8') {
+    dim3 grid((S + kBq2 - 1) / kBq2, HQ, B);
+    hipLaunchKernelGGL((fa_fwd_v2_kernel<8>), grid, 512, 0, stream,
+                       static_cast<const bf16*>(q), static_cast<const bf16*>(k),
+                       static_cast<const bf16*>(v), static_cast<bf16*>(o), lse,
+                       B, HQ, HK, S, scale, causal ? 1 : 0);
+    return;
+  }
+  // default: 4-wave blocks (faster barrier convergence, halved causal
+  // diagonal waste, 3 blocks/CU interleave)
+  dim3 grid((S + 127) / 128, HQ, B);
+  hipLaunchKernelGGL((fa_fwd_v2_kernel<4>), grid, 256, 0, stream,
                      static_cast<const bf16*>(q), static_cast<const bf16*>(k),
                      static_cast<const bf16*>(v), static_cast<bf16*>(o), lse,
                      B, HQ, HK, S, scale, causal ? 1 : 0);
