@@ -180,7 +180,7 @@ def bench(args):
     device, dtype = "cuda", torch.bfloat16
     q, k, v = make_case("random", args.batch, args.heads, args.kv_heads,
                         args.seq, args.head_dim, device, dtype)
-    fwd_only = args.impl == "axon"  # axon path times the raw fwd kernel
+    fwd_only = args.impl == "axon" or args.fwd_only
     if not fwd_only:
         q = q.requires_grad_(True)
     for _ in range(3):  # warmup
@@ -217,6 +217,7 @@ def main():
     p.add_argument("--head-dim", type=int, default=128)
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--bench", action="store_true")
+    p.add_argument("--fwd-only", action="store_true")
     p.add_argument("--bench-bwd", action="store_true")
     p.add_argument("--check-bwd", action="store_true")
     args = p.parse_args()
