@@ -389,14 +389,12 @@ __global__ __launch_bounds__(NW * 64, 1) void fa_fwd_v2_kernel(
     frag16 st;
 #pragma unroll
     for (int r = 0; r < 16; ++r) st[r] = 0.f;
-    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int c = 0; c < kD / 16; ++c) {
       frag_ab kf = *reinterpret_cast<const frag_ab*>(
           &k_lds[buf][col][c * 16 + half * 8]);
       st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[c], st, 0, 0, 0);
     }
-    __builtin_amdgcn_s_setprio(0);
 
     // ---- masking + tile max (raw-score domain) ----
     const bool full_tile = causal ? (k0 + kBc2 - 1 <= q0w) : (k0 + kBc2 <= S);
@@ -449,8 +447,9 @@ __global__ __launch_bounds__(NW * 64, 1) void fa_fwd_v2_kernel(
     // ---- O += P V (B-frag: 8 consecutive keys from transposed V) ----
     // (write_tile placed AFTER PV: hoisting it before the cluster measured
     // 384 vs 411 TF — hipcc's vmcnt(0) at the ds_write drains the t+1
-    // loads mid-compute, the failure mode the guide's T14 notes warn of.)
-    __builtin_amdgcn_s_setprio(1);
+    // loads mid-compute, the failure mode the guide's T14 notes warn of.
+    // s_setprio around the MFMA clusters also measured -5% here — without
+    // a T3-style phase split it only perturbs the scheduler.)
 #pragma unroll
     for (int cc = 0; cc < kD / 32; ++cc) {
       frag_ab v1 = *reinterpret_cast<const frag_ab*>(
@@ -462,7 +461,6 @@ __global__ __launch_bounds__(NW * 64, 1) void fa_fwd_v2_kernel(
       oacc[cc] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa2, v2, oacc[cc],
                                                          0, 0, 0);
     }
-    __builtin_amdgcn_s_setprio(0);
 
     if (t + 1 < ntiles) write_tile(buf ^ 1, kv4, vv4);
     __syncthreads();  // publish buf^1; also fences buf re-use next round
@@ -986,6 +984,265 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dkdv_v2_kernel(
     }
 }
 
+// Split dV / dK kernels: the fused dkdv kernel sits at the 256-VGPR cap
+// with 2.6x worse per-MFMA efficiency than dq (measured 1.71 ms vs an
+// expected ~1.3x of dq's 0.49 ms at bs1 h32 s4096). Recomputing S once
+// more (+25% MFMA) buys a light dV kernel (2 GEMMs, occupancy 3) and a
+// medium dK kernel (3 GEMMs) with half the transposed-image staging each.
+
+__global__ __launch_bounds__(256, 2) void fa_bwd_dv_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ k,
+    const bf16* __restrict__ dout, const float* __restrict__ lse,
+    bf16* __restrict__ dv, int B, int HQ, int HK, int S, float scale,
+    int causal) {
+  const int lane = threadIdx.x % 64;
+  const int wave = threadIdx.x / 64;
+  const int idx = threadIdx.x;
+  const int col = lane & 31;
+  const int half = lane >> 5;
+  const int kb0 = blockIdx.x * kBK;
+  const int hkv = blockIdx.y;
+  const int b = blockIdx.z;
+  const int gqa = HQ / HK;
+  const int kw0 = kb0 + wave * 32;
+
+  __shared__ bf16 q_row[2][32][kD + kKPad];
+  __shared__ bf16 do_tr[2][kD][32 + kVPad];
+  __shared__ float lse_lds[2][32];
+
+  const int64_t kbase = (static_cast<int64_t>(b) * HK + hkv) * S;
+  const int kc = kw0 + col;
+  const int krow_g = kc < S ? kc : S - 1;
+  frag_ab kreg[kD / 16];
+#pragma unroll
+  for (int c = 0; c < kD / 16; ++c)
+    kreg[c] = *reinterpret_cast<const frag_ab*>(
+        k + (kbase + krow_g) * kD + c * 16 + half * 8);
+
+  frag16 dvacc[kD / 32];
+#pragma unroll
+  for (int cc = 0; cc < kD / 32; ++cc)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dvacc[cc][r] = 0.f;
+
+  const int qsr = idx & 31, qsd = (idx >> 5) * 16;
+  const int q_lo = causal ? kb0 : 0;
+  const int nq = (S - q_lo + 31) / 32;
+  const int niter = gqa * nq;
+
+  auto stage = [&](int it, int bu) {
+    int g = it / nq, q0 = q_lo + (it % nq) * 32;
+    int64_t qb_ = (static_cast<int64_t>(b) * HQ + hkv * gqa + g) * S;
+    int qr = q0 + qsr < S ? q0 + qsr : S - 1;
+    float4 qa = *reinterpret_cast<const float4*>(q + (qb_ + qr) * kD + qsd);
+    float4 qb2 =
+        *reinterpret_cast<const float4*>(q + (qb_ + qr) * kD + qsd + 8);
+    float4 da = *reinterpret_cast<const float4*>(dout + (qb_ + qr) * kD + qsd);
+    float4 db =
+        *reinterpret_cast<const float4*>(dout + (qb_ + qr) * kD + qsd + 8);
+    float lv = 0.f;
+    if (idx < 32) lv = lse[qb_ + (q0 + idx < S ? q0 + idx : S - 1)];
+    *reinterpret_cast<float4*>(&q_row[bu][qsr][qsd]) = qa;
+    *reinterpret_cast<float4*>(&q_row[bu][qsr][qsd + 8]) = qb2;
+    const bf16* dp_ = reinterpret_cast<const bf16*>(&da);
+    const bf16* dp2 = reinterpret_cast<const bf16*>(&db);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      do_tr[bu][qsd + j][qsr] = dp_[j];
+      do_tr[bu][qsd + 8 + j][qsr] = dp2[j];
+    }
+    if (idx < 32) lse_lds[bu][idx] = lv;
+  };
+  if (niter > 0) stage(0, 0);
+  __syncthreads();
+
+  for (int it = 0; it < niter; ++it) {
+    const int q0 = q_lo + (it % nq) * 32;
+    const int bu = it & 1;
+
+    frag16 st;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) st[r] = 0.f;
+#pragma unroll
+    for (int c = 0; c < kD / 16; ++c) {
+      frag_ab qaf = *reinterpret_cast<const frag_ab*>(
+          &q_row[bu][col][c * 16 + half * 8]);
+      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qaf, kreg[c], st, 0, 0, 0);
+    }
+
+    float pv[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int qroff = (r & 3) + 8 * (r >> 2) + 4 * half;
+      int qi2 = q0 + qroff;
+      bool dead = kc >= S || qi2 >= S || (causal && kc > qi2);
+      pv[r] = dead ? 0.f : __expf(scale * st[r] - lse_lds[bu][qroff]);
+    }
+    frag_ab pa1, pa2;
+    p16_to_afrags(pv, &pa1, &pa2);
+#pragma unroll
+    for (int cc = 0; cc < kD / 32; ++cc) {
+      frag_ab d1 = *reinterpret_cast<const frag_ab*>(
+          &do_tr[bu][cc * 32 + col][half * 8]);
+      frag_ab d2 = *reinterpret_cast<const frag_ab*>(
+          &do_tr[bu][cc * 32 + col][16 + half * 8]);
+      dvacc[cc] =
+          __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa1, d1, dvacc[cc], 0, 0, 0);
+      dvacc[cc] =
+          __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa2, d2, dvacc[cc], 0, 0, 0);
+    }
+
+    if (it + 1 < niter) stage(it + 1, bu ^ 1);
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int cc = 0; cc < kD / 32; ++cc)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int ki = kw0 + (r & 3) + 8 * (r >> 2) + 4 * half;
+      if (ki < S)
+        dv[(kbase + ki) * kD + cc * 32 + col] = __float2bfloat16(dvacc[cc][r]);
+    }
+}
+
+__global__ __launch_bounds__(256, 2) void fa_bwd_dk_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ k,
+    const bf16* __restrict__ v, const bf16* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    bf16* __restrict__ dk, int B, int HQ, int HK, int S, float scale,
+    int causal) {
+  const int lane = threadIdx.x % 64;
+  const int wave = threadIdx.x / 64;
+  const int idx = threadIdx.x;
+  const int col = lane & 31;
+  const int half = lane >> 5;
+  const int kb0 = blockIdx.x * kBK;
+  const int hkv = blockIdx.y;
+  const int b = blockIdx.z;
+  const int gqa = HQ / HK;
+  const int kw0 = kb0 + wave * 32;
+
+  __shared__ bf16 q_row[2][32][kD + kKPad];
+  __shared__ bf16 do_row[2][32][kD + kKPad];
+  __shared__ bf16 q_tr[2][kD][32 + kVPad];
+  __shared__ float lse_lds[2][32], dlt_lds[2][32];
+
+  const int64_t kbase = (static_cast<int64_t>(b) * HK + hkv) * S;
+  const int kc = kw0 + col;
+  const int krow_g = kc < S ? kc : S - 1;
+  frag_ab kreg[kD / 16], vreg[kD / 16];
+#pragma unroll
+  for (int c = 0; c < kD / 16; ++c) {
+    kreg[c] = *reinterpret_cast<const frag_ab*>(
+        k + (kbase + krow_g) * kD + c * 16 + half * 8);
+    vreg[c] = *reinterpret_cast<const frag_ab*>(
+        v + (kbase + krow_g) * kD + c * 16 + half * 8);
+  }
+
+  frag16 dkacc[kD / 32];
+#pragma unroll
+  for (int cc = 0; cc < kD / 32; ++cc)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dkacc[cc][r] = 0.f;
+
+  const int qsr = idx & 31, qsd = (idx >> 5) * 16;
+  const int q_lo = causal ? kb0 : 0;
+  const int nq = (S - q_lo + 31) / 32;
+  const int niter = gqa * nq;
+
+  auto stage = [&](int it, int bu) {
+    int g = it / nq, q0 = q_lo + (it % nq) * 32;
+    int64_t qb_ = (static_cast<int64_t>(b) * HQ + hkv * gqa + g) * S;
+    int qr = q0 + qsr < S ? q0 + qsr : S - 1;
+    float4 qa = *reinterpret_cast<const float4*>(q + (qb_ + qr) * kD + qsd);
+    float4 qb2 =
+        *reinterpret_cast<const float4*>(q + (qb_ + qr) * kD + qsd + 8);
+    float4 da = *reinterpret_cast<const float4*>(dout + (qb_ + qr) * kD + qsd);
+    float4 db =
+        *reinterpret_cast<const float4*>(dout + (qb_ + qr) * kD + qsd + 8);
+    float lv = 0.f, dvl = 0.f;
+    if (idx < 32) {
+      int64_t st_ = qb_ + (q0 + idx < S ? q0 + idx : S - 1);
+      lv = lse[st_];
+      dvl = delta[st_];
+    }
+    *reinterpret_cast<float4*>(&q_row[bu][qsr][qsd]) = qa;
+    *reinterpret_cast<float4*>(&q_row[bu][qsr][qsd + 8]) = qb2;
+    *reinterpret_cast<float4*>(&do_row[bu][qsr][qsd]) = da;
+    *reinterpret_cast<float4*>(&do_row[bu][qsr][qsd + 8]) = db;
+    const bf16* qp = reinterpret_cast<const bf16*>(&qa);
+    const bf16* qp2 = reinterpret_cast<const bf16*>(&qb2);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      q_tr[bu][qsd + j][qsr] = qp[j];
+      q_tr[bu][qsd + 8 + j][qsr] = qp2[j];
+    }
+    if (idx < 32) {
+      lse_lds[bu][idx] = lv;
+      dlt_lds[bu][idx] = dvl;
+    }
+  };
+  if (niter > 0) stage(0, 0);
+  __syncthreads();
+
+  for (int it = 0; it < niter; ++it) {
+    const int q0 = q_lo + (it % nq) * 32;
+    const int bu = it & 1;
+
+    frag16 st, dp;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      st[r] = 0.f;
+      dp[r] = 0.f;
+    }
+#pragma unroll
+    for (int c = 0; c < kD / 16; ++c) {
+      frag_ab qaf = *reinterpret_cast<const frag_ab*>(
+          &q_row[bu][col][c * 16 + half * 8]);
+      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qaf, kreg[c], st, 0, 0, 0);
+      frag_ab daf = *reinterpret_cast<const frag_ab*>(
+          &do_row[bu][col][c * 16 + half * 8]);
+      dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(daf, vreg[c], dp, 0, 0, 0);
+    }
+
+    float ds[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int qroff = (r & 3) + 8 * (r >> 2) + 4 * half;
+      int qi2 = q0 + qroff;
+      bool dead = kc >= S || qi2 >= S || (causal && kc > qi2);
+      float pvr = dead ? 0.f : __expf(scale * st[r] - lse_lds[bu][qroff]);
+      ds[r] = dead ? 0.f : pvr * (dp[r] - dlt_lds[bu][qroff]) * scale;
+    }
+    frag_ab sa1, sa2;
+    p16_to_afrags(ds, &sa1, &sa2);
+#pragma unroll
+    for (int cc = 0; cc < kD / 32; ++cc) {
+      frag_ab q1 = *reinterpret_cast<const frag_ab*>(
+          &q_tr[bu][cc * 32 + col][half * 8]);
+      frag_ab q2 = *reinterpret_cast<const frag_ab*>(
+          &q_tr[bu][cc * 32 + col][16 + half * 8]);
+      dkacc[cc] =
+          __builtin_amdgcn_mfma_f32_32x32x16_bf16(sa1, q1, dkacc[cc], 0, 0, 0);
+      dkacc[cc] =
+          __builtin_amdgcn_mfma_f32_32x32x16_bf16(sa2, q2, dkacc[cc], 0, 0, 0);
+    }
+
+    if (it + 1 < niter) stage(it + 1, bu ^ 1);
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int cc = 0; cc < kD / 32; ++cc)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int ki = kw0 + (r & 3) + 8 * (r >> 2) + 4 * half;
+      if (ki < S)
+        dk[(kbase + ki) * kD + cc * 32 + col] = __float2bfloat16(dkacc[cc][r]);
+    }
+}
+
 __global__ __launch_bounds__(256, 2) void fa_bwd_dq_v2_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, const bf16* __restrict__ dout,
@@ -1156,12 +1413,27 @@ void launch_fa_bwd(const void* q, const void* k, const void* v,
     return;
   }
   dim3 gkv((S + kBK - 1) / kBK, HK, B);
-  hipLaunchKernelGGL(fa_bwd_dkdv_v2_kernel, gkv, 256, 0, stream,
-                     static_cast<const bf16*>(q), static_cast<const bf16*>(k),
-                     static_cast<const bf16*>(v),
-                     static_cast<const bf16*>(dout), lse, delta,
-                     static_cast<bf16*>(dk), static_cast<bf16*>(dv), B, HQ,
-                     HK, S, scale, causal ? 1 : 0);
+  const char* fb = getenv("AXON_FA_BWD");
+  if (fb && fb[0] == 'f') {  // fused dkdv variant (measured slower)
+    hipLaunchKernelGGL(fa_bwd_dkdv_v2_kernel, gkv, 256, 0, stream,
+                       static_cast<const bf16*>(q), static_cast<const bf16*>(k),
+                       static_cast<const bf16*>(v),
+                       static_cast<const bf16*>(dout), lse, delta,
+                       static_cast<bf16*>(dk), static_cast<bf16*>(dv), B, HQ,
+                       HK, S, scale, causal ? 1 : 0);
+  } else {
+    hipLaunchKernelGGL(fa_bwd_dv_kernel, gkv, 256, 0, stream,
+                       static_cast<const bf16*>(q), static_cast<const bf16*>(k),
+                       static_cast<const bf16*>(dout), lse,
+                       static_cast<bf16*>(dv), B, HQ, HK, S, scale,
+                       causal ? 1 : 0);
+    hipLaunchKernelGGL(fa_bwd_dk_kernel, gkv, 256, 0, stream,
+                       static_cast<const bf16*>(q), static_cast<const bf16*>(k),
+                       static_cast<const bf16*>(v),
+                       static_cast<const bf16*>(dout), lse, delta,
+                       static_cast<bf16*>(dk), B, HQ, HK, S, scale,
+                       causal ? 1 : 0);
+  }
   dim3 gq((S + kBK - 1) / kBK, HQ, B);
   hipLaunchKernelGGL(fa_bwd_dq_v2_kernel, gq, 256, 0, stream,
                      static_cast<const bf16*>(q), static_cast<const bf16*>(k),
@@ -1186,18 +1458,18 @@ void launch_fa_fwd(const void* q, const void* k, const void* v, void* o,
     return;
   }
   const char* fw = getenv("AXON_FA_W");
-  if (fw && fw[0] == '8') {
-    dim3 grid((S + kBq2 - 1) / kBq2, HQ, B);
-    hipLaunchKernelGGL((fa_fwd_v2_kernel<8>), grid, 512, 0, stream,
+  if (fw && fw[0] == '4') {
+    // 4-wave variant: measured 337 vs 390 TF (the +16 staging VGPRs drop
+    // occupancy to 2) — kept selectable for re-measurement.
+    dim3 grid((S + 127) / 128, HQ, B);
+    hipLaunchKernelGGL((fa_fwd_v2_kernel<4>), grid, 256, 0, stream,
                        static_cast<const bf16*>(q), static_cast<const bf16*>(k),
                        static_cast<const bf16*>(v), static_cast<bf16*>(o), lse,
                        B, HQ, HK, S, scale, causal ? 1 : 0);
     return;
   }
-  // default: 4-wave blocks (faster barrier convergence, halved causal
-  // diagonal waste, 3 blocks/CU interleave)
-  dim3 grid((S + 127) / 128, HQ, B);
-  hipLaunchKernelGGL((fa_fwd_v2_kernel<4>), grid, 256, 0, stream,
+  dim3 grid((S + kBq2 - 1) / kBq2, HQ, B);
+  hipLaunchKernelGGL((fa_fwd_v2_kernel<8>), grid, 512, 0, stream,
                      static_cast<const bf16*>(q), static_cast<const bf16*>(k),
                      static_cast<const bf16*>(v), static_cast<bf16*>(o), lse,
                      B, HQ, HK, S, scale, causal ? 1 : 0);
