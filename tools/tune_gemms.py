@@ -28,14 +28,60 @@ def main():
     p.add_argument("--max-ms", type=int, default=20,
                    help="max tuning duration per GEMM candidate (ms)")
     p.add_argument("--max-iter", type=int, default=30)
+    p.add_argument("--record", default=None,
+                   help="record untuned GEMM shapes to this file instead of "
+                        "tuning inline (fast; chunk-tune them later with "
+                        "--tune-file)")
+    p.add_argument("--tune-file", default=None,
+                   help="offline-tune the shapes recorded in this file "
+                        "(optionally a LINE0:LINE1 slice via --lines) and "
+                        "merge results into --out")
+    p.add_argument("--lines", default=None, help="L0:L1 slice of --tune-file")
     args = p.parse_args()
 
     import torch
 
     assert torch.cuda.is_available(), "tuning needs a GPU"
     t = torch.cuda.tunable
+
+    if args.tune_file:
+        # Offline per-shape tuning: bounded per shape, results appended to
+        # the results csv torch maintains, then merged into --out.
+        src = args.tune_file
+        if args.lines:
+            l0, l1 = (int(x) for x in args.lines.split(":"))
+            part = "/tmp/untuned_part.csv"
+            with open(src) as f, open(part, "w") as g:
+                for i, line in enumerate(f):
+                    if l0 <= i < l1:
+                        g.write(line)
+            src = part
+        t.enable(True)
+        t.tuning_enable(True)
+        t.set_max_tuning_duration(args.max_ms)
+        t.set_max_tuning_iterations(args.max_iter)
+        work = "/tmp/tunableop_work.csv"
+        out = os.path.abspath(args.out)
+        if os.path.exists(out):
+            shutil.copy(out, work)
+            t.read_file(work)
+        t.tune_gemm_in_file(src)
+        t.write_file(work)
+        shutil.copy(work, out)
+        with open(out) as f:
+            n = sum(1 for line in f if line.startswith("GemmTunableOp"))
+        print(f"[tune] wrote {out}: {n} tuned GEMM entries", flush=True)
+        return
+
+    if args.record:
+        t.enable(True)
+        t.tuning_enable(False)
+        t.record_untuned_enable(True)
+        os.environ["PYTORCH_TUNABLEOP_UNTUNED_FILENAME"] = args.record
+
     t.enable(True)
-    t.tuning_enable(True)
+    if not args.record:
+        t.tuning_enable(True)
     t.set_max_tuning_duration(args.max_ms)
     t.set_max_tuning_iterations(args.max_iter)
     out = os.path.abspath(args.out)
@@ -61,19 +107,29 @@ def main():
     else:
         _, _, step, *_ = build_llama(bargs, device, torch.bfloat16, args.model)
     # bench's _load_tunableop() (called inside build_*) turns tuning off for
-    # deterministic runs — re-enable it for this tuning pass.
-    t.tuning_enable(True)
+    # deterministic runs — re-enable it for this tuning pass (not in record
+    # mode, where GEMMs must run untuned and only be logged).
+    if args.record:
+        t.tuning_enable(False)
+        t.record_untuned_enable(True)
+    else:
+        t.tuning_enable(True)
 
     for i in range(args.steps):
         step()
         torch.cuda.synchronize()
         print(f"[tune] step {i + 1}/{args.steps} done", flush=True)
 
-    t.write_file(work)
-    shutil.copy(work, out)
-    with open(out) as f:
-        n = sum(1 for line in f if line.startswith("GemmTunableOp"))
-    print(f"[tune] wrote {out}: {n} tuned GEMM entries", flush=True)
+    if args.record:
+        import glob as _glob
+        cands = _glob.glob(args.record + "*") + _glob.glob("tunableop_untuned*")
+        print(f"[tune] recorded untuned shapes: {cands}", flush=True)
+    else:
+        t.write_file(work)
+        shutil.copy(work, out)
+        with open(out) as f:
+            n = sum(1 for line in f if line.startswith("GemmTunableOp"))
+        print(f"[tune] wrote {out}: {n} tuned GEMM entries", flush=True)
     hvd.shutdown()
 
 
