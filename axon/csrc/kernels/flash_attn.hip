@@ -816,26 +816,27 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dkdv_v2_kernel(
     int S, float scale, int causal) {
   const int lane = threadIdx.x % 64;
   const int wave = threadIdx.x / 64;
-  const int idx = threadIdx.x;
+  const int idx = threadIdx.x;  // 0..255
   const int col = lane & 31;
   const int half = lane >> 5;
   const int kb0 = blockIdx.x * kBK;
   const int hkv = blockIdx.y;
   const int b = blockIdx.z;
   const int gqa = HQ / HK;
-  const int kw0 = kb0 + wave * 32;
+  const int kw0 = kb0 + wave * 32;  // this wave's k-tile base
 
-  // Row images are PAD-0 and rotation-swizzled so global_load_lds can fill
-  // them with zero VGPR cost: LDS slot (row, dc) holds global element
-  // (dc + 8*(row&15)) & 127 — rows land on distinct bank groups (2-way
-  // worst case) and every b128 read window stays contiguous.
-  __shared__ bf16 q_row[2][32][kD];
-  __shared__ bf16 do_row[2][32][kD];
+  __shared__ bf16 q_row[2][32][kD + kKPad];
   __shared__ bf16 q_tr[2][kD][32 + kVPad];
+  __shared__ bf16 do_row[2][32][kD + kKPad];
   __shared__ bf16 do_tr[2][kD][32 + kVPad];
   __shared__ float lse_lds[2][32], dlt_lds[2][32];
 
   const int64_t kbase = (static_cast<int64_t>(b) * HK + hkv) * S;
+
+  // This wave's K/V rows live in registers for the whole kernel: the lane
+  // layout of a B fragment (col = lane&31, k-dim = (lane>>5)*8+j) matches a
+  // per-lane row load, so no LDS image is needed (keeps LDS under 40 KB ->
+  // 2 blocks/CU instead of the 108 KB 1-block version).
   const int kc = kw0 + col;
   const int krow_g = kc < S ? kc : S - 1;
   frag_ab kreg[kD / 16], vreg[kD / 16];
@@ -856,55 +857,40 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dkdv_v2_kernel(
       dkacc[cc][r] = 0.f;
     }
 
-  const int qsr = idx & 31, qsd = (idx >> 5) * 16;  // tr-staging map
+  const int qsr = idx & 31, qsd = (idx >> 5) * 16;  // staging map
   const int q_lo = causal ? kb0 : 0;
   const int nq = (S - q_lo + 31) / 32;
   const int niter = gqa * nq;
 
-  // glds row staging: wave w fills rows [8w, 8w+8) of both row images
-  // (2 instructions of 64 lanes x 16B each per image). Source addresses
-  // carry the rotation swizzle; row clamping keeps reads in-bounds (the
-  // data of a clamped row is garbage but every consumer masks dead rows).
-  const int g_row = 8 * wave + (lane >> 4);      // rows 8w .. 8w+3 (+4 for j=1)
-  const int g_dc = (lane & 15) * 8;
-  auto glds_rows = [&](int it, int bu) {
+  // Flattened (q-head, q-tile) loop with T14 double-buffered staging.
+  // Half-prefetch (T14): the FIRST 16B of each row is held across the
+  // compute (hides most of the HBM latency); the adjacent second 16B is
+  // loaded at write time from the now-hot cache line. Full prefetch spills
+  // 56B/lane at the 256-VGPR cap; this costs 24B/lane (5 dwords), the
+  // best-measuring compromise.
+  float4 qa, da;
+  int64_t pre_row = 0, pre_stat = 0;
+  auto load_qdo = [&](int it) {
     int g = it / nq, q0 = q_lo + (it % nq) * 32;
     int64_t qb_ = (static_cast<int64_t>(b) * HQ + hkv * gqa + g) * S;
-#pragma unroll
-    for (int j = 0; j < 2; ++j) {
-      int row = g_row + 4 * j;
-      int qr = q0 + row < S ? q0 + row : S - 1;
-      int e = (g_dc + ((row & 15) << 3)) & 127;
-      const bf16* qsrc = q + (qb_ + qr) * kD + e;
-      const bf16* dsrc = dout + (qb_ + qr) * kD + e;
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) uint32_t*)qsrc,
-          (__attribute__((address_space(3))) uint32_t*)&q_row[bu][8 * wave + 4 * j][0],
-          16, 0, 0);
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) uint32_t*)dsrc,
-          (__attribute__((address_space(3))) uint32_t*)&do_row[bu][8 * wave + 4 * j][0],
-          16, 0, 0);
-    }
+    int qr = q0 + qsr < S ? q0 + qsr : S - 1;
+    pre_row = qb_ + qr;
+    pre_stat = qb_ + (q0 + idx < S ? q0 + idx : S - 1);
+    qa = *reinterpret_cast<const float4*>(q + pre_row * kD + qsd);
+    da = *reinterpret_cast<const float4*>(dout + pre_row * kD + qsd);
   };
-  // Transposed images + per-row stats, read back from the freshly glds'd
-  // row image (explicit vmcnt drain: HW does not order glds vs ds_read).
-  auto write_tr = [&](int it, int bu) {
-    int g = it / nq, q0 = q_lo + (it % nq) * 32;
-    int64_t qb_ = (static_cast<int64_t>(b) * HQ + hkv * gqa + g) * S;
+  auto write_qdo = [&](int bu) {
+    float4 qb2 = *reinterpret_cast<const float4*>(q + pre_row * kD + qsd + 8);
+    float4 db = *reinterpret_cast<const float4*>(dout + pre_row * kD + qsd + 8);
     float lv = 0.f, dvl = 0.f;
     if (idx < 32) {
-      int64_t st_ = qb_ + (q0 + idx < S ? q0 + idx : S - 1);
-      lv = lse[st_];
-      dvl = delta[st_];
+      lv = lse[pre_stat];
+      dvl = delta[pre_stat];
     }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    int rot = (qsr & 15) << 3;
-    int e0 = (qsd - rot) & 127, e1 = (qsd + 8 - rot) & 127;
-    float4 qa = *reinterpret_cast<const float4*>(&q_row[bu][qsr][e0]);
-    float4 qb2 = *reinterpret_cast<const float4*>(&q_row[bu][qsr][e1]);
-    float4 da = *reinterpret_cast<const float4*>(&do_row[bu][qsr][e0]);
-    float4 db = *reinterpret_cast<const float4*>(&do_row[bu][qsr][e1]);
+    *reinterpret_cast<float4*>(&q_row[bu][qsr][qsd]) = qa;
+    *reinterpret_cast<float4*>(&q_row[bu][qsr][qsd + 8]) = qb2;
+    *reinterpret_cast<float4*>(&do_row[bu][qsr][qsd]) = da;
+    *reinterpret_cast<float4*>(&do_row[bu][qsr][qsd + 8]) = db;
     const bf16* qp = reinterpret_cast<const bf16*>(&qa);
     const bf16* qp2 = reinterpret_cast<const bf16*>(&qb2);
     const bf16* dp_ = reinterpret_cast<const bf16*>(&da);
@@ -922,8 +908,8 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dkdv_v2_kernel(
     }
   };
   if (niter > 0) {
-    glds_rows(0, 0);
-    write_tr(0, 0);
+    load_qdo(0);
+    write_qdo(0);
   }
   __syncthreads();
 
@@ -931,7 +917,7 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dkdv_v2_kernel(
     {
       const int q0 = q_lo + (it % nq) * 32;
       const int bu = it & 1;
-      if (it + 1 < niter) glds_rows(it + 1, bu ^ 1);
+      if (it + 1 < niter) load_qdo(it + 1);
 
       // ---- S = Q K^T and dP = dO V^T (C: col = k, rows = 16 q) ----
       frag16 st, dp;
@@ -940,15 +926,17 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dkdv_v2_kernel(
         st[r] = 0.f;
         dp[r] = 0.f;
       }
-      const int rot_c = (col & 15) << 3;
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int c = 0; c < kD / 16; ++c) {
-        int e = (c * 16 + half * 8 - rot_c) & 127;
-        frag_ab qaf = *reinterpret_cast<const frag_ab*>(&q_row[bu][col][e]);
+        frag_ab qaf = *reinterpret_cast<const frag_ab*>(
+            &q_row[bu][col][c * 16 + half * 8]);
         st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qaf, kreg[c], st, 0, 0, 0);
-        frag_ab daf = *reinterpret_cast<const frag_ab*>(&do_row[bu][col][e]);
+        frag_ab daf = *reinterpret_cast<const frag_ab*>(
+            &do_row[bu][col][c * 16 + half * 8]);
         dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(daf, vreg[c], dp, 0, 0, 0);
       }
+      __builtin_amdgcn_s_setprio(0);
 
       // ---- P and dS (in-register; lse/delta via LDS broadcast) ----
       float pv[16], ds[16];
@@ -966,6 +954,7 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dkdv_v2_kernel(
       frag_ab pa1, pa2, sa1, sa2;
       p16_to_afrags(pv, &pa1, &pa2);
       p16_to_afrags(ds, &sa1, &sa2);
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int cc = 0; cc < kD / 32; ++cc) {
         frag_ab d1 = *reinterpret_cast<const frag_ab*>(
@@ -985,8 +974,8 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dkdv_v2_kernel(
         dkacc[cc] =
             __builtin_amdgcn_mfma_f32_32x32x16_bf16(sa2, q2, dkacc[cc], 0, 0, 0);
       }
-
-      if (it + 1 < niter) write_tr(it + 1, bu ^ 1);
+      __builtin_amdgcn_s_setprio(0);
+      if (it + 1 < niter) write_qdo(bu ^ 1);
       __syncthreads();
     }
   }
