@@ -81,3 +81,29 @@ def test_elastic_double_failure_shrinks(tmp_path):
     assert res[0]["size"] == 2 and res[1]["size"] == 2
     assert res[0]["step"] == 10 and res[1]["step"] == 10
     assert abs(res[0]["wsum"] - res[1]["wsum"]) < 1e-9
+
+
+def test_elastic_clean_exit_excluded_from_generation(tmp_path):
+    """A worker that exits 0 mid-job must be dropped from the next
+    generation (ADVICE r1 regression): its peers fail on the broken
+    channel, respawn, and finish as a 2-rank world — pre-fix they hung
+    waiting for the finished worker to re-join."""
+    out = str(tmp_path / f"el_{uuid.uuid4().hex[:8]}")
+    script3 = os.path.join(REPO, "tests", "elastic_script3.py")
+    cmd = [sys.executable, "-m", "axon.runner.run", "-np", "3",
+           "--elastic", "--no-pin", "-v", "--min-np", "2",
+           "--max-restarts", "1", sys.executable, script3, out]
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["OMP_NUM_THREADS"] = "1"
+    env["AXON_SHUTDOWN_TIMEOUT"] = "3"
+    r = subprocess.run(cmd, env=env, capture_output=True, text=True,
+                       timeout=150)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert os.path.exists(out + ".done1")  # the clean exit happened
+    for w in (0, 2):
+        with open(f"{out}.w{w}.json") as f:
+            res = json.load(f)
+        assert res["size"] == 2, res
+        assert res["step"] == 8, res
+    assert not os.path.exists(out + ".w1.json")
