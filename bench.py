@@ -225,13 +225,21 @@ def try_graph_step(opt, fwd, n, op, compression):
         torch.cuda.current_stream().wait_stream(side)
         opt.zero_grad(set_to_none=True)
         g = torch.cuda.CUDAGraph()
-        # thread_local: the engine's poller/background threads are idle but
-        # alive; don't let their stray stream queries poison the capture.
-        with torch.cuda.graph(g, capture_error_mode="thread_local"):
+        # Default ("global") capture mode on purpose: backward runs on the
+        # autograd worker thread, and only global mode guarantees its
+        # launches are captured rather than silently executed eagerly
+        # (measured: thread_local capture produced stale grads — the
+        # trajectory degenerated to weight-decay-only updates). The
+        # engine's controller/poller threads are idle during capture (cv
+        # waits, no HIP calls with nothing in flight), so they cannot
+        # poison a global capture.
+        with torch.cuda.graph(g):
             static_loss = fwd()
             static_loss.backward()
         grads = [p.grad for group in opt.param_groups
                  for p in group["params"] if p.grad is not None]
+        if not grads:
+            raise RuntimeError("capture produced no gradients")
 
         def gstep():
             g.replay()

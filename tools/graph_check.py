@@ -58,9 +58,31 @@ def main():
 
     # Graphed trajectory from identical init.
     model, opt, fwd = build()
+    # Reference grads at init (weights untouched: no step taken).
+    loss0 = fwd()
+    loss0.backward()
+    named = list(model.named_parameters())
+    probes = [named[0], named[len(named) // 2], named[-1]]
+    ref_grads = {n: p.grad.float().clone() for n, p in probes}
     gstep = bench.try_graph_step(opt, fwd, 1, None, None)
     if gstep is None:
         print("GRAPH-CHECK FAIL: capture unavailable")
+        sys.exit(1)
+    # try_graph_step ends with one comm-free warm replay, so grads now hold
+    # the replayed backward's output at init weights — they must match the
+    # eager grads computed above.
+    grad_bad = False
+    for n, p in probes:
+        got = p.grad.float()
+        ref = ref_grads[n]
+        rel = (got - ref).abs().max().item() / max(1e-6,
+                                                   ref.abs().max().item())
+        print(f"replayed-grad rel err [{n}]: {rel:.4e} "
+              f"(|ref|max {ref.abs().max().item():.3e})")
+        if rel > 5e-2:
+            grad_bad = True
+    if grad_bad:
+        print("GRAPH-CHECK FAIL: replayed grads != eager grads")
         sys.exit(1)
     graphed = []
     for _ in range(STEPS):
