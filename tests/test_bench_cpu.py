@@ -1,0 +1,43 @@
+"""bench.py contract smoke on CPU: ws=2 over the socket control plane.
+
+Runs the driver's entry point end-to-end (init, broadcast_parameters,
+hook-overlapped allreduce, barrier, max-over-ranks elapsed, one JSON
+line on rank 0) with two CPU ranks, torchrun-style env. AXON_GRAPH=1 is
+set on purpose: graph capture is GPU-only and must degrade to the eager
+path silently when no GPU is present.
+"""
+import json
+import os
+import subprocess
+import sys
+import uuid
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_bench_resnet_cpu_ws2():
+    port = str(20000 + (uuid.uuid4().int % 20000))
+    procs = []
+    for r in range(2):
+        env = os.environ.copy()
+        env.update({"RANK": str(r), "WORLD_SIZE": "2", "LOCAL_RANK": "0",
+                    "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": port,
+                    "AXON_GRAPH": "1"})
+        procs.append(subprocess.Popen(
+            [sys.executable, "bench.py", "--steps", "1", "--warmup", "1",
+             "--batch", "2", "--image-size", "64"],
+            cwd=REPO, env=env, stdout=subprocess.PIPE,
+            stderr=subprocess.STDOUT, text=True))
+    outs = []
+    for p in procs:
+        out, _ = p.communicate(timeout=600)
+        outs.append(out)
+        assert p.returncode == 0, f"bench rank failed:\n{out}"
+    payloads = [line for out in outs for line in out.splitlines()
+                if line.startswith("{")]
+    assert len(payloads) == 1, f"expected one rank-0 JSON line:\n{outs}"
+    rec = json.loads(payloads[0])
+    assert rec["n_gpus"] == 2
+    assert rec["value"] > 0
+    assert rec["config"]["parallelism"] == "dp2"
+    assert rec["config"]["global_batch"] == 4
