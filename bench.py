@@ -196,11 +196,11 @@ def try_graph_step(opt, fwd, n, op, compression):
     and the gradient allreduce stay OUTSIDE the graph: the engine's
     background RCCL cycle (host-thread driven) is not capturable, and the
     fused-optimizer launcher stages a transient pinned chunk table whose
-    captured H2D copy would dangle on replay. Backward allocates .grad
-    from the graph's private pool at capture, so grad addresses are stable
-    across replays and the post-replay grouped allreduce + fused step can
-    reuse them in place (never call zero_grad between replays — replayed
-    producers overwrite the same buffers).
+    captured H2D copy would dangle on replay. The captured backward's
+    grad outputs live in the graph's private pool with stable addresses,
+    so the post-replay grouped allreduce + fused step reuse them in place
+    (never call zero_grad between replays — replayed producers overwrite
+    the same buffers).
     """
     try:
         # Per-parameter overlap hooks would fire at capture time (host
@@ -214,32 +214,41 @@ def try_graph_step(opt, fwd, n, op, compression):
             h.remove()
         if hasattr(opt, "_hook_handles"):
             opt._hook_handles = []
-        # Grads must be None at capture so backward's AccumulateGrad
-        # allocates them from the graph pool (assign, not accumulate).
+        params = [p for group in opt.param_groups
+                  for p in group["params"] if p.requires_grad]
+        # The backward is captured as torch.autograd.grad on the CALLING
+        # thread (multithreading off). Both choices are load-bearing on
+        # ROCm, measured on MI355X:
+        #  - loss.backward() routes through AccumulateGrad nodes; a node
+        #    cached from an earlier eager iteration is pinned to another
+        #    stream, and its cross-stream hop inside capture either
+        #    escapes capture silently (thread_local mode -> replay applied
+        #    stale grads) or kills the capture (global mode -> segfault).
+        #    autograd.grad has no AccumulateGrad nodes at all.
+        #  - the default multithreaded engine runs backward on a worker
+        #    thread, racing the capture-mode bookkeeping; single-threaded
+        #    execution keeps every launch on the capturing thread.
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(side):
+        with torch.cuda.stream(side), \
+                torch.autograd.set_multithreading_enabled(False):
             for _ in range(2):
-                opt.zero_grad(set_to_none=True)
-                fwd().backward()
+                loss = fwd()
+                torch.autograd.grad(loss, params)
         torch.cuda.current_stream().wait_stream(side)
-        opt.zero_grad(set_to_none=True)
         g = torch.cuda.CUDAGraph()
-        # Default ("global") capture mode on purpose: backward runs on the
-        # autograd worker thread, and only global mode guarantees its
-        # launches are captured rather than silently executed eagerly
-        # (measured: thread_local capture produced stale grads — the
-        # trajectory degenerated to weight-decay-only updates). The
-        # engine's controller/poller threads are idle during capture (cv
-        # waits, no HIP calls with nothing in flight), so they cannot
-        # poison a global capture.
-        with torch.cuda.graph(g):
+        with torch.cuda.graph(g), \
+                torch.autograd.set_multithreading_enabled(False):
             static_loss = fwd()
-            static_loss.backward()
-        grads = [p.grad for group in opt.param_groups
-                 for p in group["params"] if p.grad is not None]
-        if not grads:
+            grads = list(torch.autograd.grad(static_loss, params))
+        if not grads or any(gr is None for gr in grads):
             raise RuntimeError("capture produced no gradients")
+        # Bind the captured (graph-pool, stable-address) grad buffers to
+        # .grad once; every replay rewrites them in place and the fused
+        # optimizer reads them between replays.
+        opt.zero_grad(set_to_none=True)
+        for p, gr in zip(params, grads):
+            p.grad = gr
 
         def gstep():
             g.replay()

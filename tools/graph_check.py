@@ -64,6 +64,7 @@ def main():
     named = list(model.named_parameters())
     probes = [named[0], named[len(named) // 2], named[-1]]
     ref_grads = {n: p.grad.float().clone() for n, p in probes}
+    del loss0  # keep no eager graph alive across the capture
     gstep = bench.try_graph_step(opt, fwd, 1, None, None)
     if gstep is None:
         print("GRAPH-CHECK FAIL: capture unavailable")
