@@ -302,10 +302,14 @@ def main():
         step()
 
     # hipGraph replay for the launch-dense inner loop (see try_graph_step).
-    # Default on for the ResNet-50 headline config; AXON_GRAPH=0/1 forces.
+    # EXPERIMENTAL, opt-in via AXON_GRAPH=1: replay measures +16% on the
+    # ResNet-50 headline config (8545 vs 7377 img/s same-process) and the
+    # replayed gradients match eager bitwise at identical weights, but the
+    # multi-step training trajectory still diverges from eager
+    # (tools/graph_check.py fails), so results measured in this mode are
+    # not yet honest training numbers and it must stay off by default.
     graph_env = os.environ.get("AXON_GRAPH")
-    want_graph = (graph_env == "1" or
-                  (graph_env is None and args.model == "resnet50"))
+    want_graph = graph_env == "1"
     if want_graph and use_gpu and graph_env != "0":
         gstep = try_graph_step(opt, step.fwd, n, red, compression)
         ok = gstep is not None
