@@ -308,6 +308,18 @@ def main():
     # multi-step training trajectory still diverges from eager
     # (tools/graph_check.py fails), so results measured in this mode are
     # not yet honest training numbers and it must stay off by default.
+    # Diagnosis so far (MI355X, ROCm 7.0 box): the delivery logic is
+    # provably sound (a CPU emulation of stable-buffer grads + fused step
+    # tracks eager exactly), and replay-at-initial-weights reproduces
+    # eager grads bitwise — but after the first optimizer step the
+    # replayed loss stops tracking live weights for part of the network
+    # (trajectory consistent with convs frozen at their capture-time
+    # values while the BN-affine/fc layers keep training). That pattern,
+    # plus global-capture-mode hanging/segfaulting where thread_local
+    # "works", indicates MIOpen conv setup work (weight transform /
+    # workspace prep) escapes the captured stream and is snapshotted at
+    # capture instead of replayed. Next lever: MIOpen find-mode pinning
+    # or replacing the conv path before re-attempting default-on.
     graph_env = os.environ.get("AXON_GRAPH")
     want_graph = graph_env == "1"
     if want_graph and use_gpu and graph_env != "0":
