@@ -15,17 +15,17 @@ import uuid
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
-def test_bench_resnet_cpu_ws2():
+def _run_bench_ws2(extra_args, extra_env):
     port = str(20000 + (uuid.uuid4().int % 20000))
     procs = []
     for r in range(2):
         env = os.environ.copy()
         env.update({"RANK": str(r), "WORLD_SIZE": "2", "LOCAL_RANK": "0",
-                    "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": port,
-                    "AXON_GRAPH": "1"})
+                    "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": port})
+        env.update(extra_env)
         procs.append(subprocess.Popen(
             [sys.executable, "bench.py", "--steps", "1", "--warmup", "1",
-             "--batch", "2", "--image-size", "64"],
+             "--batch", "2", "--image-size", "64"] + extra_args,
             cwd=REPO, env=env, stdout=subprocess.PIPE,
             stderr=subprocess.STDOUT, text=True))
     outs = []
@@ -36,8 +36,21 @@ def test_bench_resnet_cpu_ws2():
     payloads = [line for out in outs for line in out.splitlines()
                 if line.startswith("{")]
     assert len(payloads) == 1, f"expected one rank-0 JSON line:\n{outs}"
-    rec = json.loads(payloads[0])
+    return json.loads(payloads[0])
+
+
+def test_bench_resnet_cpu_ws2():
+    rec = _run_bench_ws2([], {"AXON_GRAPH": "1"})
     assert rec["n_gpus"] == 2
     assert rec["value"] > 0
     assert rec["config"]["parallelism"] == "dp2"
     assert rec["config"]["global_batch"] == 4
+    assert rec["config"]["reduction"] == "average"
+
+
+def test_bench_resnet_adasum_cpu_ws2():
+    """Config-5 launch shape: the driver's bench contract with
+    --reduction adasum runs end-to-end (engine Adasum path at ws=2)."""
+    rec = _run_bench_ws2(["--reduction", "adasum"], {})
+    assert rec["config"]["reduction"] == "adasum"
+    assert rec["value"] > 0
