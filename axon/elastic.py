@@ -38,12 +38,25 @@ class State:
     """Commit/rollback container for arbitrary (deepcopy-able) values.
     Attribute access proxies to the tracked values."""
 
-    _RESERVED = ("_values", "_saved")
+    _RESERVED = ("_values", "_saved", "_reset_callbacks")
 
     def __init__(self, **kwargs):
         object.__setattr__(self, "_values", dict(kwargs))
         object.__setattr__(self, "_saved", None)
+        object.__setattr__(self, "_reset_callbacks", [])
         self.commit()
+
+    def register_reset_callbacks(self, callbacks):
+        """hvd.elastic parity: callbacks invoked after the worker
+        re-initializes on a membership change (new generation), once the
+        rolled-back state has been restored and re-synced — the place to
+        re-shard samplers (axon.data.ElasticSampler.set_world), rescale
+        learning rates, or rebuild world-size-dependent objects."""
+        self._reset_callbacks.extend(callbacks)
+
+    def on_reset(self):
+        for cb in self._reset_callbacks:
+            cb()
 
     def __getattr__(self, name):
         values = object.__getattribute__(self, "_values")
@@ -205,12 +218,16 @@ def run(func):
             return func(state, *args, **kwargs)
         timeout = float(os.environ.get("AXON_ELASTIC_TIMEOUT", "60"))
         known_gen = -1
+        first = True
         while True:
             gen = _wait_for_generation(path, known_gen, timeout)
             known_gen = gen["gen"]
             _apply_generation(gen)
             hvd.init()
             state.sync()
+            if not first:
+                state.on_reset()  # membership changed: notify user hooks
+            first = False
             try:
                 return func(state, *args, **kwargs)
             except RuntimeError:

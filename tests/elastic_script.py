@@ -19,6 +19,16 @@ opt = torch.optim.SGD(model.parameters(), lr=0.05)
 state = elastic.TorchState(model=model, optimizer=opt, step=0)
 
 
+def _on_reset():
+    # Survivors record the membership change (reset-callback parity).
+    wid = os.environ.get("AXON_WORKER_ID", "0")
+    with open(f"{OUT}.reset.w{wid}", "a") as f:
+        f.write(f"{hvd.size()}\n")
+
+
+state.register_reset_callbacks([_on_reset])
+
+
 @elastic.run
 def train(state):
     wid = os.environ.get("AXON_WORKER_ID", "0")

@@ -39,6 +39,10 @@ def test_elastic_shrink_on_failure(tmp_path):
     assert res[0]["step"] == 12 and res[1]["step"] == 12
     assert abs(res[0]["wsum"] - res[1]["wsum"]) < 1e-9
     assert "generation 1" in r.stderr
+    # Survivors fired their reset callbacks with the shrunken world size.
+    for w in (0, 1):
+        with open(f"{out}.reset.w{w}") as f:
+            assert f.read().strip().splitlines() == ["2"]
 
 
 def test_elastic_respawn_on_failure(tmp_path):
