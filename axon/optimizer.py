@@ -21,7 +21,9 @@ from axon.mpi_ops import ReduceOp
 class _DistributedOptimizer:
     def _distributed_init(self, named_parameters, compression,
                           backward_passes_per_step, op,
-                          gradient_predivide_factor, process_set):
+                          gradient_predivide_factor, process_set,
+                          sparse_as_dense=False):
+        self._sparse_as_dense = sparse_as_dense
         self._compression = compression
         self._bpps = backward_passes_per_step
         self._op = op
@@ -66,6 +68,16 @@ class _DistributedOptimizer:
     def _allreduce_grad_async(self, p):
         if p.grad is None:
             return
+        if p.grad.is_sparse:
+            # Sparse grads (e.g. Embedding(sparse=True)): the collective
+            # wire is dense; hvd-parity sparse_as_dense densifies before
+            # the allreduce, otherwise fail with the actionable fix.
+            if not self._sparse_as_dense:
+                raise ValueError(
+                    f"parameter '{self._names[id(p)]}' has a sparse "
+                    "gradient; pass sparse_as_dense=True to "
+                    "DistributedOptimizer (or use a dense embedding)")
+            p.grad = p.grad.to_dense()
         name = f"grad.{self._names[id(p)]}"
         if self._predivide != 1.0:
             handle = mpi_ops.allreduce_async_(
@@ -126,13 +138,17 @@ class _DistributedOptimizer:
 def DistributedOptimizer(optimizer, named_parameters=None,
                          compression=Compression.none,
                          backward_passes_per_step=1, op=ReduceOp.Average,
-                         gradient_predivide_factor=1.0, process_set=None):
+                         gradient_predivide_factor=1.0, process_set=None,
+                         sparse_as_dense=False):
     """Wrap a torch optimizer with gradient averaging over all ranks — or
-    over a registered ProcessSet's members (hvd.DistributedOptimizer API)."""
+    over a registered ProcessSet's members (hvd.DistributedOptimizer API).
+    sparse_as_dense densifies sparse gradients (Embedding(sparse=True))
+    before the allreduce, matching the reference's option."""
     cls = type(optimizer.__class__.__name__, (_DistributedOptimizer,
                                               optimizer.__class__), {})
     optimizer.__class__ = cls
     optimizer._distributed_init(named_parameters, compression,
                                 backward_passes_per_step, op,
-                                gradient_predivide_factor, process_set)
+                                gradient_predivide_factor, process_set,
+                                sparse_as_dense)
     return optimizer

@@ -164,3 +164,50 @@ def _dist_compressed(rank, size):
 
 def test_compressed_optimizer():
     run_distributed(_dist_compressed, 2)
+
+
+def _sparse_embedding_step(rank, size):
+    import axon as hvd
+
+    torch.manual_seed(5)
+    emb = nn.Embedding(12, 4, sparse=True)
+    hvd.broadcast_parameters(emb.state_dict(), root_rank=0)
+    opt = torch.optim.SGD(emb.parameters(), lr=0.5)
+    opt = hvd.DistributedOptimizer(opt,
+                                   named_parameters=emb.named_parameters(),
+                                   sparse_as_dense=True)
+    idx = torch.tensor([rank, rank + 1, 3])
+    loss = emb(idx).sum()
+    loss.backward()
+    opt.step()
+    # reference: average of both ranks' dense grads at identical weights
+    ref = torch.zeros(12, 4)
+    for r in range(size):
+        for i in [r, r + 1, 3]:
+            ref[i] += 1.0
+    ref /= size
+    assert emb.weight.grad.layout == torch.strided
+    assert torch.allclose(emb.weight.grad, ref), "sparse grad not averaged"
+
+
+def test_sparse_as_dense_ws2():
+    run_distributed(_sparse_embedding_step, 2)
+
+
+def _sparse_without_flag_errors(rank, size):
+    import axon as hvd
+
+    emb = nn.Embedding(8, 4, sparse=True)
+    opt = torch.optim.SGD(emb.parameters(), lr=0.5)
+    opt = hvd.DistributedOptimizer(opt,
+                                   named_parameters=emb.named_parameters())
+    try:
+        emb(torch.tensor([rank])).sum().backward()
+        raise AssertionError("expected sparse-grad error")
+    except (ValueError, RuntimeError) as e:
+        assert "sparse_as_dense" in str(e)
+    hvd.barrier(name="sparse.err.done")
+
+
+def test_sparse_without_flag_errors_ws2():
+    run_distributed(_sparse_without_flag_errors, 2)
