@@ -21,6 +21,8 @@ from axon.basics import (  # noqa: F401,E402
     rank,
     shutdown,
     size,
+    start_timeline,
+    stop_timeline,
 )
 from axon.compression import Compression  # noqa: F401,E402
 from axon.functions import (  # noqa: F401,E402

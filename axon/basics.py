@@ -101,3 +101,17 @@ def cross_rank():
 
 def cross_size():
     return 1
+
+
+def start_timeline(file_path, mark_cycles=False):
+    """Begin writing the chrome-trace timeline to `file_path` (rank suffix
+    appended). hvd.start_timeline parity; AXON_TIMELINE env does the same
+    from process start."""
+    if mark_cycles:
+        os.environ["AXON_TIMELINE_MARK_CYCLES"] = "1"
+    _ensure_init().start_timeline(file_path)
+
+
+def stop_timeline():
+    """Close the active timeline file (hvd.stop_timeline parity)."""
+    _ensure_init().stop_timeline()

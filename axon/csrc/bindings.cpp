@@ -245,6 +245,8 @@ PYBIND11_MODULE(_core, m) {
         py::arg("prescale") = 1.0, py::arg("postscale") = 1.0,
         py::arg("process_set") = 0);
   m.def("autotune_step_mark", &axon::pm_step_mark);
+  m.def("start_timeline", &axon::timeline_start, pybind11::arg("path"));
+  m.def("stop_timeline", &axon::timeline_stop);
   m.def("ps_register", &axon::ps_register, py::arg("ranks"));
   m.def("ps_remove", &axon::ps_remove, py::arg("id"));
   m.def("ps_count", &axon::ps_count);

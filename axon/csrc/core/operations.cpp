@@ -449,6 +449,10 @@ void pm_step_mark() {
   g.step_marks.fetch_add(1, std::memory_order_relaxed);
 }
 
+void timeline_start(const std::string& path) { g.tl.start(path); }
+
+void timeline_stop() { g.tl.stop(); }
+
 int ps_version() {
   std::lock_guard<std::mutex> lk(g.ps_mu);
   return g.ps_version;

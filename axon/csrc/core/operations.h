@@ -46,6 +46,11 @@ int ps_version();  // bumped on register and remove; controller rescan key
 // rank 0's marks feed the ParameterManager's steps/sec score.
 void pm_step_mark();
 
+// Runtime timeline control (hvd.start_timeline/stop_timeline parity);
+// rank-local, any time after initialize().
+void timeline_start(const std::string& path);
+void timeline_stop();
+
 // GPU backend factory seam (implemented in rccl_backend.cpp).
 class ControlPlane;
 class Timeline;

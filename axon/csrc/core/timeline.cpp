@@ -8,11 +8,21 @@ void Timeline::init(int rank) {
   rank_ = rank;
   std::string path = env_str("AXON_TIMELINE", "");
   if (path.empty()) return;
+  start(path);
+}
+
+void Timeline::start(const std::string& path) {
+  std::lock_guard<std::mutex> lk(mu_);
+  if (f_) {
+    std::fclose(f_);
+    f_ = nullptr;
+    open_phase_.clear();
+  }
   // One file per rank (single writer): path with rank suffix unless ws==1
   // callers pass rank-qualified path already via launcher env.
   std::string p = path;
   auto dot = p.rfind(".json");
-  std::string suffix = "." + std::to_string(rank);
+  std::string suffix = "." + std::to_string(rank_);
   if (dot != std::string::npos)
     p = p.substr(0, dot) + suffix + ".json";
   else
@@ -29,6 +39,17 @@ void Timeline::init(int rank) {
                "{\"name\":\"process_name\",\"ph\":\"M\",\"pid\":%d,"
                "\"args\":{\"name\":\"axon rank %d\"}},\n",
                rank_, rank_);
+}
+
+void Timeline::stop() {
+  std::lock_guard<std::mutex> lk(mu_);
+  if (!f_) return;
+  // Close any spans left open so the trace stays well-formed.
+  for (auto& kv : open_phase_)
+    if (!kv.second.empty()) emit("E", kv.second.c_str(), lanes_[kv.first], now_us());
+  open_phase_.clear();
+  std::fclose(f_);
+  f_ = nullptr;
 }
 
 Timeline::~Timeline() {
@@ -62,8 +83,8 @@ void Timeline::emit(const char* ph, const char* name, int tid, int64_t ts) {
 }
 
 void Timeline::begin(const std::string& tensor, const char* phase) {
-  if (!f_) return;
   std::lock_guard<std::mutex> lk(mu_);
+  if (!f_) return;
   int id = lane(tensor);
   auto& open = open_phase_[tensor];
   if (!open.empty()) emit("E", open.c_str(), id, now_us());
@@ -72,8 +93,8 @@ void Timeline::begin(const std::string& tensor, const char* phase) {
 }
 
 void Timeline::end(const std::string& tensor) {
-  if (!f_) return;
   std::lock_guard<std::mutex> lk(mu_);
+  if (!f_) return;
   auto it = open_phase_.find(tensor);
   if (it == open_phase_.end() || it->second.empty()) return;
   emit("E", it->second.c_str(), lanes_[tensor], now_us());
@@ -81,8 +102,8 @@ void Timeline::end(const std::string& tensor) {
 }
 
 void Timeline::mark_cycle() {
-  if (!f_ || !mark_cycles_) return;
   std::lock_guard<std::mutex> lk(mu_);
+  if (!f_ || !mark_cycles_) return;
   std::fprintf(f_, "{\"name\":\"cycle\",\"ph\":\"i\",\"pid\":%d,\"tid\":0,\"ts\":%lld,\"s\":\"p\"},\n",
                rank_, static_cast<long long>(now_us()));
 }

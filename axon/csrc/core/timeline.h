@@ -17,6 +17,12 @@ class Timeline {
   ~Timeline();
   bool enabled() const { return f_ != nullptr; }
 
+  // Runtime control (hvd.start_timeline/stop_timeline parity): open a new
+  // trace file (closing any active one) / close the active one. Safe to
+  // call from the user thread while the background thread records spans.
+  void start(const std::string& path);
+  void stop();
+
   // Begin/end a named phase on the tensor's lane.
   void begin(const std::string& tensor, const char* phase);
   void end(const std::string& tensor);

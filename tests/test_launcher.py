@@ -87,6 +87,31 @@ def test_timeline_written():
         assert any("allreduce.t0" in json.dumps(e) for e in lanes)
 
 
+def test_timeline_runtime_start_stop():
+    """hvd.start_timeline/stop_timeline: runtime-controlled window — only
+    the op inside the window lands in the file."""
+    with tempfile.TemporaryDirectory() as d:
+        path = os.path.join(d, "rt.json")
+        script = (
+            "import torch, axon as hvd\n"
+            "hvd.init()\n"
+            "hvd.allreduce(torch.ones(4), name='before')\n"
+            f"hvd.start_timeline({path!r})\n"
+            "hvd.allreduce(torch.ones(4), name='inside')\n"
+            "hvd.stop_timeline()\n"
+            "hvd.allreduce(torch.ones(4), name='after')\n"
+            "hvd.shutdown()\n")
+        r = _run(2, script)
+        assert r.returncode == 0, r.stderr
+        f0 = os.path.join(d, "rt.0.json")
+        assert os.path.exists(f0), os.listdir(d)
+        content = open(f0).read().rstrip().rstrip(",")
+        blob = json.dumps(json.loads(content + "]"))
+        assert "allreduce.inside" in blob
+        assert "allreduce.before" not in blob
+        assert "allreduce.after" not in blob
+
+
 def test_duplicate_inflight_name_rejected():
     """Two async ops with the same name in flight must error loudly, not
     corrupt the pending table. Race-free setup: rank 1 delays its first
