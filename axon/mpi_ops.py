@@ -103,8 +103,35 @@ def allreduce_async(tensor, average=None, name=None, op=None,
     return _register(h, output)
 
 
+class _DiffAllreduce(torch.autograd.Function):
+    """Differentiable allreduce (hvd parity): backward re-enters the same
+    collective with the same op, so gradients of cross-rank expressions
+    aggregate correctly (sum over every rank's loss)."""
+
+    @staticmethod
+    def forward(ctx, tensor, red, name, prescale, postscale, compression,
+                process_set):
+        ctx.args = (red, name, prescale, postscale, compression, process_set)
+        return synchronize(allreduce_async(tensor, None, name, red, prescale,
+                                           postscale, compression,
+                                           process_set))
+
+    @staticmethod
+    def backward(ctx, grad):
+        red, name, prescale, postscale, compression, process_set = ctx.args
+        g = synchronize(allreduce_async(
+            grad.contiguous(), None, f"{name}.grad" if name else None, red,
+            prescale, postscale, compression, process_set))
+        return (g,) + (None,) * 6
+
+
 def allreduce(tensor, average=None, name=None, op=None, prescale_factor=1.0,
               postscale_factor=1.0, compression=None, process_set=None):
+    if torch.is_grad_enabled() and tensor.requires_grad:
+        red = _resolve_op(op, average)
+        return _DiffAllreduce.apply(tensor, red, name, prescale_factor,
+                                    postscale_factor, compression,
+                                    process_set)
     return synchronize(allreduce_async(tensor, average, name, op,
                                        prescale_factor, postscale_factor,
                                        compression, process_set))
@@ -174,7 +201,29 @@ def broadcast_async(tensor, root_rank, name=None, process_set=None):
     return _register(h, output)
 
 
+class _DiffBroadcast(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, tensor, root_rank, name, process_set):
+        ctx.args = (root_rank, name, process_set)
+        return synchronize(broadcast_async(tensor, root_rank, name,
+                                           process_set))
+
+    @staticmethod
+    def backward(ctx, grad):
+        root_rank, name, process_set = ctx.args
+        # Output is root's tensor everywhere, so d(input) sums every rank's
+        # grad on the root and is zero elsewhere.
+        g = synchronize(allreduce_async(
+            grad.contiguous(), None, f"{name}.grad" if name else None,
+            ReduceOp.Sum, 1.0, 1.0, None, process_set))
+        if rank() != root_rank:
+            g = torch.zeros_like(g)
+        return g, None, None, None
+
+
 def broadcast(tensor, root_rank, name=None, process_set=None):
+    if torch.is_grad_enabled() and tensor.requires_grad:
+        return _DiffBroadcast.apply(tensor, root_rank, name, process_set)
     return synchronize(broadcast_async(tensor, root_rank, name, process_set))
 
 
@@ -204,7 +253,35 @@ def allgather_async(tensor, name=None, process_set=None):
     return _register(h, output)
 
 
+class _DiffAllgather(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, tensor, name, process_set):
+        ctx.args = (name, process_set)
+        ctx.dim0 = tensor.shape[0] if tensor.dim() > 0 else tensor.numel()
+        return synchronize(allgather_async(tensor, name, process_set))
+
+    @staticmethod
+    def backward(ctx, grad):
+        name, process_set = ctx.args
+        gname = f"{name}.grad" if name else None
+        # Sum every rank's grad of the (identical) gathered output, then
+        # take this rank's slice (hvd semantics).
+        g = synchronize(allreduce_async(grad.contiguous(), None, gname,
+                                        ReduceOp.Sum, 1.0, 1.0, None,
+                                        process_set))
+        dims = synchronize(allgather_async(
+            torch.tensor([ctx.dim0]), f"{gname}.dims" if gname else None,
+            process_set))
+        psid = resolve_id(process_set)
+        my = rank() if psid == 0 else \
+            list(_ensure_init().ps_ranks(psid)).index(rank())
+        offset = int(dims[:my].sum())
+        return g.narrow(0, offset, ctx.dim0), None, None
+
+
 def allgather(tensor, name=None, process_set=None):
+    if torch.is_grad_enabled() and tensor.requires_grad:
+        return _DiffAllgather.apply(tensor, name, process_set)
     return synchronize(allgather_async(tensor, name, process_set))
 
 
@@ -230,9 +307,32 @@ def alltoall_async(tensor, splits=None, name=None, process_set=None):
     return _register(h, (output, received_splits))
 
 
+class _DiffAlltoall(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, tensor, splits, name, process_set):
+        out, received = synchronize(alltoall_async(tensor, splits, name,
+                                                   process_set))
+        ctx.args = (name, process_set)
+        ctx.received = received.tolist()
+        ctx.mark_non_differentiable(received)
+        return out, received
+
+    @staticmethod
+    def backward(ctx, grad, _grad_received):
+        name, process_set = ctx.args
+        # The reverse exchange: what rank p sent us, we send back to p.
+        g, _ = synchronize(alltoall_async(
+            grad.contiguous(), ctx.received,
+            f"{name}.grad" if name else None, process_set))
+        return g, None, None, None
+
+
 def alltoall(tensor, splits=None, name=None, process_set=None):
-    out, received = synchronize(alltoall_async(tensor, splits, name,
-                                               process_set))
+    if torch.is_grad_enabled() and tensor.requires_grad:
+        out, received = _DiffAlltoall.apply(tensor, splits, name, process_set)
+    else:
+        out, received = synchronize(alltoall_async(tensor, splits, name,
+                                                   process_set))
     if splits is None:
         return out
     return out, received
@@ -255,8 +355,31 @@ def reducescatter_async(tensor, op=ReduceOp.Average, name=None,
     return _register(h, output)
 
 
+class _DiffReducescatter(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, tensor, op, name, prescale, postscale, process_set):
+        ctx.args = (op, name, process_set)
+        return synchronize(reducescatter_async(tensor, op, name, prescale,
+                                               postscale, process_set))
+
+    @staticmethod
+    def backward(ctx, grad):
+        op, name, process_set = ctx.args
+        g = synchronize(allgather_async(grad.contiguous(),
+                                        f"{name}.grad" if name else None,
+                                        process_set))
+        if op == ReduceOp.Average:
+            psid = resolve_id(process_set)
+            n = size() if psid == 0 else len(_ensure_init().ps_ranks(psid))
+            g = g / n
+        return (g,) + (None,) * 5
+
+
 def reducescatter(tensor, op=ReduceOp.Average, name=None, prescale_factor=1.0,
                   postscale_factor=1.0, process_set=None):
+    if torch.is_grad_enabled() and tensor.requires_grad:
+        return _DiffReducescatter.apply(tensor, op, name, prescale_factor,
+                                        postscale_factor, process_set)
     return synchronize(reducescatter_async(tensor, op, name, prescale_factor,
                                            postscale_factor, process_set))
 
