@@ -95,3 +95,21 @@ def cuda_built():
 
 def rocm_built():
     return True
+
+
+def mpi_threads_supported():
+    """hvd-compat: no MPI in this engine (socket control plane)."""
+    return False
+
+
+def mpi_enabled():
+    return False
+
+
+def gloo_enabled():
+    return False
+
+
+def is_homogeneous():
+    """Single-node engine: every rank shares the host, so always True."""
+    return True

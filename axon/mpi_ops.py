@@ -173,6 +173,12 @@ def grouped_allreduce_async(tensors, average=None, name=None, op=None,
 
 
 def grouped_allreduce(tensors, **kw):
+    if torch.is_grad_enabled() and any(t.requires_grad for t in tensors):
+        # Differentiable path: one allreduce() per tensor; same-cycle
+        # submissions still fuse in the controller like the async group.
+        gname = _auto_name("grouped", kw.pop("name", None))
+        return [allreduce(t, name=f"{gname}.{i}", **kw)
+                for i, t in enumerate(tensors)]
     return [synchronize(h) for h in grouped_allreduce_async(tensors, **kw)]
 
 

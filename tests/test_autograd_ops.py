@@ -88,3 +88,18 @@ def _alltoall_grads(rank, size):
 
 def test_alltoall_autograd():
     run_distributed(_alltoall_grads, 2)
+
+
+def _grouped_allreduce_grads(rank, size):
+    import axon as hvd
+
+    ts = [torch.full((2,), float(rank + 1 + i), requires_grad=True)
+          for i in range(3)]
+    outs = hvd.grouped_allreduce(ts, op=hvd.Average, name="gar")
+    sum(o.sum() for o in outs).backward()
+    for t in ts:
+        assert torch.allclose(t.grad, torch.ones(2)), t.grad
+
+
+def test_grouped_allreduce_autograd():
+    run_distributed(_grouped_allreduce_grads, 2)
