@@ -211,3 +211,40 @@ def _sparse_without_flag_errors(rank, size):
 
 def test_sparse_without_flag_errors_ws2():
     run_distributed(_sparse_without_flag_errors, 2)
+
+
+def _amp_scaler_pattern(rank, size):
+    """The reference's documented AMP recipe: scale -> backward ->
+    optimizer.synchronize() -> unscale_ -> clip -> skip_synchronize() +
+    scaler.step()."""
+    import axon as hvd
+
+    torch.manual_seed(3)
+    model = nn.Linear(4, 2)
+    hvd.broadcast_parameters(model.state_dict(), root_rank=0)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    opt = hvd.DistributedOptimizer(opt,
+                                   named_parameters=model.named_parameters())
+    scaler = torch.amp.GradScaler("cpu", init_scale=2.0**8)
+    g = torch.Generator().manual_seed(rank)
+    x = torch.randn(6, 4, generator=g)
+    for _ in range(3):
+        opt.zero_grad()
+        loss = (model(x) ** 2).mean()
+        scaler.scale(loss).backward()
+        opt.synchronize()
+        scaler.unscale_(opt)
+        torch.nn.utils.clip_grad_norm_(model.parameters(), 1.0)
+        with opt.skip_synchronize():
+            scaler.step(opt)
+        scaler.update()
+    # ranks stay in lockstep (same weights after averaged grads)
+    wsum = torch.tensor([sum(float(p.double().sum())
+                             for p in model.parameters())])
+    mx = hvd.allreduce(wsum, op=hvd.Max, name="amp.wmax")
+    mn = hvd.allreduce(wsum, op=hvd.Min, name="amp.wmin")
+    assert abs(float(mx - mn)) < 1e-12
+
+
+def test_amp_grad_scaler_pattern_ws2():
+    run_distributed(_amp_scaler_pattern, 2)
