@@ -69,3 +69,36 @@ def test_randomized_submission_orders():
 def test_randomized_submission_orders_ws4():
     for seed in (7, 11):
         run_distributed(_random_order_ops, 4, args=(seed,))
+
+
+def _cache_churn(rank, size, seed):
+    """Response-cache property: repeated NAMED ops (cache-id steady state)
+    with random shape/dtype churn must stay correct — every churn forces a
+    fingerprint mismatch, a full re-negotiation, and a template refresh."""
+    import axon as hvd
+
+    rng = random.Random(seed)  # same decisions on every rank
+    shapes = {}
+    for step in range(30):
+        name = f"c{rng.randint(0, 4)}"  # few names -> heavy reuse
+        if name not in shapes or rng.random() < 0.3:
+            shapes[name] = (rng.randint(1, 500),
+                            rng.choice([torch.float32, torch.float64]))
+        numel, dt = shapes[name]
+        kind = rng.choice(["allreduce", "allgather"])
+        t = torch.full((numel,), float(rank + 1), dtype=dt)
+        if kind == "allreduce":
+            out = hvd.allreduce(t, op=hvd.Sum, name=name)
+            assert out.dtype == dt and out.numel() == numel
+            assert torch.allclose(
+                out, torch.full((numel,), float(sum(range(1, size + 1))),
+                                dtype=dt)), (step, name)
+        else:
+            # allgather under the same name namespace would collide with
+            # the allreduce entries; scope it
+            out = hvd.allgather(t, name=f"ag.{name}")
+            assert out.numel() == numel * size
+
+
+def test_response_cache_churn_ws3():
+    run_distributed(_cache_churn, 3, args=(11,))
