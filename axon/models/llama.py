@@ -72,16 +72,29 @@ class LlamaAttention(nn.Module):
         self.wv = nn.Linear(cfg.dim, cfg.kv_heads * self.head_dim, bias=False)
         self.wo = nn.Linear(cfg.dim, cfg.dim, bias=False)
 
-    def forward(self, x, cos, sin):
+    def forward(self, x, cos, sin, cache=None):
         b, s, _ = x.shape
         q = rope(self.wq(x).view(b, s, self.heads, self.head_dim), cos, sin)
         k = rope(self.wk(x).view(b, s, self.kv_heads, self.head_dim), cos, sin)
         v = self.wv(x).view(b, s, self.kv_heads, self.head_dim)
+        causal = True
+        if cache is not None:
+            # Incremental decode: append this call's K/V to the layer cache
+            # ((b, t, kv_heads, d) pre-transpose layout). For the prefill
+            # call the cache is empty and causal attention applies; decode
+            # steps (q shorter than the cache) attend to everything past —
+            # torch SDPA's is_causal aligns top-left, which would mask the
+            # history, so causal turns off there.
+            if cache["k"] is not None:
+                k = torch.cat([cache["k"], k], 1)
+                v = torch.cat([cache["v"], v], 1)
+            cache["k"], cache["v"] = k, v
+            causal = s == k.shape[1]
         # AXON_FA=1 routes through the CDNA4 flash-attention kernels
         # (axon/ops/flash_attention.py); default is torch SDPA.
         from axon.ops.flash_attention import flash_attention
         o = flash_attention(q.transpose(1, 2), k.transpose(1, 2),
-                            v.transpose(1, 2), causal=True)
+                            v.transpose(1, 2), causal=causal)
         return self.wo(o.transpose(1, 2).reshape(b, s, -1))
 
 
@@ -95,8 +108,8 @@ class LlamaBlock(nn.Module):
         self.w3 = nn.Linear(cfg.dim, cfg.ffn, bias=False)  # up
         self.w2 = nn.Linear(cfg.ffn, cfg.dim, bias=False)  # down
 
-    def forward(self, x, cos, sin):
-        x = x + self.attn(self.attn_norm(x), cos, sin)
+    def forward(self, x, cos, sin, cache=None):
+        x = x + self.attn(self.attn_norm(x), cos, sin, cache)
         h = self.ffn_norm(x)
         return x + self.w2(swiglu(self.w1(h), self.w3(h)))
 
@@ -117,18 +130,26 @@ class LlamaForCausalLM(nn.Module):
             nn.init.normal_(m.weight, std=0.02)
 
     def _rope_for(self, s, device, dtype):
-        key = (s, device, dtype)
-        if key not in self._rope:
-            self._rope[key] = _rope_cache(self.cfg.dim // self.cfg.heads, s,
+        key = (device, dtype)
+        have = self._rope.get(key)
+        if have is None or have[0].shape[0] < s:
+            # Grow geometrically so incremental decoding (s+1 per step)
+            # doesn't rebuild the table every token.
+            n = max(s, 2 * have[0].shape[0] if have is not None else s)
+            self._rope[key] = _rope_cache(self.cfg.dim // self.cfg.heads, n,
                                           self.cfg.rope_theta, device, dtype)
-        return self._rope[key]
+        cos, sin = self._rope[key]
+        return cos[:s], sin[:s]
 
-    def forward(self, input_ids):
+    def forward(self, input_ids, caches=None, pos=0):
         b, s = input_ids.shape
         x = self.embed(input_ids)
-        cos, sin = self._rope_for(s, x.device, x.dtype)
-        for blk in self.blocks:
-            x = blk(x, cos, sin)
+        # Rope rows for the ABSOLUTE positions of these tokens (pos offset
+        # when decoding against a KV cache).
+        cos, sin = self._rope_for(pos + s, x.device, x.dtype)
+        cos, sin = cos[pos:pos + s], sin[pos:pos + s]
+        for i, blk in enumerate(self.blocks):
+            x = blk(x, cos, sin, caches[i] if caches is not None else None)
         return self.lm_head(self.norm(x))
 
     def lm_loss(self, input_ids):
@@ -137,3 +158,31 @@ class LlamaForCausalLM(nn.Module):
         return F.cross_entropy(
             logits[:, :-1].float().reshape(-1, self.cfg.vocab_size),
             input_ids[:, 1:].reshape(-1))
+
+    @torch.no_grad()
+    def generate(self, input_ids, max_new_tokens, temperature=0.0,
+                 top_k=None):
+        """KV-cached autoregressive decoding: one prefill pass over the
+        prompt, then one single-token pass per generated token (each decode
+        step reads the per-layer K/V cache instead of recomputing the
+        prefix). temperature 0 = greedy; otherwise softmax sampling over
+        the top_k (or full) distribution."""
+        caches = [{"k": None, "v": None} for _ in self.blocks]
+        tokens = input_ids
+        step_in = input_ids
+        pos = 0
+        for _ in range(max_new_tokens):
+            logits = self.forward(step_in, caches, pos)[:, -1].float()
+            pos += step_in.shape[1]
+            if temperature > 0:
+                logits = logits / temperature
+                if top_k is not None:
+                    kth = torch.topk(logits, top_k, dim=-1).values[..., -1:]
+                    logits = logits.masked_fill(logits < kth,
+                                                float("-inf"))
+                nxt = torch.multinomial(F.softmax(logits, -1), 1)
+            else:
+                nxt = logits.argmax(-1, keepdim=True)
+            tokens = torch.cat([tokens, nxt], 1)
+            step_in = nxt
+        return tokens
