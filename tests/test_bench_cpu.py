@@ -54,3 +54,32 @@ def test_bench_resnet_adasum_cpu_ws2():
     rec = _run_bench_ws2(["--reduction", "adasum"], {})
     assert rec["config"]["reduction"] == "adasum"
     assert rec["value"] > 0
+
+
+def test_bench_resnet_cpu_ws8():
+    """The scaling run's max launch shape: 8 ranks through the full bench
+    stack (socket star with 8 connections, broadcast_parameters, hook
+    allreduces, barrier, max-over-ranks elapsed) on CPU."""
+    port = str(20000 + (uuid.uuid4().int % 20000))
+    procs = []
+    for r in range(8):
+        env = os.environ.copy()
+        env.update({"RANK": str(r), "WORLD_SIZE": "8", "LOCAL_RANK": "0",
+                    "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": port,
+                    "OMP_NUM_THREADS": "1"})
+        procs.append(subprocess.Popen(
+            [sys.executable, "bench.py", "--steps", "1", "--warmup", "0",
+             "--batch", "1", "--image-size", "64"],
+            cwd=REPO, env=env, stdout=subprocess.PIPE,
+            stderr=subprocess.STDOUT, text=True))
+    outs = []
+    for p in procs:
+        out, _ = p.communicate(timeout=600)
+        outs.append(out)
+        assert p.returncode == 0, f"bench rank failed:\n{out}"
+    payloads = [line for out in outs for line in out.splitlines()
+                if line.startswith("{")]
+    assert len(payloads) == 1
+    rec = json.loads(payloads[0])
+    assert rec["n_gpus"] == 8
+    assert rec["config"]["global_batch"] == 8
