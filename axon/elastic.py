@@ -71,8 +71,13 @@ class State:
             self._values[name] = value
 
     def commit(self):
-        """Snapshot current values (in memory) as the rollback point."""
+        """Snapshot current values (in memory) as the rollback point. In an
+        elastic run this is also the membership-change boundary: if the
+        launcher has published a newer generation (discovery grew or shrank
+        the world), commit raises HostsUpdatedInterrupt AFTER saving so the
+        run loop re-joins with this commit intact."""
         self._saved = self._snapshot()
+        _check_hosts_updated()
 
     def restore(self):
         """Roll back to the last commit."""
@@ -169,6 +174,37 @@ class _Evicted(SystemExit):
     pass
 
 
+class HostsUpdatedInterrupt(Exception):
+    """Raised from State.commit() when the launcher has published a newer
+    generation (e.g. discovery grew the world): the worker re-joins at the
+    commit boundary instead of waiting for a failure (hvd.elastic's
+    HostsUpdatedInterrupt role)."""
+
+
+_current_gen = -1   # generation this process is currently running in
+_last_gen_check = 0.0
+
+
+def _check_hosts_updated():
+    """Cheap (throttled) probe for a newer generation; called from
+    State.commit() so growth interrupts land at commit boundaries."""
+    global _last_gen_check
+    path = _genfile_path()
+    if path is None or _current_gen < 0:
+        return
+    now = time.time()
+    if now - _last_gen_check < 0.2:
+        return
+    _last_gen_check = now
+    try:
+        gen = read_generation(path)
+    except (OSError, ValueError):
+        return
+    if gen["gen"] > _current_gen:
+        raise HostsUpdatedInterrupt(
+            f"generation {gen['gen']} published (size {gen['size']})")
+
+
 def _wait_for_generation(path, newer_than, timeout):
     deadline = time.time() + timeout
     while time.time() < deadline:
@@ -217,12 +253,14 @@ def run(func):
             state.sync()
             return func(state, *args, **kwargs)
         timeout = float(os.environ.get("AXON_ELASTIC_TIMEOUT", "60"))
+        global _current_gen
         known_gen = -1
         first = True
         while True:
             gen = _wait_for_generation(path, known_gen, timeout)
             known_gen = gen["gen"]
             _apply_generation(gen)
+            _current_gen = gen["gen"]
             hvd.init()
             state.sync()
             if not first:
@@ -230,6 +268,10 @@ def run(func):
             first = False
             try:
                 return func(state, *args, **kwargs)
+            except HostsUpdatedInterrupt:
+                # discovery changed the world: the state was just
+                # committed — leave this world and re-join, no rollback
+                hvd.shutdown()
             except RuntimeError:
                 if not _is_internal_error():
                     raise

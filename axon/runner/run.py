@@ -46,6 +46,13 @@ def main(argv=None):
                              "failures (see axon/elastic.py)")
     parser.add_argument("--min-np", type=int, default=1,
                         help="elastic: abort if the world would shrink below")
+    parser.add_argument("--discover-cmd", default=None,
+                        help="elastic world-size discovery: a shell command "
+                             "printing the DESIRED number of workers; polled "
+                             "every --discover-interval seconds, growth "
+                             "spawns workers / shrink evicts the "
+                             "highest-numbered ones at a new generation")
+    parser.add_argument("--discover-interval", type=float, default=1.0)
     parser.add_argument("--max-restarts", type=int, default=1,
                         help="elastic: respawns allowed per worker before it "
                              "is evicted (world shrinks)")
@@ -179,9 +186,65 @@ def run_elastic(args):
     for w in members:
         spawn(w)
 
+    def probe_discovery():
+        try:
+            r = subprocess.run(args.discover_cmd, shell=True,
+                               capture_output=True, text=True, timeout=5)
+            return int(r.stdout.strip())
+        except (ValueError, subprocess.SubprocessError) as e:
+            if args.verbose:
+                print(f"[axonrun] discovery probe failed: {e}",
+                      file=sys.stderr)
+            return None
+
     exit_code = 0
+    evicted_by_discovery = []
+    next_probe = time.time() + args.discover_interval
+    next_worker_id = max(members) + 1
     try:
         while True:
+            # Discovery-driven membership changes (hvd.elastic discovery
+            # parity, single node): grow by spawning fresh worker ids,
+            # shrink by evicting the highest-numbered members. Running
+            # survivors pick the new generation up at their next
+            # state.commit() (HostsUpdatedInterrupt).
+            if args.discover_cmd and time.time() >= next_probe:
+                next_probe = time.time() + args.discover_interval
+                target = probe_discovery()
+                if target is not None:
+                    target = min(target, 32)  # engine rank masks are 32-bit
+                if target is not None and target >= args.min_np and \
+                        target != len(members):
+                    if target > len(members):
+                        new = list(range(next_worker_id,
+                                         next_worker_id + target - len(members)))
+                        next_worker_id += len(new)
+                        members = members + new
+                        for w in new:
+                            restarts[w] = 0
+                        gen += 1
+                        write_genfile()
+                        for w in new:
+                            spawn(w)
+                    else:
+                        drop = members[target:]
+                        members = members[:target]
+                        gen += 1
+                        write_genfile()  # published BEFORE the kill: the
+                        # survivors' next commit sees it as a resize, not
+                        # a failure
+                        for w in drop:
+                            evicted_by_discovery.append(w)
+                            if procs[w].poll() is None:
+                                try:
+                                    os.killpg(os.getpgid(procs[w].pid),
+                                              signal.SIGTERM)
+                                except (ProcessLookupError, PermissionError):
+                                    pass
+            for w in list(evicted_by_discovery):
+                if procs[w].poll() is not None:
+                    evicted_by_discovery.remove(w)  # reaped
+
             running = [w for w in members if procs[w].poll() is None]
             done_ok = [w for w in members if procs[w].poll() == 0]
             dead = [w for w in members

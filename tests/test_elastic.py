@@ -111,3 +111,75 @@ def test_elastic_clean_exit_excluded_from_generation(tmp_path):
         assert res["size"] == 2, res
         assert res["step"] == 8, res
     assert not os.path.exists(out + ".w1.json")
+
+
+SCRIPT4 = os.path.join(REPO, "tests", "elastic_script4.py")
+
+
+def _run_discover(np_, out, flips, total=60):
+    import time
+
+    disc = out + ".target"
+    with open(disc, "w") as f:
+        f.write(str(np_))
+    cmd = [sys.executable, "-m", "axon.runner.run", "-np", str(np_),
+           "--elastic", "--no-pin", "-v", "--min-np", "1",
+           "--max-restarts", "0", "--discover-cmd", f"cat {disc}",
+           "--discover-interval", "0.3",
+           sys.executable, SCRIPT4, out, str(total)]
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["OMP_NUM_THREADS"] = "1"
+    p = subprocess.Popen(cmd, env=env, stdout=subprocess.PIPE,
+                         stderr=subprocess.PIPE, text=True)
+    for wait_marker, value in flips:
+        # wait until training in the CURRENT world is demonstrably under
+        # way before resizing (a flip during startup would just be picked
+        # up as the boot generation, not a mid-training interrupt)
+        deadline = time.time() + 90
+        while not os.path.exists(wait_marker) and time.time() < deadline:
+            time.sleep(0.1)
+        assert os.path.exists(wait_marker), "training never started"
+        with open(disc + ".tmp", "w") as f:
+            f.write(str(value))
+        os.replace(disc + ".tmp", disc)
+    out_, err = p.communicate(timeout=180)
+    return p.returncode, out_, err
+
+
+def test_elastic_discovery_grow(tmp_path):
+    """Discovery raises the target 2 -> 3 mid-run: a third worker is
+    spawned, survivors re-join at a commit boundary
+    (HostsUpdatedInterrupt), and every worker finishes in the ws=3
+    world with identical weights."""
+    out = str(tmp_path / f"gr_{uuid.uuid4().hex[:8]}")
+    rc, _o, err = _run_discover(2, out, [(out + ".started.2", 3)])
+    assert rc == 0, err[-2000:]
+    res = {}
+    for w in (0, 1, 2):
+        with open(f"{out}.w{w}.json") as f:
+            res[w] = json.load(f)
+    for w in (0, 1, 2):
+        assert res[w]["size"] == 3, res
+        assert res[w]["step"] == 60
+    assert res[0]["sizes"] == [2, 3]
+    assert abs(res[0]["wsum"] - res[2]["wsum"]) < 1e-9
+    assert "generation 1: size=3" in err
+
+
+def test_elastic_discovery_shrink(tmp_path):
+    """Discovery lowers the target 3 -> 2: the highest worker is evicted
+    (not treated as a failure), survivors continue in the ws=2 world."""
+    out = str(tmp_path / f"sh_{uuid.uuid4().hex[:8]}")
+    rc, _o, err = _run_discover(3, out, [(out + ".started.3", 2)])
+    assert rc == 0, err[-2000:]
+    res = {}
+    for w in (0, 1):
+        with open(f"{out}.w{w}.json") as f:
+            res[w] = json.load(f)
+    assert not os.path.exists(f"{out}.w2.json")  # evicted mid-run
+    for w in (0, 1):
+        assert res[w]["size"] == 2
+        assert res[w]["step"] == 60
+    assert res[0]["sizes"] == [3, 2]
+    assert abs(res[0]["wsum"] - res[1]["wsum"]) < 1e-9
