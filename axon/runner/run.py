@@ -32,6 +32,11 @@ def main(argv=None):
         prog="axonrun", description="axon single-node launcher")
     parser.add_argument("-np", "--num-proc", type=int, required=True,
                         help="number of ranks (one per GPU)")
+    parser.add_argument("-H", "--hosts", "--hostfile", default=None,
+                        help="NOT SUPPORTED: this engine is single-node by "
+                             "design (xGMI is the fabric); multi-host "
+                             "launches are rejected with this message "
+                             "instead of silently running on one host")
     parser.add_argument("--no-pin", action="store_true",
                         help="do not pin ranks to GPUs via HIP_VISIBLE_DEVICES")
     parser.add_argument("--timeline", default=None,
@@ -62,6 +67,14 @@ def main(argv=None):
 
     if not args.command:
         parser.error("no command given")
+    if args.hosts:
+        only_local = all(h.split(":")[0] in ("localhost", "127.0.0.1")
+                         for h in args.hosts.split(",") if h)
+        if not only_local:
+            parser.error(
+                "-H/--hosts: multi-host launch is not supported — this "
+                "engine targets ONE node of MI355X GPUs over xGMI "
+                "(BASELINE single-node scope); run one job per node")
     if args.elastic:
         return run_elastic(args)
     np_ = args.num_proc

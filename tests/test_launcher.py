@@ -222,3 +222,22 @@ def test_autotune_end_to_end_signal():
     # refinement happened: more samples than the 12-point base grid
     assert r.stderr.count("autotune sample") > 12, \
         r.stderr.count("autotune sample")
+
+
+def test_multi_host_rejected():
+    """-H with a remote host fails fast with the single-node scope message
+    (horovodrun's -H exists; silently running on one host would be worse
+    than an error)."""
+    r = subprocess.run(
+        [sys.executable, "-m", "axon.runner.run", "-np", "2",
+         "-H", "nodeA:4,nodeB:4", sys.executable, "-c", "pass"],
+        capture_output=True, text=True, timeout=60)
+    assert r.returncode != 0
+    assert "multi-host launch is not supported" in r.stderr
+    r2 = subprocess.run(
+        [sys.executable, "-m", "axon.runner.run", "-np", "1", "--no-pin",
+         "-H", "localhost:1", sys.executable, "-c",
+         "import axon; axon.init(); axon.shutdown()"],
+        capture_output=True, text=True, timeout=120,
+        env={**os.environ, "PYTHONPATH": REPO})
+    assert r2.returncode == 0, r2.stderr
