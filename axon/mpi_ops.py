@@ -9,6 +9,7 @@ import threading
 
 import torch
 
+from axon import metrics
 from axon.basics import _ensure_init, rank, size
 from axon.process_sets import resolve_id
 
@@ -96,6 +97,7 @@ def allreduce_async(tensor, average=None, name=None, op=None,
     psid = resolve_id(process_set)
     t = tensor if _dense(tensor) else tensor.contiguous()
     output = torch.empty_like(t)
+    metrics.record_op('allreduce', t.numel() * t.element_size())
     core = _ensure_init()
     h = core.allreduce_async(t, output, _auto_name("allreduce", name, psid),
                              red, prescale_factor, postscale_factor,
@@ -145,6 +147,7 @@ def allreduce_async_(tensor, average=None, name=None, op=None,
         raise ValueError("allreduce_ requires a dense tensor")
     red = _resolve_op(op, average)
     psid = resolve_id(process_set)
+    metrics.record_op('allreduce', tensor.numel() * tensor.element_size())
     core = _ensure_init()
     h = core.allreduce_async(tensor, tensor,
                              _auto_name("allreduce", name, psid), red,
@@ -201,6 +204,7 @@ def broadcast_async(tensor, root_rank, name=None, process_set=None):
     psid = resolve_id(process_set)
     t = tensor if _dense(tensor) else tensor.contiguous()
     output = torch.empty_like(t)
+    metrics.record_op('broadcast', t.numel() * t.element_size())
     core = _ensure_init()
     h = core.broadcast_async(t, output, root_rank,
                              _auto_name("broadcast", name, psid), psid)
@@ -237,6 +241,7 @@ def broadcast_async_(tensor, root_rank, name=None, process_set=None):
     if not _dense(tensor):
         raise ValueError("broadcast_ requires a dense tensor")
     psid = resolve_id(process_set)
+    metrics.record_op('broadcast', tensor.numel() * tensor.element_size())
     core = _ensure_init()
     h = core.broadcast_async(tensor, tensor, root_rank,
                              _auto_name("broadcast", name, psid), psid)
@@ -253,6 +258,7 @@ def allgather_async(tensor, name=None, process_set=None):
     psid = resolve_id(process_set)
     t = tensor.contiguous()
     output = t.new_empty(0)
+    metrics.record_op('allgather', t.numel() * t.element_size())
     core = _ensure_init()
     h = core.allgather_async(t, output, _auto_name("allgather", name, psid),
                              psid)
@@ -307,6 +313,7 @@ def alltoall_async(tensor, splits=None, name=None, process_set=None):
         splits_list = [int(x) for x in splits]
     output = t.new_empty(0)
     received_splits = torch.empty(n, dtype=torch.int64)
+    metrics.record_op('alltoall', t.numel() * t.element_size())
     core = _ensure_init()
     h = core.alltoall_async(t, output, splits_list, received_splits,
                             _auto_name("alltoall", name, psid), psid)
@@ -354,6 +361,7 @@ def reducescatter_async(tensor, op=ReduceOp.Average, name=None,
     psid = resolve_id(process_set)
     t = tensor.contiguous()
     output = t.new_empty(0)
+    metrics.record_op('reducescatter', t.numel() * t.element_size())
     core = _ensure_init()
     h = core.reducescatter_async(t, output,
                                  _auto_name("reducescatter", name, psid),

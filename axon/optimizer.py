@@ -123,8 +123,9 @@ class _DistributedOptimizer:
         ret = super().step(closure)
         # End-to-end autotune signal: steps/sec on rank 0 scores the
         # fusion-threshold/cycle-time sweep (no-op unless AXON_AUTOTUNE=1).
-        from axon import _core
+        from axon import _core, metrics
         _core.autotune_step_mark()
+        metrics.record_step()
         return ret
 
     def zero_grad(self, set_to_none=True):
