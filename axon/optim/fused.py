@@ -55,7 +55,22 @@ class FusedSGD(Optimizer):
                     continue
                 st = self.state[p]
                 if mu != 0 and "momentum_buffer" not in st:
-                    st["momentum_buffer"] = torch.zeros_like(p)
+                    damp = group["dampening"]
+                    if damp == 0:
+                        st["momentum_buffer"] = torch.zeros_like(p)
+                    else:
+                        # torch.optim.SGD seeds the buffer with the raw
+                        # first-step update direction (dampening is NOT
+                        # applied on init). A zero buffer only reproduces
+                        # that when dampening == 0; seeding with
+                        # (damp/mu)*d0 makes the uniform kernel update
+                        # mu*m + (1-damp)*d equal d on the first step.
+                        g0 = _like_param(p, p.grad)
+                        d0 = (-g0 if group["maximize"] else g0).float()
+                        if group["weight_decay"] != 0:
+                            d0 = d0 + group["weight_decay"] * p.detach().float()
+                        st["momentum_buffer"] = \
+                            (d0 * (damp / mu)).to(p.dtype)
                 elif mu != 0:
                     st["momentum_buffer"] = _like_param(p, st["momentum_buffer"])
                 work.append((p, _like_param(p, p.grad),

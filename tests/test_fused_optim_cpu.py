@@ -92,3 +92,40 @@ def test_fused_state_dict_interchange():
     st = list(o2.state.values())[0]
     assert "exp_avg" in st and "exp_avg_sq" in st
     assert float(st["step"]) == 4
+
+
+# ------------------------- property sweeps (hypothesis) -------------------
+
+from hypothesis import given, settings, strategies as st  # noqa: E402
+
+
+@settings(max_examples=25, deadline=None)
+@given(lr=st.floats(1e-4, 0.5), momentum=st.floats(0.0, 0.99),
+       dampening=st.floats(0.0, 0.9), wd=st.floats(0.0, 0.05),
+       maximize=st.booleans(), seed=st.integers(0, 10_000))
+def test_fused_sgd_property_sweep(lr, momentum, dampening, wd, maximize,
+                                  seed):
+    m1, m2 = _pair(seed % 100)
+    kw = dict(lr=lr, momentum=momentum, dampening=dampening,
+              weight_decay=wd, maximize=maximize)
+    _train(m1, FusedSGD(m1.parameters(), **kw), steps=3, seed=seed)
+    _train(m2, torch.optim.SGD(m2.parameters(), **kw), steps=3, seed=seed)
+    for p, q in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p, q, rtol=1e-4, atol=1e-6), kw
+
+
+@settings(max_examples=20, deadline=None)
+@given(lr=st.floats(1e-5, 1e-2), b1=st.floats(0.5, 0.95),
+       b2=st.floats(0.9, 0.9999), eps=st.floats(1e-10, 1e-6),
+       wd=st.floats(0.0, 0.05), adamw=st.booleans(),
+       seed=st.integers(0, 10_000))
+def test_fused_adam_property_sweep(lr, b1, b2, eps, wd, adamw, seed):
+    m1, m2 = _pair(seed % 100)
+    kw = dict(lr=lr, betas=(b1, b2), eps=eps, weight_decay=wd)
+    fused = (FusedAdamW if adamw else FusedAdam)(m1.parameters(), **kw)
+    ref = (torch.optim.AdamW if adamw else torch.optim.Adam)(
+        m2.parameters(), **kw)
+    _train(m1, fused, steps=3, seed=seed)
+    _train(m2, ref, steps=3, seed=seed)
+    for p, q in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p, q, rtol=1e-4, atol=1e-6), kw
