@@ -563,3 +563,36 @@ def _bf16_wire(rank, size):
 
 def test_bf16_wire_compression():
     run_distributed(_bf16_wire, 2)
+
+
+def _adasum_properties(rank, size):
+    """Mathematical invariants of the Adasum combine (ws=2): orthogonal
+    gradients ADD (no conflict -> keep both), identical gradients AVERAGE
+    (full conflict -> no double step), and the result is scale-covariant
+    for a common positive factor."""
+    import axon as hvd
+
+    # orthogonal: disjoint supports
+    a = torch.zeros(8, dtype=torch.float64)
+    a[:4] = torch.tensor([1.0, 2.0, -1.0, 0.5], dtype=torch.float64)
+    b = torch.zeros(8, dtype=torch.float64)
+    b[4:] = torch.tensor([3.0, -2.0, 1.0, 1.0], dtype=torch.float64)
+    mine = a if rank == 0 else b
+    out = hvd.allreduce(mine.clone(), op=hvd.Adasum, name="prop.orth")
+    assert torch.allclose(out, a + b, atol=1e-12), out
+
+    # identical: average
+    c = torch.tensor([1.0, -2.0, 3.0, 4.0], dtype=torch.float64)
+    out2 = hvd.allreduce(c.clone(), op=hvd.Adasum, name="prop.same")
+    assert torch.allclose(out2, c, atol=1e-12), out2  # avg of c and c
+
+    # scale covariance: adasum(s*a, s*b) == s * adasum(a, b)
+    g = torch.Generator().manual_seed(7 + rank)
+    d = torch.randn(16, generator=g, dtype=torch.float64)
+    base = hvd.allreduce(d.clone(), op=hvd.Adasum, name="prop.base")
+    scaled = hvd.allreduce(3.5 * d, op=hvd.Adasum, name="prop.scaled")
+    assert torch.allclose(scaled, 3.5 * base, atol=1e-10)
+
+
+def test_adasum_properties_ws2():
+    run_distributed(_adasum_properties, 2)
