@@ -596,3 +596,30 @@ def _adasum_properties(rank, size):
 
 def test_adasum_properties_ws2():
     run_distributed(_adasum_properties, 2)
+
+
+def _alltoall_zero_splits(rank, size):
+    """Zero entries in the splits vector (a rank sends nothing to some
+    peers, possibly nothing at all) must round-trip cleanly."""
+    import axon as hvd
+
+    # rank 0 sends 2 rows to itself only; rank 1 sends 1 row to each;
+    # rank 2 sends everything to rank 0.
+    table = {0: [2, 0, 0], 1: [1, 1, 1], 2: [3, 0, 0]}
+    splits = table[rank]
+    rows = sum(splits)
+    t = torch.full((rows, 2), float(rank), dtype=torch.float32)
+    out, received = hvd.alltoall(t, splits=torch.tensor(splits),
+                                 name="a2a.zeros")
+    expect_rows = sum(table[src][rank] for src in range(size))
+    assert received.tolist() == [table[src][rank] for src in range(size)]
+    assert out.shape[0] == expect_rows
+    off = 0
+    for src in range(size):
+        n = table[src][rank]
+        assert torch.all(out[off:off + n] == float(src)), (rank, src)
+        off += n
+
+
+def test_alltoall_zero_splits_ws3():
+    run_distributed(_alltoall_zero_splits, 3)
