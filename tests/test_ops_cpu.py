@@ -623,3 +623,27 @@ def _alltoall_zero_splits(rank, size):
 
 def test_alltoall_zero_splits_ws3():
     run_distributed(_alltoall_zero_splits, 3)
+
+
+def _degenerate_edges(rank, size):
+    """Empty tensors and fewer rows than ranks must round-trip: empty
+    allreduce/broadcast, reducescatter with rows < world (some ranks own
+    zero rows), and an all-zero-row allgather."""
+    import axon as hvd
+
+    out = hvd.allreduce(torch.empty(0), name="e.ar")
+    assert out.numel() == 0
+    t = torch.full((2, 3), float(rank + 1))
+    rs = hvd.reducescatter(t, op=hvd.Sum, name="e.rs")
+    expect_rows = 1 if rank < 2 else 0
+    assert rs.shape[0] == expect_rows
+    if expect_rows:
+        assert torch.allclose(rs, torch.full((1, 3), 6.0))
+    g = hvd.allgather(torch.zeros(0, 4), name="e.ag")
+    assert g.shape == (0, 4)
+    b = hvd.broadcast(torch.empty(0), root_rank=1, name="e.bc")
+    assert b.numel() == 0
+
+
+def test_degenerate_edges_ws3():
+    run_distributed(_degenerate_edges, 3)
