@@ -647,3 +647,38 @@ def _degenerate_edges(rank, size):
 
 def test_degenerate_edges_ws3():
     run_distributed(_degenerate_edges, 3)
+
+
+def _layout_dtype_edges(rank, size):
+    """Layout/dtype edges: dense-but-transposed tensors allreduce in
+    place correctly (all ranks share strides -> storage order agrees);
+    truly non-dense views work out-of-place via a contiguous copy and
+    raise in place; integer dtypes reduce; multi-MB broadcast_object."""
+    import axon as hvd
+
+    v = (torch.arange(12.0).reshape(3, 4) + rank).t().clone().t()
+    vt = v.t()
+    hvd.allreduce_(vt, op=hvd.Sum, name="nc.ip")
+    assert torch.allclose(vt, (torch.arange(12.0).reshape(3, 4) * 2 + 1).t())
+
+    s = (torch.arange(8.0) + rank)[::2]
+    out = hvd.allreduce(s, op=hvd.Sum, name="nd.oop")
+    assert torch.allclose(out, torch.tensor([1.0, 5, 9, 13]))
+    try:
+        hvd.allreduce_(s, name="nd.ip")
+        raise AssertionError("in-place on strided slice should raise")
+    except ValueError:
+        pass
+
+    i32 = torch.tensor([1, 2, 3], dtype=torch.int32) * (rank + 1)
+    assert hvd.allreduce(i32, op=hvd.Sum, name="i32").tolist() == [3, 6, 9]
+    assert hvd.allreduce(torch.tensor([5 - rank]), op=hvd.Min,
+                         name="i64").item() == 4
+
+    obj = {"blob": bytes(5 << 20), "rank": 0}
+    got = hvd.broadcast_object(obj if rank == 0 else None, root_rank=0)
+    assert len(got["blob"]) == 5 << 20
+
+
+def test_layout_dtype_edges_ws2():
+    run_distributed(_layout_dtype_edges, 2)
