@@ -19,10 +19,12 @@ class Bottleneck(nn.Module):
         out_ch = width * self.expansion
         fuses = norm_layer is FusedBatchNorm2d
         self.conv1 = nn.Conv2d(in_ch, width, 1, bias=False)
-        self.bn1 = norm_layer(width, fuse_relu=True) if fuses             else norm_layer(width)
+        self.bn1 = (norm_layer(width, fuse_relu=True) if fuses
+                    else norm_layer(width))
         self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1,
                                bias=False)
-        self.bn2 = norm_layer(width, fuse_relu=True) if fuses             else norm_layer(width)
+        self.bn2 = (norm_layer(width, fuse_relu=True) if fuses
+                    else norm_layer(width))
         self.conv3 = nn.Conv2d(width, out_ch, 1, bias=False)
         self.bn3 = norm_layer(out_ch)
         self.relu = nn.ReLU(inplace=True)
