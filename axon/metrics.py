@@ -12,6 +12,7 @@ import threading
 _lock = threading.Lock()
 _ops = {}      # op kind -> submissions
 _bytes = {}    # op kind -> tensor bytes submitted
+_lat = {}      # op kind -> [count, sum_seconds, max_seconds]
 _steps = 0     # optimizer steps (DistributedOptimizer.step)
 
 
@@ -19,6 +20,16 @@ def record_op(kind, nbytes):
     with _lock:
         _ops[kind] = _ops.get(kind, 0) + 1
         _bytes[kind] = _bytes.get(kind, 0) + int(nbytes)
+
+
+def record_latency(kind, seconds):
+    """Submit-to-synchronize latency of one completed op (includes the
+    caller's wait; recorded by mpi_ops.synchronize)."""
+    with _lock:
+        e = _lat.setdefault(kind, [0, 0.0, 0.0])
+        e[0] += 1
+        e[1] += seconds
+        e[2] = max(e[2], seconds)
 
 
 def record_step():
@@ -29,9 +40,11 @@ def record_step():
 
 def collect():
     """Snapshot: {"ops": {kind: n}, "bytes": {kind: n},
-    "optimizer_steps": n}."""
+    "latency": {kind: {"count", "sum", "max"}}, "optimizer_steps": n}."""
     with _lock:
         return {"ops": dict(_ops), "bytes": dict(_bytes),
+                "latency": {k: {"count": v[0], "sum": v[1], "max": v[2]}
+                            for k, v in _lat.items()},
                 "optimizer_steps": _steps}
 
 
@@ -40,6 +53,7 @@ def reset():
     with _lock:
         _ops.clear()
         _bytes.clear()
+        _lat.clear()
         _steps = 0
 
 
@@ -48,7 +62,8 @@ def start_http_server(port, addr="127.0.0.1"):
     format). Returns the (server, thread) pair from prometheus_client."""
     from prometheus_client import start_http_server as _start
     from prometheus_client.core import (CounterMetricFamily, REGISTRY,
-                                        GaugeMetricFamily)
+                                        GaugeMetricFamily,
+                                        SummaryMetricFamily)
 
     from axon import basics
 
@@ -66,6 +81,19 @@ def start_http_server(port, addr="127.0.0.1"):
             for k, v in snap["bytes"].items():
                 byt.add_metric([k], v)
             yield byt
+            lat = SummaryMetricFamily(
+                "axon_op_latency_seconds",
+                "submit-to-synchronize latency", labels=["op"])
+            for k, v in snap["latency"].items():
+                lat.add_metric([k], count_value=v["count"],
+                               sum_value=v["sum"])
+            yield lat
+            lmax = GaugeMetricFamily("axon_op_latency_max_seconds",
+                                     "max observed op latency",
+                                     labels=["op"])
+            for k, v in snap["latency"].items():
+                lmax.add_metric([k], v["max"])
+            yield lmax
             yield CounterMetricFamily("axon_optimizer_steps",
                                       "DistributedOptimizer steps",
                                       value=snap["optimizer_steps"])

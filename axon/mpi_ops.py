@@ -6,6 +6,7 @@ API"). Ops are enqueued to the C++ core (axon/_core.so) which coordinates
 order across ranks and runs RCCL on a dedicated HIP stream.
 """
 import threading
+import time
 
 import torch
 
@@ -51,9 +52,9 @@ def _auto_name(kind, name, psid=0):
     return base if psid == 0 else f"ps{psid}:{base}"
 
 
-def _register(handle, payload):
+def _register(handle, payload, kind=None):
     with _lock:
-        _outputs[handle] = payload
+        _outputs[handle] = (payload, kind, time.perf_counter())
     return handle
 
 
@@ -102,7 +103,7 @@ def allreduce_async(tensor, average=None, name=None, op=None,
     h = core.allreduce_async(t, output, _auto_name("allreduce", name, psid),
                              red, prescale_factor, postscale_factor,
                              _wire(compression), psid)
-    return _register(h, output)
+    return _register(h, output, 'allreduce')
 
 
 class _DiffAllreduce(torch.autograd.Function):
@@ -153,7 +154,7 @@ def allreduce_async_(tensor, average=None, name=None, op=None,
                              _auto_name("allreduce", name, psid), red,
                              prescale_factor, postscale_factor,
                              _wire(compression), psid)
-    return _register(h, tensor)
+    return _register(h, tensor, 'allreduce')
 
 
 def allreduce_(tensor, average=None, name=None, op=None, prescale_factor=1.0,
@@ -208,7 +209,7 @@ def broadcast_async(tensor, root_rank, name=None, process_set=None):
     core = _ensure_init()
     h = core.broadcast_async(t, output, root_rank,
                              _auto_name("broadcast", name, psid), psid)
-    return _register(h, output)
+    return _register(h, output, 'broadcast')
 
 
 class _DiffBroadcast(torch.autograd.Function):
@@ -245,7 +246,7 @@ def broadcast_async_(tensor, root_rank, name=None, process_set=None):
     core = _ensure_init()
     h = core.broadcast_async(tensor, tensor, root_rank,
                              _auto_name("broadcast", name, psid), psid)
-    return _register(h, tensor)
+    return _register(h, tensor, 'broadcast')
 
 
 def broadcast_(tensor, root_rank, name=None, process_set=None):
@@ -262,7 +263,7 @@ def allgather_async(tensor, name=None, process_set=None):
     core = _ensure_init()
     h = core.allgather_async(t, output, _auto_name("allgather", name, psid),
                              psid)
-    return _register(h, output)
+    return _register(h, output, 'allgather')
 
 
 class _DiffAllgather(torch.autograd.Function):
@@ -317,7 +318,7 @@ def alltoall_async(tensor, splits=None, name=None, process_set=None):
     core = _ensure_init()
     h = core.alltoall_async(t, output, splits_list, received_splits,
                             _auto_name("alltoall", name, psid), psid)
-    return _register(h, (output, received_splits))
+    return _register(h, (output, received_splits), 'alltoall')
 
 
 class _DiffAlltoall(torch.autograd.Function):
@@ -366,7 +367,7 @@ def reducescatter_async(tensor, op=ReduceOp.Average, name=None,
     h = core.reducescatter_async(t, output,
                                  _auto_name("reducescatter", name, psid),
                                  op, prescale_factor, postscale_factor, psid)
-    return _register(h, output)
+    return _register(h, output, 'reducescatter')
 
 
 class _DiffReducescatter(torch.autograd.Function):
@@ -427,7 +428,13 @@ def synchronize(handle):
     core = _ensure_init()
     core.synchronize(handle)
     with _lock:
-        return _outputs.pop(handle, None)
+        entry = _outputs.pop(handle, None)
+    if entry is None:
+        return None
+    payload, kind, t0 = entry
+    if kind is not None:
+        metrics.record_latency(kind, time.perf_counter() - t0)
+    return payload
 
 
 def poll(handle):

@@ -19,6 +19,9 @@ def _body(rank, size):
     assert snap["ops"]["allreduce"] == 2
     assert snap["bytes"]["allreduce"] == 150 * 4
     assert snap["ops"]["broadcast"] == 1
+    lat = snap["latency"]["allreduce"]
+    assert lat["count"] == 2 and lat["sum"] > 0
+    assert lat["max"] >= lat["sum"] / lat["count"]
 
     if rank == 0:
         port = 21000 + (uuid.uuid4().int % 9000)
@@ -27,6 +30,7 @@ def _body(rank, size):
             f"http://127.0.0.1:{port}/metrics", timeout=10).read().decode()
         assert 'axon_ops_total{op="allreduce"} 2.0' in text, text[-800:]
         assert 'axon_bytes_total{op="allreduce"} 600.0' in text
+        assert 'axon_op_latency_seconds_count{op="allreduce"} 2.0' in text
         assert "axon_world_size 2.0" in text
     hvd.barrier(name="metrics.done")
 
