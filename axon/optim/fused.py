@@ -34,6 +34,13 @@ class FusedSGD(Optimizer):
                  weight_decay=0.0, nesterov=False, maximize=False):
         if nesterov and (momentum <= 0 or dampening != 0):
             raise ValueError("nesterov requires momentum > 0, dampening 0")
+        if dampening != 0 and 0 < momentum < 1e-8:
+            # The fused kernel's uniform m = mu*m + (1-damp)*d update
+            # reproduces torch's undamped first step via a (damp/mu)-scaled
+            # seed buffer, which overflows for denormal momentum. No real
+            # configuration pairs ~zero momentum with dampening.
+            raise ValueError("fused SGD: momentum in (0, 1e-8) with "
+                             "nonzero dampening is unsupported")
         super().__init__(params, dict(lr=lr, momentum=momentum,
                                       dampening=dampening,
                                       weight_decay=weight_decay,

@@ -100,7 +100,8 @@ from hypothesis import given, settings, strategies as st  # noqa: E402
 
 
 @settings(max_examples=25, deadline=None)
-@given(lr=st.floats(1e-4, 0.5), momentum=st.floats(0.0, 0.99),
+@given(lr=st.floats(1e-4, 0.5),
+       momentum=st.one_of(st.just(0.0), st.floats(1e-3, 0.99)),
        dampening=st.floats(0.0, 0.9), wd=st.floats(0.0, 0.05),
        maximize=st.booleans(), seed=st.integers(0, 10_000))
 def test_fused_sgd_property_sweep(lr, momentum, dampening, wd, maximize,
