@@ -99,7 +99,7 @@ def test_fused_state_dict_interchange():
 from hypothesis import given, settings, strategies as st  # noqa: E402
 
 
-@settings(max_examples=25, deadline=None)
+@settings(max_examples=25, deadline=None, derandomize=True)
 @given(lr=st.floats(1e-4, 0.5),
        momentum=st.one_of(st.just(0.0), st.floats(1e-3, 0.99)),
        dampening=st.floats(0.0, 0.9), wd=st.floats(0.0, 0.05),
@@ -115,7 +115,7 @@ def test_fused_sgd_property_sweep(lr, momentum, dampening, wd, maximize,
         assert torch.allclose(p, q, rtol=1e-4, atol=1e-6), kw
 
 
-@settings(max_examples=20, deadline=None)
+@settings(max_examples=20, deadline=None, derandomize=True)
 @given(lr=st.floats(1e-5, 1e-2), b1=st.floats(0.5, 0.95),
        b2=st.floats(0.9, 0.9999), eps=st.floats(1e-10, 1e-6),
        wd=st.floats(0.0, 0.05), adamw=st.booleans(),
