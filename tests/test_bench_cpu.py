@@ -83,3 +83,30 @@ def test_bench_resnet_cpu_ws8():
     rec = json.loads(payloads[0])
     assert rec["n_gpus"] == 8
     assert rec["config"]["global_batch"] == 8
+
+
+def test_bench_json_contract():
+    """The driver parses one JSON line with these exact keys; pin them so
+    a future edit can't silently break the contract."""
+    port = str(20000 + (uuid.uuid4().int % 20000))
+    env = os.environ.copy()
+    env.update({"RANK": "0", "WORLD_SIZE": "1", "LOCAL_RANK": "0",
+                "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": port})
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "1", "--warmup", "0",
+         "--batch", "1", "--image-size", "64"],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stdout + r.stderr
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][0]
+    rec = json.loads(line)
+    required = {"metric": str, "value": (int, float), "unit": str,
+                "n_gpus": int, "steps": int, "warmup": int,
+                "ms_per_step": (int, float), "higher_is_better": bool,
+                "scaling": str, "dtype": str, "data": str, "config": dict}
+    for k, t in required.items():
+        assert k in rec and isinstance(rec[k], t), (k, rec.get(k))
+    assert "vs_baseline" in rec  # null allowed (no published baseline)
+    assert rec["scaling"] == "weak"
+    assert rec["data"] == "synthetic"
+    assert rec["config"]["model"] == "resnet50"
+    assert "global_batch" in rec["config"]
