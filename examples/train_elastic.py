@@ -23,6 +23,11 @@ torch.manual_seed(0)
 model = LeNet()
 opt = torch.optim.SGD(model.parameters(), lr=0.05, momentum=0.9)
 state = elastic.TorchState(model=model, optimizer=opt, step=0)
+# Reset callbacks fire on survivors after every membership change —
+# the place to re-shard samplers (axon.data.ElasticSampler.set_world)
+# or rescale the learning rate to the new world size.
+state.register_reset_callbacks(
+    [lambda: print(f"[train] new world: ws={hvd.size()}", flush=True)])
 
 
 @elastic.run
