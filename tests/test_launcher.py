@@ -241,3 +241,17 @@ def test_multi_host_rejected():
         capture_output=True, text=True, timeout=120,
         env={**os.environ, "PYTHONPATH": REPO})
     assert r2.returncode == 0, r2.stderr
+
+
+def test_sweep_tool_cpu_smoke():
+    """tools/sweep.py: one-point sweep end-to-end on CPU (the xGMI
+    re-tuning deliverable's harness must stay runnable)."""
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tools", "sweep.py"),
+         "--model", "resnet50", "--gpus", "1", "--steps", "1",
+         "--warmup", "0", "--batch", "1", "--thresholds", "32",
+         "--cycles", "1"],
+        capture_output=True, text=True, timeout=300,
+        env={**os.environ, "PYTHONPATH": REPO})
+    assert r.returncode == 0, r.stderr[-1500:]
+    assert "best: thr=32MiB" in r.stdout, r.stdout
