@@ -27,6 +27,17 @@ import time
 import uuid
 
 
+def _redirect_kwargs(args, rank):
+    """--output-filename FILE -> ranks > 0 append stdout+stderr to
+    FILE.rank<N>.log (horovodrun's per-rank log capture). Rank 0 stays on
+    the console so single-line stdout contracts (e.g. the bench JSON
+    line) are never swallowed."""
+    if not args.output_filename or rank == 0:
+        return {}
+    f = open(f"{args.output_filename}.rank{rank}.log", "ab", buffering=0)
+    return {"stdout": f, "stderr": subprocess.STDOUT}
+
+
 def main(argv=None):
     parser = argparse.ArgumentParser(
         prog="axonrun", description="axon single-node launcher")
@@ -41,6 +52,10 @@ def main(argv=None):
                         help="do not pin ranks to GPUs via HIP_VISIBLE_DEVICES")
     parser.add_argument("--timeline", default=None,
                         help="write a chrome-trace timeline to this path")
+    parser.add_argument("--output-filename", default=None,
+                        help="redirect each non-zero rank's stdout+stderr "
+                             "to FILE.rank<N>.log (horovodrun parity); "
+                             "rank 0 stays on the console")
     parser.add_argument("--fusion-threshold-mb", type=int, default=None,
                         help="override AXON_FUSION_THRESHOLD (MiB)")
     parser.add_argument("--cycle-time-ms", type=float, default=None,
@@ -99,7 +114,8 @@ def main(argv=None):
         if args.verbose:
             print(f"[axonrun] rank {r}: {' '.join(args.command)}", file=sys.stderr)
         procs.append(subprocess.Popen(args.command, env=env,
-                                      start_new_session=True))
+                                      start_new_session=True,
+                                      **_redirect_kwargs(args, r)))
 
     def kill_all(sig=signal.SIGTERM):
         for p in procs:
@@ -182,7 +198,8 @@ def run_elastic(args):
         if args.cycle_time_ms is not None:
             env["AXON_CYCLE_TIME"] = str(args.cycle_time_ms)
         procs[worker] = subprocess.Popen(args.command, env=env,
-                                         start_new_session=True)
+                                         start_new_session=True,
+                                         **_redirect_kwargs(args, worker))
 
     def kill_all(sig=signal.SIGTERM):
         for p in procs.values():

@@ -255,3 +255,24 @@ def test_sweep_tool_cpu_smoke():
         env={**os.environ, "PYTHONPATH": REPO})
     assert r.returncode == 0, r.stderr[-1500:]
     assert "best: thr=32MiB" in r.stdout, r.stdout
+
+
+def test_output_filename_per_rank_logs():
+    with tempfile.TemporaryDirectory() as d:
+        base = os.path.join(d, "job")
+        script = (
+            "import axon as hvd\n"
+            "hvd.init()\n"
+            "print(f'hello from rank {hvd.rank()}', flush=True)\n"
+            "hvd.shutdown()\n")
+        r = subprocess.run(
+            [sys.executable, "-m", "axon.runner.run", "-np", "2",
+             "--no-pin", "--output-filename", base,
+             sys.executable, "-c", script],
+            capture_output=True, text=True, timeout=120,
+            env={**os.environ, "PYTHONPATH": REPO})
+        assert r.returncode == 0, r.stderr
+        assert "hello from rank 0" in r.stdout  # rank 0 on the console
+        with open(base + ".rank1.log") as f:
+            assert "hello from rank 1" in f.read()
+        assert "hello from rank 1" not in r.stdout
