@@ -276,3 +276,28 @@ def test_output_filename_per_rank_logs():
         with open(base + ".rank1.log") as f:
             assert "hello from rank 1" in f.read()
         assert "hello from rank 1" not in r.stdout
+
+
+def test_timeline_report_tool():
+    """tools/timeline_report.py parses a real ws=2 timeline and reports
+    phase stats + comm-busy fraction."""
+    with tempfile.TemporaryDirectory() as d:
+        path = os.path.join(d, "tl.json")
+        script = (
+            "import torch, axon as hvd\n"
+            "hvd.init()\n"
+            "for i in range(4):\n"
+            "    hvd.allreduce(torch.ones(64), name=f'r{i}')\n"
+            "hvd.shutdown()\n")
+        r = _run(2, script, extra_args=("--timeline", path))
+        assert r.returncode == 0, r.stderr
+        rep = subprocess.run(
+            [sys.executable, os.path.join(REPO, "tools", "timeline_report.py"),
+             "--json", os.path.join(d, "tl.0.json")],
+            capture_output=True, text=True, timeout=60)
+        assert rep.returncode == 0, rep.stderr
+        doc = json.loads(rep.stdout)
+        assert doc["phases"]["CPU_OP"]["count"] == 4
+        assert doc["phases"]["NEGOTIATE"]["count"] == 4
+        assert 0 <= doc["comm_busy_frac"] <= 1
+        assert any(t["tensor"].endswith("r0") for t in doc["top_tensors"])
