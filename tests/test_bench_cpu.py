@@ -110,3 +110,20 @@ def test_bench_json_contract():
     assert rec["data"] == "synthetic"
     assert rec["config"]["model"] == "resnet50"
     assert "global_batch" in rec["config"]
+
+
+def test_bench_under_real_torchrun_ws2():
+    """The driver's exact launch form: torch.distributed.run --nnodes=1
+    --nproc-per-node N --master-addr 127.0.0.1 ... bench.py."""
+    port = str(20000 + (uuid.uuid4().int % 20000))
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", port, "bench.py", "--gpus", "2", "--steps", "1",
+         "--warmup", "0", "--batch", "1", "--image-size", "64"],
+        cwd=REPO, env=os.environ.copy(), capture_output=True, text=True,
+        timeout=600)
+    assert r.returncode == 0, (r.stdout + r.stderr)[-2000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][0]
+    rec = json.loads(line)
+    assert rec["n_gpus"] == 2 and rec["config"]["parallelism"] == "dp2"
