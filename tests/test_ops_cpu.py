@@ -682,3 +682,37 @@ def _layout_dtype_edges(rank, size):
 
 def test_layout_dtype_edges_ws2():
     run_distributed(_layout_dtype_edges, 2)
+
+
+def _alltoall_bad_splits(rank, size):
+    """splits vector length != world size must surface the controller's
+    validation error on every rank (not hang)."""
+    import axon as hvd
+
+    t = torch.ones(4, 2)
+    try:
+        hvd.alltoall(t, splits=torch.tensor([2, 2, 0]), name="bad.a2a")
+        raise AssertionError("expected splits-size error")
+    except RuntimeError as e:
+        assert "splits" in str(e), e
+
+
+def test_alltoall_bad_splits_ws2():
+    run_distributed(_alltoall_bad_splits, 2)
+
+
+def _broadcast_bad_root(rank, size):
+    """broadcast root outside the process set errors cleanly."""
+    import axon as hvd
+
+    ps = hvd.add_process_set([0, 1])
+    t = torch.ones(3)
+    try:
+        hvd.broadcast(t, root_rank=7, name="bad.bc", process_set=ps)
+        raise AssertionError("expected bad-root error")
+    except RuntimeError as e:
+        assert "root" in str(e), e
+
+
+def test_broadcast_bad_root_ws2():
+    run_distributed(_broadcast_bad_root, 2)
