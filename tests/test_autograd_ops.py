@@ -103,3 +103,32 @@ def _grouped_allreduce_grads(rank, size):
 
 def test_grouped_allreduce_autograd():
     run_distributed(_grouped_allreduce_grads, 2)
+
+
+def _subset_autograd(rank, size):
+    """Differentiable collectives over a process SET (ws=3, members
+    {0, 2}): the backward must use set-relative ranks/sizes."""
+    import axon as hvd
+
+    ps = hvd.add_process_set([0, 2])
+    if not ps.included():
+        hvd.barrier(name="subset.done")
+        return
+    mem = ps.rank()  # 0 for rank0, 1 for rank2
+    x = torch.full((2, 2), float(mem + 1), requires_grad=True)
+    out = hvd.allreduce(x, op=hvd.Average, name="ps.ar", process_set=ps)
+    assert torch.allclose(out.detach(), torch.full((2, 2), 1.5))
+    out.sum().backward()
+    assert torch.allclose(x.grad, torch.ones(2, 2)), x.grad
+
+    y = torch.full((1 + mem, 3), float(mem), requires_grad=True)
+    g = hvd.allgather(y, name="ps.ag", process_set=ps)
+    assert g.shape == (3, 3)
+    g.pow(2).sum().backward()
+    # 2 members' losses each see my shard: grad = 2 * 2 * y
+    assert torch.allclose(y.grad, 4.0 * y.detach()), y.grad
+    hvd.barrier(name="subset.done")
+
+
+def test_subset_autograd_ws3():
+    run_distributed(_subset_autograd, 3)
