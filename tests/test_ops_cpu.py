@@ -716,3 +716,24 @@ def _broadcast_bad_root(rank, size):
 
 def test_broadcast_bad_root_ws2():
     run_distributed(_broadcast_bad_root, 2)
+
+
+def _mixed_dtype_group(rank, size):
+    """Grouped allreduce over mixed dtypes: the controller fuses only
+    same-(dtype, wire) runs, so results stay exact per tensor (with and
+    without an fp16 wire)."""
+    import axon as hvd
+
+    ts = [torch.ones(10, dtype=torch.float32) * (rank + 1),
+          torch.ones(5, dtype=torch.float64) * (rank + 1),
+          torch.ones(7, dtype=torch.float32) * (rank + 1)]
+    for o in hvd.grouped_allreduce(ts, op=hvd.Sum, name="mixed"):
+        assert torch.allclose(o.float(), torch.full_like(o.float(), 3.0))
+    for o in hvd.grouped_allreduce(ts, op=hvd.Sum, name="mixedw",
+                                   compression=hvd.Compression.fp16):
+        assert torch.allclose(o.float(), torch.full_like(o.float(), 3.0),
+                              atol=1e-2)
+
+
+def test_mixed_dtype_group_ws2():
+    run_distributed(_mixed_dtype_group, 2)
