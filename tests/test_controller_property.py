@@ -102,3 +102,45 @@ def _cache_churn(rank, size, seed):
 
 def test_response_cache_churn_ws3():
     run_distributed(_cache_churn, 3, args=(11,))
+
+
+def _mixed_set_ops(rank, size, seed):
+    """Global and process-set collectives interleaved in DIFFERENT
+    per-rank submission orders: set-aware readiness must keep the global
+    response order consistent and every result correct (ws=3, set {0,2})."""
+    import time
+
+    import axon as hvd
+
+    ps = hvd.add_process_set([0, 2])
+    rng = random.Random(seed)
+    ops = []
+    for i in range(10):
+        ops.append((i, rng.choice(["global", "subset"]),
+                    rng.randint(1, 300)))
+    order = list(range(len(ops)))
+    random.Random(seed * 31 + rank).shuffle(order)
+    jitter = random.Random(seed * 7 + rank)
+    handles = {}
+    for idx in order:
+        i, kind, numel = ops[idx]
+        if jitter.random() < 0.4:
+            time.sleep(jitter.random() * 0.003)
+        t = torch.full((numel,), float(rank + 1))
+        if kind == "global":
+            handles[i] = ("g", hvd.allreduce_async(t, op=hvd.Sum,
+                                                   name=f"mx{i}"), numel)
+        elif ps.included():
+            handles[i] = ("s", hvd.allreduce_async(t, op=hvd.Sum,
+                                                   name=f"mx{i}",
+                                                   process_set=ps), numel)
+    for i, (kind, h, numel) in sorted(handles.items()):
+        out = hvd.synchronize(h)
+        expect = 6.0 if kind == "g" else 4.0  # ranks 1+2+3 vs 1+3
+        assert torch.allclose(out, torch.full((numel,), expect)), (i, kind)
+    hvd.barrier(name="mx.done")
+
+
+def test_mixed_set_randomized_orders_ws3():
+    for seed in (3, 17):
+        run_distributed(_mixed_set_ops, 3, args=(seed,))
