@@ -14,6 +14,7 @@
 #include "core/gpu_util.h"
 #include "core/message.h"
 #include "core/operations.h"
+#include "core/parameter_manager.h"
 
 namespace axon {
 namespace {
@@ -245,6 +246,20 @@ PYBIND11_MODULE(_core, m) {
         py::arg("prescale") = 1.0, py::arg("postscale") = 1.0,
         py::arg("process_set") = 0);
   m.def("autotune_step_mark", &axon::pm_step_mark);
+  m.def("_autotune_bo_propose",
+        [](const std::vector<std::tuple<int64_t, double, double>>& obs)
+            -> pybind11::object {
+          // (threshold_bytes, cycle_ms, score) triples -> EI-max proposal
+          // (deterministic; exposed for the unit test of the GP/EI math).
+          std::vector<axon::ParameterManager::Sample> v;
+          for (auto& t : obs)
+            v.push_back({std::get<0>(t), std::get<1>(t), std::get<2>(t)});
+          int64_t thr;
+          double cyc;
+          if (!axon::autotune_bo_propose(v, &thr, &cyc))
+            return pybind11::none();
+          return pybind11::make_tuple(thr, cyc);
+        });
   m.def("start_timeline", &axon::timeline_start, pybind11::arg("path"));
   m.def("stop_timeline", &axon::timeline_stop);
   m.def("ps_register", &axon::ps_register, py::arg("ranks"));

@@ -1,12 +1,16 @@
 // Online autotuner for the two engine knobs: fusion threshold (bytes) and
 // cycle time (ms) — capability parity with the reference's ParameterManager
-// (SURVEY.md §2.1 "Parameter manager (autotuner)"), re-designed as a
-// coordinator-driven grid hill-climb scored end-to-end by optimizer
-// steps/sec (reported by DistributedOptimizer.step via autotune_step_mark;
-// falls back to allreduced bytes/sec for op-only workloads), followed by
-// refinement rounds around the best point:
-// xGMI-era defaults are already large (128 MiB), so the tuner explores a
-// coarse 2-D neighborhood instead of running full Bayesian optimization.
+// + BayesianOptimization (SURVEY.md §2.1 "Parameter manager (autotuner)"),
+// scored end-to-end by optimizer steps/sec (reported by
+// DistributedOptimizer.step via autotune_step_mark; falls back to
+// allreduced bytes/sec for op-only workloads). Three phases:
+//   1. coarse 4x3 grid over (threshold, cycle);
+//   2. AXON_AUTOTUNE_BO (default 6) Bayesian-optimization rounds — a GP
+//      with an RBF kernel over (log2 threshold, log2 cycle) fits all
+//      observations and the expected-improvement maximum over a
+//      half-octave lattice is sampled next;
+//   3. optional legacy 3x3 refinement (AXON_AUTOTUNE_REFINE) around the
+//      incumbent.
 //
 // Enable with AXON_AUTOTUNE=1; AXON_AUTOTUNE_LOG=1 logs every sample.
 // Runs on rank 0 only: the fusion threshold is a coordinator decision, and
@@ -33,12 +37,13 @@ class ParameterManager {
   // changed; outputs the new values (to apply locally + broadcast).
   bool maybe_update(int64_t* threshold, double* cycle_ms);
 
- private:
   struct Sample {
     int64_t threshold;
     double cycle_ms;
     double score = -1.0;
   };
+
+ private:
   void advance();
 
   void refine();  // build a finer plan around the current best
@@ -46,6 +51,9 @@ class ParameterManager {
   bool enabled_;
   bool log_;
   bool done_ = false;
+  int bo_rounds_ = 6;
+  int bo_done_ = 0;
+  std::vector<Sample> history_;  // every scored sample (GP observations)
   int64_t bytes_ = 0;
   int64_t steps_ = 0;
   bool using_steps_ = false;
@@ -60,5 +68,14 @@ class ParameterManager {
   Sample best_;
   bool started_ = false;
 };
+
+// Bayesian-optimization proposal (exposed for deterministic unit tests):
+// fit a fixed-hyperparameter GP (RBF over normalized log2 coordinates) to
+// the scored samples and return the expected-improvement maximizer over a
+// half-octave (threshold, cycle) lattice that is not already observed.
+// Returns false when every lattice point has been observed or inputs are
+// degenerate.
+bool autotune_bo_propose(const std::vector<ParameterManager::Sample>& obs,
+                         int64_t* threshold, double* cycle_ms);
 
 }  // namespace axon

@@ -181,13 +181,14 @@ def test_autotune_converges():
         "import torch, axon as hvd\n"
         "hvd.init()\n"
         "t = torch.ones(1 << 16)\n"
-        "for i in range(400):\n"
+        "for i in range(1200):\n"
         "    hvd.allreduce(t, average=True, name=f'a{i % 7}')\n"
         "hvd.shutdown()\n")
     r = _run(2, script, env={"AXON_AUTOTUNE": "1", "AXON_AUTOTUNE_LOG": "1",
                              "AXON_AUTOTUNE_SAMPLE_SEC": "0.05",
                              "AXON_AUTOTUNE_WARMUP_SEC": "0.1",
                              "AXON_AUTOTUNE_REFINE": "0",
+                             "AXON_AUTOTUNE_BO": "2",
                              "AXON_LOG_LEVEL": "info"}, timeout=150)
     assert r.returncode == 0, r.stderr
     assert "autotune sample" in r.stderr
@@ -301,3 +302,35 @@ def test_timeline_report_tool():
         assert doc["phases"]["NEGOTIATE"]["count"] == 4
         assert 0 <= doc["comm_busy_frac"] <= 1
         assert any(t["tensor"].endswith("r0") for t in doc["top_tensors"])
+
+
+def test_autotune_bayesian_proposals():
+    """The GP/EI proposal engine (phase 2 of the tuner): seeded with the
+    coarse grid over a smooth score surface whose optimum lies OFF the
+    grid, the loop must find a half-octave-accurate optimum within a few
+    proposals (here: the very first proposal hits it)."""
+    import math
+
+    from axon import _core
+
+    MiB = 1 << 20
+
+    def score(thr, cyc):
+        d = (math.log2(thr / MiB) - 7.5) ** 2 + (math.log2(cyc) - 0.5) ** 2
+        return 100.0 - 10.0 * d  # optimum (181 MiB, 1.41 ms), off-grid
+
+    obs = [(t * MiB, c, score(t * MiB, c))
+           for t in (32, 64, 128, 256) for c in (0.5, 1.0, 2.0)]
+    grid_best = max(o[2] for o in obs)
+    for _ in range(8):
+        p = _core._autotune_bo_propose(obs)
+        if p is None:
+            break
+        obs.append((p[0], p[1], score(p[0], p[1])))
+    best = max(obs, key=lambda o: o[2])
+    assert best[2] > grid_best  # BO beat the grid
+    assert abs(math.log2(best[0] / MiB) - 7.5) <= 0.5
+    assert abs(math.log2(best[1]) - 0.5) <= 0.5
+    # degenerate flat scores: no model, no proposal
+    flat = [(64 * MiB, 1.0, 5.0), (128 * MiB, 1.0, 5.0)]
+    assert _core._autotune_bo_propose(flat) is None
