@@ -248,3 +248,46 @@ def _amp_scaler_pattern(rank, size):
 
 def test_amp_grad_scaler_pattern_ws2():
     run_distributed(_amp_scaler_pattern, 2)
+
+
+def _accum_fp16_combo(rank, size):
+    """Interaction coverage: gradient accumulation (bpps=2) + fp16 wire
+    compression together — the allreduce fires on the second backward
+    with the ACCUMULATED grad, cast on the wire, and the result matches
+    a single-process run on all shards at fp16-wire tolerance."""
+    import axon as hvd
+
+    model = _make_model()
+    hvd.broadcast_parameters(model.state_dict(), root_rank=0)
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    opt = hvd.DistributedOptimizer(opt,
+                                   named_parameters=model.named_parameters(),
+                                   backward_passes_per_step=2,
+                                   compression=hvd.Compression.fp16)
+    for micro in range(2):
+        x, y = _data_for(rank, seed=500 + micro)
+        loss = nn.functional.mse_loss(model(x), y)
+        loss.backward()
+    opt.step()
+    opt.zero_grad()
+
+    # single-process oracle: mean over the 4 micro-batches' grads
+    ref = _make_model()
+    xs, ys = [], []
+    for r in range(size):
+        for micro in range(2):
+            x, y = _data_for(r, seed=500 + micro)
+            xs.append(x)
+            ys.append(y)
+    ref_opt = torch.optim.SGD(ref.parameters(), lr=0.05)
+    ref_opt.zero_grad()
+    for x, y in zip(xs, ys):
+        # accumulated-sum-then-averaged-over-ranks == mean of per-rank sums
+        (nn.functional.mse_loss(ref(x), y) / size).backward()
+    ref_opt.step()
+    for p, q in zip(model.parameters(), ref.parameters()):
+        assert torch.allclose(p, q, rtol=2e-3, atol=2e-3), (p - q).abs().max()
+
+
+def test_accumulation_with_fp16_wire_ws2():
+    run_distributed(_accum_fp16_combo, 2)
