@@ -156,3 +156,34 @@ def test_vanilla_checkpoint_interchange():
     with tempfile.TemporaryDirectory() as d:
         run_distributed(_vanilla_interchange, 2,
                         args=(os.path.join(d, "ckpt.pt"),))
+
+
+def _fused_opt_state_broadcast(rank, size):
+    """broadcast_optimizer_state over a FUSED optimizer's state dict
+    (exp_avg/exp_avg_sq/step tensors): non-root states rebuilt + equal."""
+    import axon as hvd
+    from axon.optim import FusedAdamW
+
+    model = _model(seed=4)
+    hvd.broadcast_parameters(model.state_dict(), root_rank=0)
+    opt = FusedAdamW(model.parameters(), lr=1e-3)
+    if rank == 0:  # only root trains; others start stateless
+        for _ in range(2):
+            opt.zero_grad()
+            (model(torch.ones(2, 4)).sum()).backward()
+            opt.step()
+    hvd.broadcast_optimizer_state(opt, root_rank=0)
+    sums = []
+    for p in model.parameters():
+        st = opt.state[p]
+        assert float(st["step"]) == 2
+        sums.append(float(st["exp_avg"].double().sum())
+                    + float(st["exp_avg_sq"].double().sum()))
+    t = torch.tensor(sums, dtype=torch.float64)
+    mx = hvd.allreduce(t, op=hvd.Max, name="fs.max")
+    mn = hvd.allreduce(t, op=hvd.Min, name="fs.min")
+    assert torch.allclose(mx, mn, atol=0, rtol=0), (mx - mn)
+
+
+def test_fused_optimizer_state_broadcast():
+    run_distributed(_fused_opt_state_broadcast, 2)
