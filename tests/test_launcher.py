@@ -334,3 +334,35 @@ def test_autotune_bayesian_proposals():
     # degenerate flat scores: no model, no proposal
     flat = [(64 * MiB, 1.0, 5.0), (128 * MiB, 1.0, 5.0)]
     assert _core._autotune_bo_propose(flat) is None
+
+
+def test_autotune_bo_fuzz_invariants():
+    """Fuzz the GP/EI proposal: for arbitrary observation sets the
+    proposal is either None or an in-range lattice point that was not
+    already observed; never NaN, never out of the knob envelope."""
+    import math
+    import random
+
+    from axon import _core
+
+    MiB = 1 << 20
+    rng = random.Random(7)
+    for trial in range(40):
+        n = rng.randint(0, 12)
+        obs = []
+        for _ in range(n):
+            thr = int(2 ** rng.uniform(3, 10) * MiB)
+            cyc = 2 ** rng.uniform(-2, 2)
+            score = rng.choice([rng.uniform(-1e9, 1e9),
+                                rng.uniform(0, 1), 0.0])
+            obs.append((thr, cyc, score))
+        p = _core._autotune_bo_propose(obs)
+        if p is None:
+            continue
+        thr, cyc = p
+        assert 8 * MiB <= thr <= 1024 * MiB + MiB
+        assert 0.24 <= cyc <= 4.01
+        assert math.isfinite(cyc)
+        for o in obs:  # never re-proposes an observed point
+            assert not (abs(math.log2(max(1, o[0]) / thr)) < 1e-6 and
+                        abs(math.log2(o[1] / cyc)) < 1e-6), (o, p)
