@@ -737,3 +737,31 @@ def _mixed_dtype_group(rank, size):
 
 def test_mixed_dtype_group_ws2():
     run_distributed(_mixed_dtype_group, 2)
+
+
+def _poll_semantics(rank, size):
+    """hvd.poll(handle): False while a peer withholds its submission,
+    True after completion; synchronize still returns the result."""
+    import time
+
+    import axon as hvd
+
+    t = torch.full((4,), float(rank + 1))
+    if rank == 0:
+        h = hvd.allreduce_async(t, op=hvd.Sum, name="pollme")
+        assert hvd.poll(h) is False  # rank 1 hasn't submitted yet
+        time.sleep(0.3)
+        assert hvd.poll(h) is False
+    hvd.barrier(name="poll.gate")
+    if rank != 0:
+        h = hvd.allreduce_async(t, op=hvd.Sum, name="pollme")
+    deadline = time.time() + 30
+    while not hvd.poll(h) and time.time() < deadline:
+        time.sleep(0.01)
+    assert hvd.poll(h) is True
+    out = hvd.synchronize(h)
+    assert torch.allclose(out, torch.full((4,), 3.0))
+
+
+def test_poll_semantics_ws2():
+    run_distributed(_poll_semantics, 2)
