@@ -19,6 +19,18 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: test requires an AMD GPU (MI355X)")
+    # Keep the pytest process itself single-threaded: run_distributed
+    # forks children from it, and threads created by in-process tests
+    # (torch intra-op pools, HTTP test clients) can leave locked mutexes
+    # behind in the fork — the occasional 90s child deadlock. Children
+    # re-assert this for themselves in _child.
+    os.environ.setdefault("OMP_NUM_THREADS", "1")
+    try:
+        import torch
+
+        torch.set_num_threads(1)
+    except Exception:
+        pass
 
 
 def _child(rank, size, sock, fn, args, q):
